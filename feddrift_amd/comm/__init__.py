@@ -1,0 +1,85 @@
+"""Distributed communicator: RCCL over xGMI on GPU, gloo on CPU.
+
+Replaces the reference's entire fedml_core/distributed stack (mpi4py send/recv
+threads + 0.3 s polling dispatch, com_manager.py:71-79) with torch.distributed
+collectives: aggregation is ONE all_reduce of the fused [K, P+1] weighted-sum
+tensor per round, overlapping nothing host-side; the tiny control decisions
+(clustering, sampling) are computed lockstep on every rank from identical
+inputs so no control-plane messages are needed at all.
+
+Bucket note: with 7 xGMI links x ~153 GB/s per GPU, ring all-reduce is
+per-link bound; the aggregation tensors here are small (K*P floats), so a
+single fused all_reduce per round is the right shape — one launch, no
+bucketing overhead.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+class Communicator:
+    def __init__(self, device: Optional[torch.device] = None,
+                 backend: str = "auto"):
+        self.rank = int(os.environ.get("RANK", "0"))
+        self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        self.local_rank = int(os.environ.get("LOCAL_RANK", str(self.rank)))
+        if device is not None:
+            self.device = device
+        elif torch.cuda.is_available():
+            torch.cuda.set_device(self.local_rank % torch.cuda.device_count())
+            self.device = torch.device("cuda", torch.cuda.current_device())
+        else:
+            self.device = torch.device("cpu")
+
+        self.distributed = self.world_size > 1
+        if self.distributed and not dist.is_initialized():
+            if backend == "auto":
+                backend = "nccl" if self.device.type == "cuda" else "gloo"
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29571")
+            dist.init_process_group(
+                backend=backend, rank=self.rank, world_size=self.world_size,
+                timeout=datetime.timedelta(seconds=600))
+        self.backend = dist.get_backend() if self.distributed else "local"
+
+    # -- sharding ----------------------------------------------------------
+    def owns_worker(self, w: int) -> bool:
+        return w % self.world_size == self.rank
+
+    def owned_workers(self, n_workers: int) -> List[int]:
+        return [w for w in range(n_workers) if self.owns_worker(w)]
+
+    def owns_client(self, c: int) -> bool:
+        return c % self.world_size == self.rank
+
+    # -- collectives -------------------------------------------------------
+    def all_reduce_(self, t: torch.Tensor) -> torch.Tensor:
+        if self.distributed:
+            dist.all_reduce(t)
+        return t
+
+    def broadcast_(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
+        if self.distributed:
+            dist.broadcast(t, src)
+        return t
+
+    def all_gather_object(self, obj):
+        if not self.distributed:
+            return [obj]
+        out = [None] * self.world_size
+        dist.all_gather_object(out, obj)
+        return out
+
+    def barrier(self) -> None:
+        if self.distributed:
+            dist.barrier()
+
+    @property
+    def is_root(self) -> bool:
+        return self.rank == 0

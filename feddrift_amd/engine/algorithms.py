@@ -1,0 +1,937 @@
+"""Per-algorithm policies: local-training plans, aggregation rules,
+prequential testing, checkpoint wiring.
+
+Each class re-expresses one reference (data_loader, trainer, aggregator)
+triple (fedml_api/distributed/fedavg_ens/*) on the batched engine. The
+DRIFT_ALGO surface matches the reference dispatch
+(FedAvgEnsAPI.py:32-60, :99-141, :150-173):
+
+  aue / auepc / driftsurf / mmacc / mmgeni / mmgeniex / softcluster(+win-1,
+  +reset) / ada / exp / lin / kue, plus 'single' (the fedavg_cont_one
+  single-model baselines win-1/win-2/all driven by --retrain_data).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ..config import (Config, driftsurf_delta, parse_ada_arg,
+                      parse_softcluster_arg)
+from ..data.generators import load_change_points
+from ..drift.ada import AdaState
+from ..drift.driftsurf import DriftSurfState
+from ..drift.kue import KueState
+from ..drift.mmacc import MultiModelAccState
+from ..drift.softcluster import SoftClusterState
+from .fljob import FLJob, TaskList, TrainPlan
+
+
+def make(cfg: Config):
+    algo = cfg.concept_drift_algo
+    if algo in ("aue", "auepc"):
+        return AueAlgo(per_client=(algo == "auepc"))
+    if algo == "driftsurf":
+        return DriftSurfAlgo()
+    if algo in ("mmacc", "mmgeni", "mmgeniex"):
+        return MultiModelAlgo(algo)
+    if cfg.is_softcluster:
+        return SoftClusterAlgo()
+    if algo == "ada":
+        return AdaAlgo()
+    if algo in ("exp", "lin"):
+        return DecayAlgo(algo)
+    if algo == "kue":
+        return KueAlgo()
+    if algo == "single":
+        return SingleAlgo()
+    raise NameError(algo)
+
+
+class AlgoBase:
+    n_models_hint: Optional[int] = None
+
+    # ---- data / model setup ----
+    def build_views(self, job: FLJob) -> list:
+        raise NotImplementedError
+
+    def add_extra_segments(self, job: FLJob) -> None:
+        pass
+
+    def load_checkpoint(self, job: FLJob) -> None:
+        """General model_params.pt reload rule (main_fedavg.py:354-357)."""
+        if job.curr_iter != 0 and not job.cfg.reset_models and \
+                os.path.exists(job.ckpt_path("model_params.pt")):
+            job.load_model_params_general()
+
+    def init_iteration(self, job: FLJob) -> None:
+        pass
+
+    # ---- round phases ----
+    def plan(self, job: FLJob, round_idx: int,
+             client_idx: np.ndarray) -> TrainPlan:
+        return self.default_plan(job, client_idx)
+
+    def aggregate(self, job: FLJob, round_idx: int, plan: TrainPlan,
+                  client_idx: np.ndarray) -> None:
+        job.aggregate(plan)
+
+    def post_aggregate(self, job: FLJob, round_idx: int) -> None:
+        pass
+
+    def test(self, job: FLJob, round_idx: int) -> None:
+        if round_idx % job.cfg.frequency_of_the_test == 0 or \
+                round_idx == job.cfg.comm_round - 1:
+            C = job.cfg.client_num_in_total
+            tr, te = self.client_eval_views(job, np.zeros(C, dtype=int))
+            job.log_round_stats(round_idx, tr, te)
+
+    def finalize(self, job: FLJob) -> None:
+        pass
+
+    # ---- shared helpers ----
+    def default_plan(self, job: FLJob, client_idx: np.ndarray,
+                     lr: Optional[float] = None) -> TrainPlan:
+        """Reference FedAvgEnsTrainer semantics: per model, E random batch
+        picks from its retrain view; num_samples = the view's per-client
+        sample count (FedAvgEnsTrainer.py:47-95)."""
+        E = job.cfg.epochs
+        K = job.n_models
+        rows, offs, lens = [], [], []
+        nW = len(job.owned_workers)
+        sample = np.zeros((nW, K))
+        if lr is not None:
+            job.opt["lr"].fill_(lr)
+        for wi, w in enumerate(job.owned_workers):
+            c = int(client_idx[w])
+            for m in range(K):
+                n = job.views[m].train_n(c)
+                sample[wi, m] = n
+                if n == 0:
+                    continue
+                wins = job.view_train_ref[m][c].windows
+                picks = job.pick_rng.integers(0, len(wins), size=E)
+                rows.append(job.row(wi, m))
+                offs.append([wins[p][0] for p in picks])
+                lens.append([wins[p][1] for p in picks])
+        return TrainPlan(np.asarray(rows, dtype=np.int64),
+                         np.asarray(offs, dtype=np.int64).reshape(-1, E),
+                         np.asarray(lens, dtype=np.int64).reshape(-1, E),
+                         sample)
+
+    def client_eval_views(self, job: FLJob, model_per_client: np.ndarray,
+                          train_model_per_client: Optional[np.ndarray] = None,
+                          train_on_view: bool = True):
+        """test_on_all_clients core: per-client _infer of the chosen model on
+        the view train data (or current-iter all_data) + the test set."""
+        C = job.cfg.client_num_in_total
+        if train_model_per_client is None:
+            train_model_per_client = model_per_client
+        clients = range(C) if job.cfg.ci != 1 else range(1)
+        tl = TaskList()
+        for c in clients:
+            tid_tr = tl.new_task()
+            tid_te = tl.new_task()
+            if job.comm.owns_client(c):
+                mt = int(train_model_per_client[c])
+                me = int(model_per_client[c])
+                if train_on_view:
+                    if c in job.view_train_ref[mt]:
+                        tl.add_windows(tid_tr, mt,
+                                       job.view_train_ref[mt][c].windows)
+                else:
+                    tl.add_windows(tid_tr, mt,
+                                   job.all_ref[c][job.curr_iter].windows)
+                if c in job.test_ref:
+                    tl.add_windows(tid_te, me, job.test_ref[c].windows)
+        correct, total, loss, _ = job.run_eval(job.global_params, tl)
+        buf = torch.from_numpy(np.stack([correct, total, loss])).to(job.device)
+        job.comm.all_reduce_(buf)
+        correct, total, loss = buf.cpu().numpy()
+        tr = (correct[0::2], total[0::2], loss[0::2])
+        te = (correct[1::2], total[1::2], loss[1::2])
+        return tr, te
+
+
+# ---------------------------------------------------------------------------
+# hooks adapter: drift state machines -> engine
+# ---------------------------------------------------------------------------
+class Hooks:
+    """EngineHooks implementation over an FLJob (see drift/softcluster.py)."""
+
+    def __init__(self, job: FLJob):
+        self.job = job
+
+    def train_acc_matrix(self, models_in_use):
+        return self.job.train_acc_matrix_rows(list(models_in_use))
+
+    def cluster_pair_acc(self, models_in_use, cluster_batches):
+        return self.job.cluster_pair_acc_windows(list(models_in_use),
+                                                 cluster_batches)
+
+    def pooled_cluster_batches(self, weights, model, curr_iter):
+        return self.job.pooled_cluster_windows(weights, model, curr_iter)
+
+    def merge_models(self, base, second, w1, w2):
+        self.job.merge_models(base, second, w1, w2)
+
+    def reinit_model(self, m):
+        self.job.reinit_model(m)
+
+    def copy_model(self, dst, src):
+        self.job.copy_model(dst, src)
+
+    def log_client(self, key_fmt, client, value, round_idx):
+        if self.job.comm.is_root:
+            self.job.logger.log({key_fmt.format(client): value}, round_idx)
+
+    def log_summary(self, key, value):
+        if self.job.comm.is_root:
+            self.job.logger.set_summary(key, value)
+
+
+# ---------------------------------------------------------------------------
+# SoftCluster family: FedDrift (H_*), FedDrift-Eager (mmacc), IFCA (hard),
+# CFL (cfl_*), softmax, gmm, geni
+# ---------------------------------------------------------------------------
+class SoftClusterAlgo(AlgoBase):
+    def build_views(self, job: FLJob) -> list:
+        # views are win-1 placeholders; the trainer uses all_data
+        # (reference SoftCluster_data_loader:1334-1341)
+        v = job.cdata.view("win-1")
+        return [v for _ in range(job.cfg.concept_num)]
+
+    def _state_path(self, job):
+        return job.ckpt_path("sc_state.pkl")
+
+    def init_iteration(self, job: FLJob) -> None:
+        cfg = job.cfg
+        hooks = Hooks(job)
+        if job.curr_iter == 0 or not os.path.exists(self._state_path(job)):
+            p = parse_softcluster_arg(cfg.concept_drift_algo_arg, cfg.dataset_norm)
+            st = SoftClusterState(
+                cfg.client_num_in_total, cfg.concept_num, p.cluster_alg,
+                p.mmacc_delta, p.softmax_alpha,
+                load_change_points(cfg.data_dir, cfg.change_points)
+                if p.cluster_alg == "geni" else None,
+                cfg.time_stretch, p.h_delta, p.h_deltap, p.h_w, p.h_distance,
+                p.h_cluster, p.cfl_gamma, p.cfl_retrain, seed=cfg.dummy_arg)
+        else:
+            st = job.load_state_pickle("sc_state.pkl")
+        self.state = st
+        arg = cfg.concept_drift_algo_arg
+
+        # iteration-start clustering (FedAvgEnsAggregatorSoftCluster.py:46-118)
+        if "H" in arg:
+            if job.curr_iter == 0:
+                st.cluster_init(hooks)
+            else:
+                st.cluster_hierarchical(hooks, job.curr_iter)
+        elif "cfl" in arg:
+            if job.curr_iter == 0:
+                st.cluster_init(hooks)
+            else:
+                st.cluster_cfl_init(hooks, job.curr_iter)
+        elif "hard" in arg:
+            if job.curr_iter == 0:
+                job.randomize_models_unseeded()
+            acc = job.train_acc_matrix_rows(list(range(job.n_models)))
+            st.cluster(hooks, acc, job.curr_iter, 0)
+        elif "mmacc" in arg:
+            if job.curr_iter == 0:
+                st.cluster_init(hooks)
+            else:
+                st.cluster_mmacc2(hooks, job.curr_iter)
+        else:
+            if job.curr_iter == 0:
+                st.cluster_init(hooks)
+            else:
+                if cfg.concept_drift_algo == "softclusterreset":
+                    # delete epsilon-dominated models
+                    # (FedAvgEnsAggregatorSoftCluster.py:85-97)
+                    acc = job.train_acc_matrix_rows(list(range(job.n_models)))
+                    deleted: List[int] = []
+                    for m in reversed(range(job.n_models)):
+                        rest = np.delete(acc, deleted + [m], axis=0)
+                        if rest.shape[0] > 0 and \
+                                np.all(acc[m] < np.max(rest, axis=0) + 0.01):
+                            deleted.append(m)
+                            hooks.log_summary(f"Reset-{m}", 1)
+                            st.set_weights_zero_model(m)
+                            job.reinit_model(m)
+                acc = job.train_acc_matrix_rows(list(range(job.n_models)))
+                st.cluster(hooks, acc, job.curr_iter, 0)
+
+        if cfg.concept_drift_algo == "softclusterwin-1":
+            st.set_weights_win1(job.curr_iter)
+
+        # record accuracy at first iteration so the drift detector starts
+        # initialized (FedAvgEnsAggregatorSoftCluster.py:105-116)
+        if job.curr_iter == 0:
+            accs = job.train_acc_matrix_rows(list(range(job.n_models)))
+            for c in range(cfg.client_num_in_total):
+                st.set_acc(c, accs[st.get_test_model_idx(0, c)][c])
+
+        job.save_state_pickle("sc_state.pkl", st)
+
+    def plan(self, job: FLJob, round_idx: int,
+             client_idx: np.ndarray) -> TrainPlan:
+        """TrainerSoftCluster: pairs weighted by sc_weights x iteration batch
+        counts; unit-weight fast path pools batches across selected
+        iterations (FedAvgEnsTrainerSoftCluster.py:63-135). NOTE the
+        reference weighs by len(all_local_data[t]) = the BATCH count of
+        iteration t, not the sample count — preserved."""
+        st = self.state
+        E = job.cfg.epochs
+        K = job.n_models
+        T = job.curr_iter + 1
+        rows, offs, lens = [], [], []
+        nW = len(job.owned_workers)
+        sample = np.zeros((nW, K))
+        w_iter = st.get_weights()
+        for wi, w in enumerate(job.owned_workers):
+            c = int(client_idx[w])
+            for m in range(K):
+                if not np.any(w_iter[job.curr_iter][m]):
+                    continue
+                unnorm = np.array([
+                    w_iter[t][m][c] * len(job.all_ref[c][t].windows)
+                    for t in range(T)])
+                tot = unnorm.sum()
+                if tot == 0:
+                    continue
+                sample[wi, m] = tot
+                pool = []
+                for t in range(T):
+                    if unnorm[t] > 0:
+                        pool.extend(job.all_ref[c][t].windows)
+                picks = job.pick_rng.integers(0, len(pool), size=E)
+                rows.append(job.row(wi, m))
+                offs.append([pool[p][0] for p in picks])
+                lens.append([pool[p][1] for p in picks])
+        return TrainPlan(np.asarray(rows, dtype=np.int64),
+                         np.asarray(offs, dtype=np.int64).reshape(-1, E),
+                         np.asarray(lens, dtype=np.int64).reshape(-1, E),
+                         sample)
+
+    def aggregate(self, job: FLJob, round_idx: int, plan: TrainPlan,
+                  client_idx: np.ndarray) -> None:
+        st = self.state
+        if "cfl" in job.cfg.concept_drift_algo_arg:
+            if self._cfl_round(job, round_idx, plan, client_idx):
+                return  # split: skip this round's aggregation (:141-148)
+        mask = np.array([bool(np.any(st.get_weights()[job.curr_iter][m]))
+                         for m in range(job.n_models)])
+        job.aggregate(plan, model_mask=mask)
+        if job.cfg.concept_drift_algo_arg == "hard-r":
+            acc = job.train_acc_matrix_rows(list(range(job.n_models)))
+            st.cluster(Hooks(job), acc, job.curr_iter, round_idx + 1)
+
+    def _cfl_round(self, job: FLJob, round_idx: int, plan: TrainPlan,
+                   client_idx: np.ndarray) -> bool:
+        """CFL split check needs each client's raw weight update; gather the
+        flattened deltas (small: active pairs only) to every rank
+        (reference cluster_cfl, FedAvgEnsDataLoader.py:1159-1223)."""
+        st = self.state
+        K, P = job.n_models, job.spec.n_params
+        local: Dict[tuple, np.ndarray] = {}
+        nW = len(job.owned_workers)
+        reps = job.replicas.reshape(nW, K, P)
+        for wi, w in enumerate(job.owned_workers):
+            for m in range(K):
+                if plan.sample_num[wi, m] > 0:
+                    delta = (reps[wi, m] - job.global_params[m]).cpu().numpy()
+                    local[(int(client_idx[w]), m)] = delta
+        gathered = job.comm.all_gather_object(local)
+        updates: Dict[tuple, np.ndarray] = {}
+        for d in gathered:
+            updates.update(d)
+        models_in_use = [m for m in range(K)
+                         if np.any(st.get_weights()[job.curr_iter][m] > 0)]
+        clients_by_model = {
+            m: np.nonzero(st.get_weights()[job.curr_iter][m])[0]
+            for m in models_in_use}
+        upd_by_model = {
+            m: [updates[(c, m)] for c in clients_by_model[m]
+                if (c, m) in updates]
+            for m in models_in_use}
+        # keep client lists aligned with the updates actually present
+        clients_aligned = {
+            m: np.array([c for c in clients_by_model[m]
+                         if (c, m) in updates])
+            for m in models_in_use}
+        return st.cluster_cfl(Hooks(job), job.curr_iter, round_idx + 1,
+                              upd_by_model, clients_aligned)
+
+    def test(self, job: FLJob, round_idx: int) -> None:
+        cfg = job.cfg
+        if round_idx % cfg.frequency_of_the_test == 0 or \
+                round_idx == cfg.comm_round - 1:
+            st = self.state
+            C = cfg.client_num_in_total
+            mpc = np.array([st.get_test_model_idx(job.curr_iter, c)
+                            for c in range(C)])
+            # softcluster tests train data on the CURRENT-iteration all_data
+            # (FedAvgEnsAggregatorSoftCluster.py:227-231)
+            tr, te = self.client_eval_views(job, mpc, train_on_view=False)
+            job.log_round_stats(round_idx, tr, te)
+        if round_idx > (cfg.comm_round - 5):
+            job.save_state_pickle("sc_state.pkl", self.state)
+
+    def finalize(self, job: FLJob) -> None:
+        job.save_state_pickle("sc_state.pkl", self.state)
+
+
+# ---------------------------------------------------------------------------
+# Single-model baselines (fedavg_cont_one: win-1 / win-2 / all via
+# --retrain_data; reference fedml_experiments/distributed/fedavg_cont_one)
+# ---------------------------------------------------------------------------
+class SingleAlgo(AlgoBase):
+    def build_views(self, job: FLJob) -> list:
+        return [job.cdata.view(job.cfg.retrain_data)]
+
+
+# ---------------------------------------------------------------------------
+# AUE / AUE-PC
+# ---------------------------------------------------------------------------
+class AueAlgo(AlgoBase):
+    EPS = 1e-20
+
+    def __init__(self, per_client: bool):
+        self.per_client = per_client
+
+    def build_views(self, job: FLJob) -> list:
+        # model m trains on win-(m+1) (reference AUE_data_loader:20-29)
+        n = min(job.curr_iter + 1, job.cfg.ensemble_window)
+        return [job.cdata.view(f"win-{m + 1}") for m in range(n)]
+
+    def load_checkpoint(self, job: FLJob) -> None:
+        # circular shift: model m loads previous model m-1
+        # (main_fedavg.py:342-344)
+        if job.curr_iter != 0 and not job.cfg.reset_models and \
+                os.path.exists(job.ckpt_path("model_params.pt")):
+            mp = torch.load(job.ckpt_path("model_params.pt"))
+            for m_idx in range(1, job.n_models):
+                if m_idx - 1 in mp:
+                    job.global_params[m_idx] = job.packer.flatten(
+                        mp[m_idx - 1]).to(job.device)
+
+    def init_iteration(self, job: FLJob) -> None:
+        # all models start "perfect" (FedAvgEnsAggregatorAue.py:45-54)
+        py = 1.0 / job.dataset.class_num
+        mser = (1 - py) ** 2
+        M, C = job.n_models, job.cfg.client_num_in_total
+        if self.per_client:
+            w = np.full((C, M), 1.0 / (mser + self.EPS))
+            self.ens_weights = w / w.sum(axis=1, keepdims=True)
+        else:
+            w = np.full(M, 1.0 / (mser + self.EPS))
+            self.ens_weights = w / w.sum()
+
+    def _update_ens_weights(self, job: FLJob) -> None:
+        """AUE weights w_m = 1/(MSE_r + MSE_m + eps) over the newest batch;
+        newest model gets the perfect score. The reference's index shift
+        (enumerate over models[1:] writing ens_weights[m_idx]) is preserved
+        verbatim (FedAvgEnsAggregatorAue.py:55-87)."""
+        py = 1.0 / job.dataset.class_num
+        mser = (1 - py) ** 2
+        M, C = job.n_models, job.cfg.client_num_in_total
+        # per-(model>=1, client) MSE on the win-1 view train data
+        tl = TaskList()
+        ids = {}
+        for mi in range(1, M):
+            for c in range(C):
+                tid = tl.new_task()
+                ids[(mi, c)] = tid
+                if job.comm.owns_client(c) and c in job.view_train_ref[0]:
+                    tl.add_windows(tid, mi, job.view_train_ref[0][c].windows)
+        _, total, _, mse = job.run_eval(job.global_params, tl, want_mse=True)
+        buf = torch.from_numpy(np.stack([total, mse])).to(job.device)
+        job.comm.all_reduce_(buf)
+        total, mse = buf.cpu().numpy()
+
+        if self.per_client:
+            for c in range(C):
+                for mi in range(1, M):
+                    tid = ids[(mi, c)]
+                    msei = mse[tid] / total[tid] if total[tid] else 0.0
+                    self.ens_weights[c][mi - 1] = 1.0 / (mser + msei + self.EPS)
+                self.ens_weights[c][0] = 1.0 / (mser + self.EPS)
+                self.ens_weights[c] = self.ens_weights[c] / \
+                    self.ens_weights[c].sum()
+        else:
+            for mi in range(1, M):
+                msei = sum(mse[ids[(mi, c)]] for c in range(C))
+                tot = sum(total[ids[(mi, c)]] for c in range(C))
+                msei = msei / tot if tot else 0.0
+                self.ens_weights[mi - 1] = 1.0 / (mser + msei + self.EPS)
+            self.ens_weights[0] = 1.0 / (mser + self.EPS)
+            self.ens_weights = self.ens_weights / self.ens_weights.sum()
+
+    def post_aggregate(self, job: FLJob, round_idx: int) -> None:
+        if round_idx % 10 == 0 or round_idx > (job.cfg.comm_round - 10):
+            self._update_ens_weights(job)
+
+    def test(self, job: FLJob, round_idx: int) -> None:
+        cfg = job.cfg
+        if not (round_idx % cfg.frequency_of_the_test == 0 or
+                round_idx == cfg.comm_round - 1):
+            return
+        C = cfg.client_num_in_total
+        clients = range(C) if cfg.ci != 1 else range(1)
+        # train: model 0 on view-0 train data (FedAvgEnsAggregatorAue.py:172)
+        tl = TaskList()
+        for c in clients:
+            tid = tl.new_task()
+            if job.comm.owns_client(c) and c in job.view_train_ref[0]:
+                tl.add_windows(tid, 0, job.view_train_ref[0][c].windows)
+        trc, trt, trl, _ = job.run_eval(job.global_params, tl)
+        # test: weighted-vote ensemble (:256-283)
+        tec = np.zeros(len(list(clients)))
+        tet = np.zeros(len(list(clients)))
+        M = job.n_models
+        for i, c in enumerate(clients):
+            if not job.comm.owns_client(c) or c not in job.test_ref:
+                continue
+            w = self.ens_weights[c] if self.per_client else self.ens_weights
+            cc, tt = job.backend.ens_vote_eval(
+                job.spec, job.global_params,
+                torch.as_tensor(w[:M], dtype=torch.float32,
+                                device=job.device),
+                job.arena.x, job.arena.y, job.test_ref[c].windows,
+                mode="hard")
+            tec[i], tet[i] = cc, tt
+        buf = torch.from_numpy(np.stack(
+            [trc, trt, trl, tec, tet])).to(job.device)
+        job.comm.all_reduce_(buf)
+        trc, trt, trl, tec, tet = buf.cpu().numpy()
+        job.log_round_stats(round_idx, (trc, trt, trl),
+                            (tec, tet, np.zeros_like(tec)))
+
+
+# ---------------------------------------------------------------------------
+# DriftSurf
+# ---------------------------------------------------------------------------
+class DriftSurfAlgo(AlgoBase):
+    def build_views(self, job: FLJob) -> list:
+        cfg = job.cfg
+        if job.curr_iter == 0:
+            self.state = DriftSurfState(
+                delta=driftsurf_delta(cfg.concept_drift_algo_arg,
+                                      cfg.dataset_norm))
+            views = [job.cdata.view("sel-0") for _ in range(2)]
+        else:
+            self.state = job.load_state_pickle("ds_state.pkl")
+            newest = job.cdata.view("win-1")
+            self._score_newest(job, newest)
+            views = []
+            for key in self.state.get_train_keys():
+                sel = ",".join(str(x) for x in self.state.get_train_data(key))
+                views.append(job.cdata.view(f"sel-{sel}"))
+        job.save_state_pickle("ds_state.pkl", self.state)
+        return views
+
+    def _score_newest(self, job: FLJob, view) -> None:
+        """run_ds_algo at data-load time (DriftSurf_data_loader:269-314):
+        score the pickled models on the newest global batch (CPU eval; runs
+        once per iteration, data is host-side at this point)."""
+        from ..ops import mlp_torch
+        spec = job.spec
+
+        def score(key: str) -> float:
+            flat = self.state.models.get(key)
+            if flat is None:
+                return 0.0
+            params = torch.from_numpy(np.asarray(flat, dtype=np.float32)) \
+                .reshape(1, -1)
+            correct = total = 0.0
+            for c, seg in view.train.items():
+                if seg.n == 0:
+                    continue
+                x = torch.from_numpy(seg.x)
+                y = torch.from_numpy(seg.y)
+                logits = mlp_torch.forward_logits(spec, params,
+                                                  x.unsqueeze(0))
+                correct += (logits.argmax(-1).squeeze(0) == y).sum().item()
+                total += seg.n
+            return correct / total if total else 0.0
+
+        self.state.run_ds_algo(score, job.curr_iter)
+
+    def load_checkpoint(self, job: FLJob) -> None:
+        # handled via ds_state (FedAvgEnsAggregatorDriftSurf.py:45-64)
+        st = self.state
+        self.test_model_idx = 0
+        if job.curr_iter != 0 and not job.cfg.reset_models:
+            for idx, key in enumerate(st.get_train_keys()):
+                flat = st.models.get(key)
+                if flat is not None and idx < job.n_models:
+                    job.global_params[idx] = torch.from_numpy(
+                        np.asarray(flat, dtype=np.float32)).to(job.device)
+        for idx, key in enumerate(st.get_train_keys()):
+            if key == st.get_model_key():
+                self.test_model_idx = idx
+
+    def test(self, job: FLJob, round_idx: int) -> None:
+        cfg = job.cfg
+        if round_idx % cfg.frequency_of_the_test == 0 or \
+                round_idx == cfg.comm_round - 1:
+            C = cfg.client_num_in_total
+            mpc = np.full(C, self.test_model_idx)
+            tr, te = self.client_eval_views(job, mpc, mpc)
+            job.log_round_stats(round_idx, tr, te)
+
+    def _sync_state_models(self, job: FLJob) -> None:
+        for idx, key in enumerate(self.state.get_train_keys()):
+            if idx < job.n_models:
+                self.state.set_model(
+                    key, job.global_params[idx].cpu().numpy())
+
+    def aggregate(self, job: FLJob, round_idx: int, plan: TrainPlan,
+                  client_idx: np.ndarray) -> None:
+        job.aggregate(plan)
+        self._sync_state_models(job)
+        job.save_state_pickle("ds_state.pkl", self.state)
+
+    def finalize(self, job: FLJob) -> None:
+        self._sync_state_models(job)
+        job.save_state_pickle("ds_state.pkl", self.state)
+
+
+# ---------------------------------------------------------------------------
+# MultiModelAcc (mmacc) + oracles (mmgeni / mmgeniex)
+# ---------------------------------------------------------------------------
+class MultiModelAlgo(AlgoBase):
+    def __init__(self, variant: str):
+        self.variant = variant
+
+    def build_views(self, job: FLJob) -> list:
+        cfg = job.cfg
+        from ..config import DEFAULT_DELTAS
+        if job.curr_iter == 0:
+            self.state = MultiModelAccState(
+                cfg.client_num_in_total, cfg.concept_num,
+                DEFAULT_DELTAS.get(cfg.dataset_norm, 0.1))
+        else:
+            self.state = job.load_state_pickle("mm_state.pkl")
+
+        if self.variant == "mmacc":
+            if job.curr_iter == 0:
+                self.state.run_model_select(None, 0)
+            else:
+                newest = job.cdata.view("win-1")
+                self._run_select(job, newest)
+        else:
+            cps = load_change_points(cfg.data_dir, cfg.change_points)
+            if self.variant == "mmgeni":
+                self.state.model_select_geni(job.curr_iter, cps,
+                                             cfg.time_stretch)
+            else:
+                self.state.model_select_geniex(job.curr_iter, cps,
+                                               cfg.time_stretch)
+
+        views = []
+        self.model_ids = []   # original model id per view position
+        for m in range(cfg.concept_num):
+            td = self.state.get_train_data_by_model(m)
+            if td != "":
+                views.append(job.cdata.view("clientsel-" + td))
+                self.model_ids.append(m)
+        job.save_state_pickle("mm_state.pkl", self.state)
+        return views
+
+    def _run_select(self, job: FLJob, newest_view) -> None:
+        """run_model_select with per-(model, client) scoring on the newest
+        local batch (reference FedAvgEnsDataLoader.py:350-390)."""
+        from ..ops import mlp_torch
+        spec = job.spec
+        cache: Dict[tuple, float] = {}
+
+        def score(m: int, c: int) -> float:
+            if (m, c) in cache:
+                return cache[(m, c)]
+            flat = self.state.models.get(m)
+            if flat is None or c not in newest_view.train:
+                return 0.0
+            seg = newest_view.train[c]
+            if seg.n == 0:
+                return 0.0
+            params = torch.from_numpy(np.asarray(flat, dtype=np.float32)) \
+                .reshape(1, -1)
+            logits = mlp_torch.forward_logits(
+                spec, params, torch.from_numpy(seg.x).unsqueeze(0))
+            acc = float((logits.argmax(-1).squeeze(0) ==
+                         torch.from_numpy(seg.y)).sum().item()) / seg.n
+            cache[(m, c)] = acc
+            return acc
+
+        self.state.run_model_select(score, job.curr_iter)
+
+    def load_checkpoint(self, job: FLJob) -> None:
+        cfg = job.cfg
+        if job.curr_iter != 0 and not cfg.reset_models and \
+                os.path.exists(job.ckpt_path("model_params.pt")):
+            mp = torch.load(job.ckpt_path("model_params.pt"))
+            if self.variant in ("mmacc", "mmgeni", "mmgeniex") and \
+                    job.n_models == 1 and len(mp) == 2:
+                # post-transition special case (main_fedavg.py:337-339)
+                job.global_params[0] = job.packer.flatten(mp[1]).to(job.device)
+            else:
+                for m_idx, sd in mp.items():
+                    if m_idx < job.n_models:
+                        job.global_params[m_idx] = job.packer.flatten(sd).to(
+                            job.device)
+
+    def _sync_state_models(self, job: FLJob) -> None:
+        for pos, m in enumerate(self.model_ids):
+            self.state.set_model(m, job.global_params[pos].cpu().numpy())
+
+    def init_iteration(self, job: FLJob) -> None:
+        # record accuracy for the drift detector at iteration 0
+        # (reference records via test_on_all_clients; mmacc uses acc_dict)
+        if self.variant == "mmacc" and job.curr_iter == 0:
+            accs = job.train_acc_matrix_rows([0])
+            for c in range(job.cfg.client_num_in_total):
+                self.state.set_acc(c, accs[0][c])
+            job.save_state_pickle("mm_state.pkl", self.state)
+
+    def test(self, job: FLJob, round_idx: int) -> None:
+        cfg = job.cfg
+        if round_idx % cfg.frequency_of_the_test == 0 or \
+                round_idx == cfg.comm_round - 1:
+            C = cfg.client_num_in_total
+            pos_of = {m: i for i, m in enumerate(self.model_ids)}
+            mpc = np.array([pos_of.get(self.state.get_test_model_idx(c), 0)
+                            for c in range(C)])
+            tpc = np.array([pos_of.get(self.state.get_train_model_idx(c), 0)
+                            for c in range(C)])
+            tr, te = self.client_eval_views(job, mpc, tpc)
+            job.log_round_stats(round_idx, tr, te)
+        if round_idx > (cfg.comm_round - 5):
+            self._sync_state_models(job)
+            job.save_state_pickle("mm_state.pkl", self.state)
+
+    def finalize(self, job: FLJob) -> None:
+        self._sync_state_models(job)
+        job.save_state_pickle("mm_state.pkl", self.state)
+
+
+# ---------------------------------------------------------------------------
+# Adaptive-FedAvg
+# ---------------------------------------------------------------------------
+class AdaAlgo(AlgoBase):
+    def build_views(self, job: FLJob) -> list:
+        retrain, gran = parse_ada_arg(job.cfg.concept_drift_algo_arg)
+        self.update_each_round = (gran == "round")
+        return [job.cdata.view(retrain)]
+
+    def init_iteration(self, job: FLJob) -> None:
+        if job.curr_iter == 0 or not os.path.exists(
+                job.ckpt_path("ada_state.pkl")):
+            self.state = AdaState(init_lr=job.cfg.lr)
+        else:
+            self.state = job.load_state_pickle("ada_state.pkl")
+        job.save_state_pickle("ada_state.pkl", self.state)
+
+    def plan(self, job: FLJob, round_idx: int,
+             client_idx: np.ndarray) -> TrainPlan:
+        # trainer uses the server-determined lr (FedAvgEnsTrainerAda.py:62-65)
+        return self.default_plan(job, client_idx,
+                                 lr=self.state.current_lr())
+
+    def aggregate(self, job: FLJob, round_idx: int, plan: TrainPlan,
+                  client_idx: np.ndarray) -> None:
+        job.aggregate(plan)
+        theta = job.global_params[0].cpu().numpy()
+        if self.update_each_round:
+            t = round_idx + job.curr_iter * job.cfg.comm_round
+            self.state.update(theta, t)
+        else:
+            if round_idx == job.cfg.comm_round - 5:
+                self.state.update(theta, job.curr_iter)
+
+    def test(self, job: FLJob, round_idx: int) -> None:
+        super().test(job, round_idx)
+        if round_idx > (job.cfg.comm_round - 5):
+            job.save_state_pickle("ada_state.pkl", self.state)
+
+
+# ---------------------------------------------------------------------------
+# Exp / Lin time-decay single-model baselines
+# ---------------------------------------------------------------------------
+class DecayAlgo(AlgoBase):
+    def __init__(self, kind: str):
+        self.kind = kind
+
+    def build_views(self, job: FLJob) -> list:
+        return [job.cdata.view("win-1")]
+
+    def plan(self, job: FLJob, round_idx: int,
+             client_idx: np.ndarray) -> TrainPlan:
+        """Single model trained on all history with time-decay sampling:
+        probs ~ 2**t (exp) or t+1 (lin); per step sample an iteration then a
+        uniform batch within it (FedAvgEnsTrainerExp.py:47-96 / Lin:66)."""
+        E = job.cfg.epochs
+        T = job.curr_iter + 1
+        rows, offs, lens = [], [], []
+        nW = len(job.owned_workers)
+        sample = np.zeros((nW, 1))
+        unnorm = np.array([2.0 ** t if self.kind == "exp" else t + 1.0
+                           for t in range(T)])
+        probs = unnorm / unnorm.sum()
+        for wi, w in enumerate(job.owned_workers):
+            c = int(client_idx[w])
+            total_batches = sum(len(job.all_ref[c][t].windows)
+                                for t in range(T))
+            if total_batches == 0:
+                continue
+            sample[wi, 0] = total_batches
+            so, sl = [], []
+            for _ in range(E):
+                t = int(job.pick_rng.choice(T, p=probs))
+                wins = job.all_ref[c][t].windows
+                if len(wins) == 0:
+                    so.append(0)
+                    sl.append(0)   # reference skips the step (:73-74)
+                    continue
+                o, l = wins[int(job.pick_rng.integers(0, len(wins)))]
+                so.append(o)
+                sl.append(l)
+            rows.append(job.row(wi, 0))
+            offs.append(so)
+            lens.append(sl)
+        return TrainPlan(np.asarray(rows, dtype=np.int64),
+                         np.asarray(offs, dtype=np.int64).reshape(-1, E),
+                         np.asarray(lens, dtype=np.int64).reshape(-1, E),
+                         sample)
+
+
+# ---------------------------------------------------------------------------
+# KUE
+# ---------------------------------------------------------------------------
+class KueAlgo(AlgoBase):
+    def build_views(self, job: FLJob) -> list:
+        # poisson bootstrap view per ensemble member
+        # (Kue_data_loader:58-72)
+        return [job.cdata.view("poisson")
+                for _ in range(job.cfg.concept_num)]
+
+    def init_iteration(self, job: FLJob) -> None:
+        cfg = job.cfg
+        if job.curr_iter == 0 or not os.path.exists(
+                job.ckpt_path("kue_state.pkl")):
+            self.state = KueState(cfg.concept_num, job.dataset.feature_num,
+                                  rng=np.random.RandomState(cfg.dummy_arg))
+        else:
+            self.state = job.load_state_pickle("kue_state.pkl")
+            # re-draw the worst model's mask + params
+            # (FedAvgEnsAggregatorKue.py:47-57)
+            worst = self.state.get_worst_idx()
+            self.state.initialize_mask(worst)
+            job.reinit_model(worst)
+        job.save_state_pickle("kue_state.pkl", self.state)
+        self.ens_weights = np.ones(job.n_models)
+
+    def _masks_tensor(self, job: FLJob) -> torch.Tensor:
+        return torch.as_tensor(
+            self.state.get_masks().astype(np.float32), device=job.device)
+
+    def plan(self, job: FLJob, round_idx: int,
+             client_idx: np.ndarray) -> TrainPlan:
+        plan = self.default_plan(job, client_idx)
+        if plan.rows.size:
+            masks = self._masks_tensor(job)
+            model_of_row = torch.as_tensor(plan.rows % job.n_models,
+                                           dtype=torch.int64,
+                                           device=job.device)
+            plan.x_mask = masks[model_of_row]
+        return plan
+
+    def aggregate(self, job: FLJob, round_idx: int, plan: TrainPlan,
+                  client_idx: np.ndarray) -> None:
+        job.aggregate(plan)
+        if round_idx % 10 == 0 or round_idx > (job.cfg.comm_round - 10):
+            self._update_ens_weights(job)
+
+    def _update_ens_weights(self, job: FLJob) -> None:
+        """Cohen's kappa per model from the global confusion matrix over the
+        model's own (masked) train view (FedAvgEnsAggregatorKue.py:59-76)."""
+        O = job.dataset.class_num
+        masks = self._masks_tensor(job)
+        for m in range(job.n_models):
+            tl = TaskList()
+            tid = tl.new_task()
+            for c in range(job.cfg.client_num_in_total):
+                if job.comm.owns_client(c) and c in job.view_train_ref[m]:
+                    tl.add_windows(tid, m, job.view_train_ref[m][c].windows)
+            A = job.backend.confusion_tasks(
+                job.spec, job.global_params, job.arena.x, job.arena.y,
+                torch.as_tensor(tl.task_row, dtype=torch.int64,
+                                device=job.device),
+                torch.as_tensor(tl.task_id, dtype=torch.int64,
+                                device=job.device),
+                torch.as_tensor(tl.off, dtype=torch.int64, device=job.device),
+                torch.as_tensor(tl.ln, dtype=torch.int64, device=job.device),
+                1, O, x_mask=masks[m])
+            job.comm.all_reduce_(A)
+            An = A[0].cpu().numpy()
+            n = An.sum()
+            left = np.trace(An)
+            right = sum(An[i, :].sum() * An[:, i].sum() for i in range(O))
+            denom = n ** 2 - right
+            self.ens_weights[m] = (n * left - right) / denom if denom else 0.0
+        if job.curr_iter != 0:
+            self.state.set_worst_idx(int(np.argmin(self.ens_weights)))
+
+    def plan_masks(self, job: FLJob, plan: TrainPlan) -> torch.Tensor:
+        masks = self._masks_tensor(job)
+        model_of_row = torch.as_tensor(
+            plan.rows % job.n_models, dtype=torch.int64, device=job.device)
+        return masks[model_of_row]
+
+    def test(self, job: FLJob, round_idx: int) -> None:
+        cfg = job.cfg
+        if not (round_idx % cfg.frequency_of_the_test == 0 or
+                round_idx == cfg.comm_round - 1):
+            return
+        C = cfg.client_num_in_total
+        clients = range(C) if cfg.ci != 1 else range(1)
+        tl = TaskList()
+        for c in clients:
+            tid = tl.new_task()
+            if job.comm.owns_client(c) and c in job.view_train_ref[0]:
+                tl.add_windows(tid, 0, job.view_train_ref[0][c].windows)
+        trc, trt, trl, _ = job.run_eval(job.global_params, tl)
+        # soft-vote ensemble with masks, excluding the worst model
+        # (FedAvgEnsAggregatorKue.py:234-264)
+        masks = self._masks_tensor(job)
+        include = np.array([m != self.state.get_worst_idx() and
+                            self.ens_weights[m] > 0
+                            for m in range(job.n_models)])
+        w = np.where(include, self.ens_weights, 0.0)
+        nc = len(list(clients))
+        tec = np.zeros(nc)
+        tet = np.zeros(nc)
+        for i, c in enumerate(clients):
+            if not job.comm.owns_client(c) or c not in job.test_ref:
+                continue
+            cc, tt = job.backend.ens_vote_eval(
+                job.spec, job.global_params,
+                torch.as_tensor(w, dtype=torch.float32, device=job.device),
+                job.arena.x, job.arena.y, job.test_ref[c].windows,
+                mode="soft", masks=masks)
+            tec[i], tet[i] = cc, tt
+        buf = torch.from_numpy(np.stack(
+            [trc, trt, trl, tec, tet])).to(job.device)
+        job.comm.all_reduce_(buf)
+        trc, trt, trl, tec, tet = buf.cpu().numpy()
+        job.log_round_stats(round_idx, (trc, trt, trl),
+                            (tec, tet, np.zeros_like(tec)))
+        if round_idx > (cfg.comm_round - 5):
+            job.save_state_pickle("kue_state.pkl", self.state)
+
+    def finalize(self, job: FLJob) -> None:
+        job.save_state_pickle("kue_state.pkl", self.state)

@@ -1,0 +1,428 @@
+"""FLJob: one training iteration (= one reference MPI job) on the new engine.
+
+One process per GPU; worker slots are sharded across ranks
+(w % world_size == rank). Each round:
+
+  1. sync_replicas: every owned (worker, model) replica row <- global model
+     row (a device-side broadcast copy; the reference re-sends pickled
+     state_dicts to every worker process instead,
+     FedAvgEnsServerManager.py:66-67).
+  2. the algorithm plans which pairs train on which minibatch windows
+     (host-side RNG, statistical parity with the reference's client-side
+     np.random.choice picks, FedAvgEnsTrainer.py:67).
+  3. ops.train_fused runs ALL local training for this rank in one batched
+     op (HIP kernel on GPU / vectorized torch on CPU).
+  4. aggregation: one fused [K, P+1] weighted-sum + all_reduce over
+     RCCL/gloo (replaces FedAvgEnsAggregatorSoftCluster.py:148-195).
+  5. algorithm post-aggregation hooks (clustering, ensemble weights, lr
+     adaptation) run LOCKSTEP on every rank from identical allreduced
+     inputs — no control-plane messages at all.
+  6. prequential evaluation, sharded by client + allreduced.
+
+Checkpoint layout is reference-compatible: model_params.pt =
+dict(enumerate(state_dicts)) + per-algorithm state pickles
+(FedAvgEnsServerManager.py:84-86).
+"""
+
+from __future__ import annotations
+
+import os
+import pickle
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from .. import ops
+from ..comm import Communicator
+from ..config import Config
+from ..data.loader import ClientData, DriftDataset, RetrainView
+from ..eval.metrics import MetricLogger
+from ..models import packed, zoo
+from .arena import DeviceArena, SegRef
+
+
+@dataclass
+class TrainPlan:
+    """Task arrays for one round of local training on this rank."""
+    rows: np.ndarray            # [G] row ids into the replica buffer
+    step_off: np.ndarray        # [G, E]
+    step_len: np.ndarray        # [G, E]
+    # aggregation weights per (local worker, model): float (batch counts or
+    # sample counts, per-algorithm — reference passes them as num_samples)
+    sample_num: np.ndarray      # [nW_local, K]
+    x_mask: Optional[torch.Tensor] = None   # [G, D] per-pair input mask (KUE)
+
+
+class TaskList:
+    """Builder for batched evaluation sweeps."""
+
+    def __init__(self):
+        self.task_row: List[int] = []
+        self.task_id: List[int] = []
+        self.off: List[int] = []
+        self.ln: List[int] = []
+        self.n_tasks = 0
+
+    def new_task(self) -> int:
+        self.n_tasks += 1
+        return self.n_tasks - 1
+
+    def add_windows(self, task_id: int, row: int,
+                    windows: Sequence[Tuple[int, int]]) -> None:
+        for o, l in windows:
+            if l <= 0:
+                continue
+            self.task_row.append(row)
+            self.task_id.append(task_id)
+            self.off.append(o)
+            self.ln.append(l)
+
+
+class FLJob:
+    def __init__(self, cfg: Config, comm: Communicator,
+                 logger: Optional[MetricLogger] = None):
+        self.cfg = cfg
+        self.comm = comm
+        self.device = comm.device
+        self.logger = logger or MetricLogger(cfg.log_dir,
+                                             enabled=comm.is_root)
+        self.backend = ops.backend_for(self.device, cfg.use_hip_kernels)
+
+        # seeds: np partition/batch RNG + torch init seed keyed on dummy_arg
+        # (reference main_fedavg.py:292-298)
+        np.random.seed(cfg.dummy_arg)
+        torch.manual_seed(cfg.dummy_arg)
+        zoo.set_torch_seed(cfg.dummy_arg)
+
+        self.curr_iter = cfg.curr_train_iteration
+        self.dataset = DriftDataset(cfg.data_dir, cfg.dataset,
+                                    cfg.client_num_in_total)
+        self.spec = packed.spec_for(cfg.model, self.dataset.feature_num,
+                                    self.dataset.class_num)
+        self.packer = packed.PackedMLP(self.spec)
+
+        from . import algorithms
+        self.algo = algorithms.make(cfg)
+
+        # data: all_data history + the algorithm's per-model retrain views
+        self.cdata = ClientData(self.dataset.store, self.curr_iter,
+                                cfg.batch_size,
+                                seed=cfg.dummy_arg * 1000003 + self.curr_iter)
+        self.views: List[RetrainView] = self.algo.build_views(self)
+        self.n_models = len(self.views)
+
+        # device arena
+        self.arena = DeviceArena(self.dataset.feature_num, self.device)
+        C, T = cfg.client_num_in_total, self.curr_iter + 1
+        self.all_ref: List[List[SegRef]] = [
+            [self.arena.add(self.cdata.all_data[c][t]) for t in range(T)]
+            for c in range(C)]
+        self.view_train_ref: List[Dict[int, SegRef]] = [
+            {c: self.arena.add(v.train[c]) for c in v.train}
+            for v in self.views]
+        # test set is iteration t+1 for every view — upload once (accuracy
+        # sums are shuffle-invariant, so one view's shuffle suffices)
+        self.test_ref: Dict[int, SegRef] = {
+            c: self.arena.add(self.views[0].test[c])
+            for c in self.views[0].test}
+        self.algo.add_extra_segments(self)
+        self.arena.freeze()
+
+        # model bank: K global rows + canonical init row
+        P = self.spec.n_params
+        proto = zoo.create_model(cfg.model, self.dataset.class_num,
+                                 self.dataset.feature_num)
+        self.init_flat = self.packer.flatten(proto.state_dict()).to(self.device)
+        self.global_params = self.init_flat.unsqueeze(0).repeat(
+            self.n_models, 1).contiguous()
+        self.algo.load_checkpoint(self)
+
+        # replicas + optimizer state for owned worker slots
+        self.n_workers = cfg.client_num_per_round
+        self.owned_workers = comm.owned_workers(self.n_workers)
+        nW = len(self.owned_workers)
+        self.replicas = torch.zeros(nW * self.n_models, P, device=self.device)
+        self.opt = ops.mlp_torch.make_opt_state(
+            cfg.client_optimizer, nW * self.n_models, P, cfg.lr, cfg.wd,
+            self.device)
+
+        # per-rank batch-pick RNG (statistical parity; reference draws on
+        # each worker process's own global np RNG)
+        self.pick_rng = np.random.default_rng(
+            (cfg.dummy_arg * 7919 + self.curr_iter) * 1009 + comm.rank)
+
+        self.algo.init_iteration(self)
+
+    # ------------------------------------------------------------------
+    # helpers
+    # ------------------------------------------------------------------
+    def row(self, wi: int, m: int) -> int:
+        """Replica row for owned-worker index wi, model m."""
+        return wi * self.n_models + m
+
+    def sync_replicas(self) -> None:
+        nW = len(self.owned_workers)
+        if nW:
+            self.replicas.copy_(
+                self.global_params.unsqueeze(0).expand(nW, -1, -1)
+                .reshape(-1, self.spec.n_params))
+
+    def run_eval(self, params: torch.Tensor, tl: TaskList,
+                 want_mse: bool = False):
+        dev = self.device
+        tr = torch.as_tensor(tl.task_row, dtype=torch.int64, device=dev)
+        ti = torch.as_tensor(tl.task_id, dtype=torch.int64, device=dev)
+        wo = torch.as_tensor(tl.off, dtype=torch.int64, device=dev)
+        wl = torch.as_tensor(tl.ln, dtype=torch.int64, device=dev)
+        c, t, l, m = self.backend.eval_tasks(
+            self.spec, params, self.arena.x, self.arena.y,
+            tr, ti, wo, wl, tl.n_tasks, want_mse=want_mse)
+        out = [c.cpu().numpy(), t.cpu().numpy(), l.cpu().numpy()]
+        out.append(m.cpu().numpy() if m is not None else None)
+        return out
+
+    def train(self, plan: TrainPlan) -> None:
+        if plan.rows.size == 0:
+            return
+        dev = self.device
+        self.backend.train_fused(
+            self.spec, self.replicas,
+            torch.as_tensor(plan.rows, dtype=torch.int64, device=dev),
+            self.arena.x, self.arena.y,
+            torch.as_tensor(plan.step_off, dtype=torch.int64, device=dev),
+            torch.as_tensor(plan.step_len, dtype=torch.int64, device=dev),
+            self.opt, x_mask=plan.x_mask)
+
+    def aggregate(self, plan: TrainPlan,
+                  model_mask: Optional[np.ndarray] = None) -> np.ndarray:
+        """Per-model sample-weighted averaging across all workers.
+
+        model_mask[m]=False skips the model entirely (its global params are
+        left untouched), matching the softcluster skip rule
+        (FedAvgEnsAggregatorSoftCluster.py:151-153). Returns total weights
+        per model. Models with zero total weight are skipped
+        (:167-169)."""
+        K, P = self.n_models, self.spec.n_params
+        partial = torch.zeros(K, P + 1, device=self.device)
+        nW = len(self.owned_workers)
+        if nW:
+            w = torch.as_tensor(plan.sample_num, dtype=torch.float32,
+                                device=self.device)          # [nW, K]
+            reps = self.replicas.reshape(nW, K, P)
+            partial[:, :P] = torch.einsum("wk,wkp->kp", w, reps)
+            partial[:, P] = w.sum(dim=0)
+        self.comm.all_reduce_(partial)
+        totals = partial[:, P].cpu().numpy()
+        for m in range(K):
+            if model_mask is not None and not model_mask[m]:
+                continue
+            if totals[m] > 0:
+                self.global_params[m] = partial[m, :P] / partial[m, P]
+        return totals
+
+    def client_sampling(self, round_idx: int) -> np.ndarray:
+        """Reference client_sampling (FedAvgEnsAggregatorSoftCluster.py:197-204):
+        identity when every client participates, else a per-round seeded
+        draw without replacement."""
+        cfg = self.cfg
+        if cfg.client_num_in_total == cfg.client_num_per_round:
+            return np.arange(cfg.client_num_in_total)
+        rs = np.random.RandomState(round_idx)
+        return rs.choice(range(cfg.client_num_in_total),
+                         min(cfg.client_num_per_round,
+                             cfg.client_num_in_total), replace=False)
+
+    # ------------------------------------------------------------------
+    # model-bank operations (EngineHooks for the drift state machines)
+    # ------------------------------------------------------------------
+    def merge_models(self, base: int, second: int, w1: float, w2: float):
+        self.global_params[base].mul_(w1).add_(self.global_params[second],
+                                               alpha=w2)
+
+    def reinit_model(self, m: int) -> None:
+        self.global_params[m] = self.init_flat
+
+    def copy_model(self, dst: int, src: int) -> None:
+        self.global_params[dst] = self.global_params[src]
+
+    def randomize_models_unseeded(self) -> None:
+        """reference 'hard' iter-0 quirk: reset_parameters() WITHOUT
+        re-seeding, so the K models diverge (needed for IFCA;
+        FedAvgEnsAggregatorSoftCluster.py:64-69)."""
+        proto = zoo.create_model(self.cfg.model, self.dataset.class_num,
+                                 self.dataset.feature_num)
+        for m in range(self.n_models):
+            for layer in proto.children():
+                if hasattr(layer, "reset_parameters"):
+                    layer.reset_parameters()
+            self.global_params[m] = self.packer.flatten(
+                proto.state_dict()).to(self.device)
+
+    def train_acc_matrix_rows(self, model_rows: List[int]) -> np.ndarray:
+        """[len(rows), C] acc of each global model row on each client's
+        current-iteration data; sharded by client + allreduced."""
+        C = self.cfg.client_num_in_total
+        tl = TaskList()
+        ids = {}
+        for i, m in enumerate(model_rows):
+            for c in range(C):
+                tid = tl.new_task()
+                ids[(i, c)] = tid
+                if self.comm.owns_client(c):
+                    tl.add_windows(tid, m,
+                                   self.all_ref[c][self.curr_iter].windows)
+        correct, total, _, _ = self.run_eval(self.global_params, tl)
+        cv = torch.zeros(2, tl.n_tasks, dtype=torch.float64)
+        cv[0] = torch.from_numpy(correct)
+        cv[1] = torch.from_numpy(total)
+        cv = cv.to(self.device)
+        self.comm.all_reduce_(cv)
+        cv = cv.cpu().numpy()
+        acc = np.zeros((len(model_rows), C))
+        for (i, c), tid in ids.items():
+            if cv[1][tid] != 0:
+                acc[i][c] = cv[0][tid] / cv[1][tid]
+        return acc
+
+    def pooled_cluster_windows(self, weights: Dict[int, np.ndarray], m: int,
+                               curr_iter: int) -> list:
+        """Reference pooling order (FedAvgEnsDataLoader.py:901-910):
+        client-major, then time."""
+        out = []
+        for c in range(self.cfg.client_num_in_total):
+            for t in range(curr_iter + 1):
+                if t in weights and weights[t][m][c] == 1:
+                    out.extend(self.all_ref[c][t].windows)
+        return out
+
+    def cluster_pair_acc_windows(self, model_rows: List[int],
+                                 cluster_windows: Dict[int, list],
+                                 cap: int = 21) -> np.ndarray:
+        """[K, K] acc of model i on cluster j's pooled (already shuffled)
+        windows, capped at `cap` batches per pair (reference :923-931 with
+        _infer_subset's 21-batch cap, :1111-1138). Evaluated identically on
+        every rank (inputs are lockstep-identical)."""
+        k = len(model_rows)
+        tl = TaskList()
+        ids = {}
+        for i, mrow in enumerate(model_rows):
+            for j, mj in enumerate(model_rows):
+                tid = tl.new_task()
+                ids[(i, j)] = tid
+                tl.add_windows(tid, mrow, cluster_windows[mj][:cap])
+        correct, total, _, _ = self.run_eval(self.global_params, tl)
+        acc = np.zeros((k, k))
+        for (i, j), tid in ids.items():
+            if total[tid] != 0:
+                acc[i][j] = correct[tid] / total[tid]
+        return acc
+
+    # per-client eval of chosen model rows on train (curr iter) + test (t+1)
+    def client_eval(self, model_idx_per_client: np.ndarray):
+        """Returns per-client (train_correct, train_total, train_loss,
+        test_correct, test_total, test_loss) arrays, allreduced.
+
+        ci==1 keeps only client 0 (reference CI shortcut,
+        FedAvgEnsAggregatorSoftCluster.py:259-264)."""
+        C = self.cfg.client_num_in_total
+        clients = range(C) if self.cfg.ci != 1 else range(1)
+        tl = TaskList()
+        for c in clients:
+            tid_tr = tl.new_task()
+            tid_te = tl.new_task()
+            if self.comm.owns_client(c):
+                m = int(model_idx_per_client[c])
+                tl.add_windows(tid_tr, m,
+                               self.all_ref[c][self.curr_iter].windows)
+                if c in self.test_ref:
+                    tl.add_windows(tid_te, m, self.test_ref[c].windows)
+        correct, total, loss, _ = self.run_eval(self.global_params, tl)
+        buf = torch.from_numpy(np.stack([correct, total, loss])).to(self.device)
+        self.comm.all_reduce_(buf)
+        correct, total, loss = buf.cpu().numpy()
+        n = len(list(clients))
+        tr = (correct[0::2][:n], total[0::2][:n], loss[0::2][:n])
+        te = (correct[1::2][:n], total[1::2][:n], loss[1::2][:n])
+        return tr, te
+
+    def log_round_stats(self, round_idx: int, tr, te) -> None:
+        (trc, trt, trl) = tr
+        (tec, tet, tel) = te
+        if self.cfg.report_client == 1 and self.comm.is_root:
+            for c in range(len(trc)):
+                self.logger.log(
+                    {f"Train/Acc-CL-{c}": trc[c] / trt[c] if trt[c] else -1},
+                    round_idx)
+                self.logger.log(
+                    {f"Test/Acc-CL-{c}": tec[c] / tet[c] if tet[c] else -1},
+                    round_idx)
+        train_acc = trc.sum() / trt.sum() if trt.sum() else 0.0
+        train_loss = trl.sum() / trt.sum() if trt.sum() else 0.0
+        test_acc = tec.sum() / tet.sum() if tet.sum() else 0.0
+        test_loss = tel.sum() / tet.sum() if tet.sum() else 0.0
+        if self.comm.is_root:
+            self.logger.log({"Train/Acc": train_acc, "Train/Loss": train_loss},
+                            round_idx)
+            self.logger.log({"Test/Acc": test_acc, "Test/Loss": test_loss},
+                            round_idx)
+
+    # ------------------------------------------------------------------
+    # checkpointing (reference-compatible layout)
+    # ------------------------------------------------------------------
+    def ckpt_path(self, name: str) -> str:
+        return os.path.join(self.cfg.log_dir, name)
+
+    def save_model_params(self) -> None:
+        if not self.comm.is_root:
+            return
+        sds = {m: self.packer.unflatten(self.global_params[m])
+               for m in range(self.n_models)}
+        torch.save(sds, self.ckpt_path("model_params.pt"))
+
+    def load_model_params_general(self) -> None:
+        """General reload rule: load models in saved order
+        (main_fedavg.py:354-357)."""
+        mp = torch.load(self.ckpt_path("model_params.pt"))
+        for m_idx, sd in mp.items():
+            if m_idx < self.n_models:
+                self.global_params[m_idx] = self.packer.flatten(sd).to(
+                    self.device)
+
+    def save_state_pickle(self, name: str, state) -> None:
+        if self.comm.is_root:
+            with open(self.ckpt_path(name), "wb") as f:
+                pickle.dump(state, f)
+        self.comm.barrier()
+
+    def load_state_pickle(self, name: str):
+        with open(self.ckpt_path(name), "rb") as f:
+            return pickle.load(f)
+
+    # ------------------------------------------------------------------
+    # the round loop
+    # ------------------------------------------------------------------
+    def run(self) -> None:
+        client_idx = self.client_sampling(0)
+        for r in range(self.cfg.comm_round):
+            self.sync_replicas()
+            plan = self.algo.plan(self, r, client_idx)
+            self.train(plan)
+            self.algo.aggregate(self, r, plan, client_idx)
+            self.algo.post_aggregate(self, r)
+            self.algo.test(self, r)
+            client_idx = self.client_sampling(r + 1)
+        self.save_model_params()
+        self.algo.finalize(self)
+        self.comm.barrier()
+        self.logger.flush()
+
+
+def run_iteration(cfg: Config, comm: Optional[Communicator] = None,
+                  logger: Optional[MetricLogger] = None) -> FLJob:
+    comm = comm or Communicator(backend=cfg.backend if cfg.backend != "auto"
+                                else "auto")
+    job = FLJob(cfg, comm, logger)
+    job.run()
+    return job

@@ -1,0 +1,100 @@
+"""Model zoo for the drift path.
+
+Matches the reference architectures exactly (fedml_api/model/...):
+  * LogisticRegression (linear/lr.py:4-11): linear + in-graph sigmoid, CE on
+    the sigmoid output — a reference quirk kept for metric parity.
+  * FeedForwardNN (fnn/fnn.py:4-15): D -> 2D -> O with ReLU (SEA: 3->6->2).
+  * CNN_DropOut (cv/cnn.py:71-135): 784 -> 28x28, conv3x3(32) -> conv3x3(64)
+    -> maxpool -> dropout(.25) -> fc128 -> dropout(.5) -> fc10 with an
+    IN-GRAPH nn.Softmax before CrossEntropyLoss (double-softmax quirk,
+    cv/cnn.py:134) — preserved, not "fixed".
+  * resnet18 (torchvision in the reference, main_fedavg.py:219-222):
+    torchvision is unavailable offline, so models/resnet.py provides an
+    equivalent random-init ResNet-18.
+
+reinitialize() re-seeds torch with the module-level torch_seed and resets
+every child layer, so all K ensemble slots start identical
+(fedml_api/model/utils.py:7-24).
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import nn
+
+torch_seed = 42
+
+
+def set_torch_seed(seed: int) -> None:
+    global torch_seed
+    torch_seed = seed
+
+
+class LogisticRegression(nn.Module):
+    def __init__(self, input_dim: int, output_dim: int):
+        super().__init__()
+        self.linear = nn.Linear(input_dim, output_dim)
+
+    def forward(self, x):
+        return torch.sigmoid(self.linear(x))
+
+
+class FeedForwardNN(nn.Module):
+    def __init__(self, input_dim: int, output_dim: int, hidden_dim: int):
+        super().__init__()
+        self.fc1 = nn.Linear(input_dim, hidden_dim)
+        self.relu = nn.ReLU()
+        self.fc2 = nn.Linear(hidden_dim, output_dim)
+
+    def forward(self, x):
+        return self.fc2(self.relu(self.fc1(x)))
+
+
+class CNN_DropOut(nn.Module):
+    def __init__(self, only_digits: bool = True):
+        super().__init__()
+        self.conv2d_1 = nn.Conv2d(1, 32, kernel_size=3)
+        self.max_pooling = nn.MaxPool2d(2, stride=2)
+        self.conv2d_2 = nn.Conv2d(32, 64, kernel_size=3)
+        self.dropout_1 = nn.Dropout(0.25)
+        self.flatten = nn.Flatten()
+        self.linear_1 = nn.Linear(9216, 128)
+        self.dropout_2 = nn.Dropout(0.5)
+        self.linear_2 = nn.Linear(128, 10 if only_digits else 62)
+        self.relu = nn.ReLU()
+        self.softmax = nn.Softmax(dim=1)
+
+    def forward(self, x):
+        x = torch.unsqueeze(x.reshape(-1, 28, 28), 1)
+        x = self.relu(self.conv2d_1(x))
+        x = self.relu(self.conv2d_2(x))
+        x = self.max_pooling(x)
+        x = self.dropout_1(x)
+        x = self.flatten(x)
+        x = self.relu(self.linear_1(x))
+        x = self.dropout_2(x)
+        x = self.linear_2(x)
+        return self.softmax(x)
+
+
+def reinitialize(model: nn.Module) -> None:
+    torch.manual_seed(torch_seed)
+    for layer in model.children():
+        if hasattr(layer, "reset_parameters"):
+            layer.reset_parameters()
+
+
+def create_model(model_name: str, output_dim: int, feature_dim: int) -> nn.Module:
+    if model_name == "lr":
+        model = LogisticRegression(feature_dim, output_dim)
+    elif model_name == "fnn":
+        model = FeedForwardNN(feature_dim, output_dim, feature_dim * 2)
+    elif model_name == "cnn":
+        model = CNN_DropOut()
+    elif model_name == "resnet":
+        from .resnet import resnet18
+        model = resnet18(num_classes=output_dim if output_dim > 2 else 1000)
+    else:
+        raise NameError(model_name)
+    reinitialize(model)
+    return model
