@@ -1,0 +1,158 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: FedDrift-style FL rounds/sec on SEA-4 FNN.
+
+One timed "step" = one full FL round of the softcluster (FedDrift) engine in
+steady state: local training of every (client, active-model) pair (E=5 Adam
+steps on batch-500 minibatches), fused weighted-parameter aggregation with an
+RCCL all_reduce, and the full prequential evaluation of every client on its
+current-iteration and next-iteration data — the same per-round work the
+reference performs (FedAvgEnsServerManager round handler + trainers +
+test_on_all_clients).
+
+Scaling is WEAK: 10 simulated clients per GPU (10 at N=1, 80 at N=8), one
+model per client cluster capped at K=10 ensemble slots, matching
+BASELINE.json's SEA-4 FNN FedDrift configuration shape. Data is synthetic
+SEA-4 (in-memory, random-init weights; there is no network access).
+
+Contract (driver): prints ONE JSON line from rank 0; the timed region is
+bracketed by dist barrier + torch.cuda.synchronize on both sides; the
+reported time is the MAX over ranks.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from feddrift_amd.comm import Communicator
+from feddrift_amd.config import Config
+from feddrift_amd.data.generators import sample_sea
+from feddrift_amd.data.loader import DriftDataset
+from feddrift_amd.engine.fljob import FLJob
+from feddrift_amd.eval.metrics import MetricLogger
+
+BASE_CLIENTS_PER_GPU = 10
+N_MODELS = 10
+CURR_ITER = 3           # steady state: 4 iterations of history resident
+SEQ = 500               # samples per (client, iteration); batch size 500
+EPOCHS = 5
+ROUNDS_DEFAULT = 60
+WARMUP_DEFAULT = 10
+
+
+def build_dataset(n_clients: int, seed: int) -> DriftDataset:
+    ds = DriftDataset(data_dir="/nonexistent", dataset="sea",
+                      num_client=n_clients)
+    rng = np.random.default_rng(seed)
+    for c in range(n_clients):
+        for t in range(CURR_ITER + 2):     # +1 for the prequential test set
+            arr = sample_sea(SEQ, (c + t) % 4, rng)
+            ds.store.put(c, t, arr[:, :3], arr[:, 3])
+    return ds
+
+
+def one_round(job: FLJob, r: int, client_idx) -> None:
+    job.sync_replicas()
+    plan = job.algo.plan(job, r, client_idx)
+    job.train(plan)
+    job.algo.aggregate(job, r, plan, client_idx)
+    job.algo.post_aggregate(job, r)
+    job.algo.test(job, r)
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=ROUNDS_DEFAULT)
+    p.add_argument("--warmup", type=int, default=WARMUP_DEFAULT)
+    a = p.parse_args()
+
+    comm = Communicator()
+    world = comm.world_size
+    n_gpus = max(a.gpus, world)
+    n_clients = BASE_CLIENTS_PER_GPU * max(world, 1)
+
+    cfg = Config(model="fnn", dataset="sea", data_dir="/nonexistent",
+                 client_num_in_total=n_clients,
+                 client_num_per_round=n_clients, batch_size=SEQ,
+                 client_optimizer="adam", lr=0.01, epochs=EPOCHS,
+                 comm_round=10 ** 9, total_train_iteration=CURR_ITER + 1,
+                 curr_train_iteration=CURR_ITER, concept_num=N_MODELS,
+                 concept_drift_algo="softcluster",
+                 concept_drift_algo_arg="H_A_F_1_06_0",
+                 change_points="A", dummy_arg=0, report_client=0,
+                 bench_mode=1)
+
+    dataset = build_dataset(n_clients, seed=1234)
+    logger = MetricLogger(enabled=comm.is_root, to_file=False)
+    job = FLJob(cfg, comm, logger, dataset=dataset)
+    client_idx = np.arange(n_clients)
+
+    dev = job.device
+    is_cuda = dev.type == "cuda"
+
+    for r in range(a.warmup):
+        one_round(job, r, client_idx)
+
+    comm.barrier()
+    if is_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for r in range(a.warmup, a.warmup + a.steps):
+        one_round(job, r, client_idx)
+    comm.barrier()
+    if is_cuda:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64, device=dev
+                           if is_cuda else "cpu")
+    if comm.distributed:
+        import torch.distributed as dist
+        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+    elapsed = float(elapsed.item())
+
+    rounds_per_sec = a.steps / elapsed
+    ms_per_step = elapsed / a.steps * 1e3
+
+    if comm.is_root:
+        acc = logger.mean("Test/Acc")
+        print(json.dumps({
+            "metric": "fl_rounds_per_sec",
+            "value": rounds_per_sec,
+            "unit": "rounds/s",
+            "n_gpus": world,
+            "steps": a.steps,
+            "warmup": a.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "fnn(3-6-2)",
+                "algo": "softcluster H_A_F_1_06_0 (FedDrift)",
+                "clients_per_gpu": BASE_CLIENTS_PER_GPU,
+                "clients_total": n_clients,
+                "ensemble_models": N_MODELS,
+                "epochs_per_round": EPOCHS,
+                "global_batch": SEQ * n_clients,
+                "seq_len": SEQ,
+                "optimizer": "adam(amsgrad,wd=1e-3)",
+                "history_iterations": CURR_ITER + 1,
+                "prequential_eval_per_round": True,
+                "parallelism": f"client-sharded dp{world}",
+                "steady_state_test_acc": acc,
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

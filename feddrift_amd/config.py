@@ -58,6 +58,7 @@ class Config:
     use_hip_kernels: str = "auto"        # auto|always|never
     wandb: int = 0                       # optional wandb mirror of metric logs
     log_dir: str = "."                   # where checkpoints/state files go
+    bench_mode: int = 0                  # 1: synthetic steady-state (bench.py)
 
     def __post_init__(self):
         self.dataset_norm = "MNIST" if self.dataset.lower() == "mnist" else self.dataset

@@ -77,6 +77,10 @@ class RawStore:
             self._cache[key] = _read_client_iter_csv(path)
         return self._cache[key]
 
+    def put(self, c: int, t: int, x: np.ndarray, y: np.ndarray) -> None:
+        """In-memory injection (synthetic benchmarks / tests — no CSV IO)."""
+        self._cache[(c, t)] = (x.astype(np.float32), y.astype(np.int64))
+
 
 def resolve_retrain_rows(store: RawStore, c: int, curr_iter: int,
                          method: str, rng: np.random.Generator

@@ -210,6 +210,19 @@ class SoftClusterAlgo(AlgoBase):
     def init_iteration(self, job: FLJob) -> None:
         cfg = job.cfg
         hooks = Hooks(job)
+        if cfg.bench_mode:
+            # synthetic steady-state: client c trains/tests model c % K at
+            # every iteration — K active clusters, full per-round load,
+            # no iteration-start clustering (bench.py)
+            st = SoftClusterState(cfg.client_num_in_total, cfg.concept_num,
+                                  "hard", seed=cfg.dummy_arg)
+            for t in range(job.curr_iter + 1):
+                w = np.zeros((cfg.concept_num, cfg.client_num_in_total))
+                for c in range(cfg.client_num_in_total):
+                    w[c % cfg.concept_num][c] = 1.0
+                st.train_data_weights[t] = w
+            self.state = st
+            return
         if job.curr_iter == 0 or not os.path.exists(self._state_path(job)):
             p = parse_softcluster_arg(cfg.concept_drift_algo_arg, cfg.dataset_norm)
             st = SoftClusterState(

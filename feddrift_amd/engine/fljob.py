@@ -82,7 +82,8 @@ class TaskList:
 
 class FLJob:
     def __init__(self, cfg: Config, comm: Communicator,
-                 logger: Optional[MetricLogger] = None):
+                 logger: Optional[MetricLogger] = None,
+                 dataset: Optional[DriftDataset] = None):
         self.cfg = cfg
         self.comm = comm
         self.device = comm.device
@@ -97,8 +98,8 @@ class FLJob:
         zoo.set_torch_seed(cfg.dummy_arg)
 
         self.curr_iter = cfg.curr_train_iteration
-        self.dataset = DriftDataset(cfg.data_dir, cfg.dataset,
-                                    cfg.client_num_in_total)
+        self.dataset = dataset or DriftDataset(cfg.data_dir, cfg.dataset,
+                                               cfg.client_num_in_total)
         self.spec = packed.spec_for(cfg.model, self.dataset.feature_num,
                                     self.dataset.class_num)
         self.packer = packed.PackedMLP(self.spec)
@@ -391,6 +392,8 @@ class FLJob:
                     self.device)
 
     def save_state_pickle(self, name: str, state) -> None:
+        if self.cfg.bench_mode:
+            return
         if self.comm.is_root:
             with open(self.ckpt_path(name), "wb") as f:
                 pickle.dump(state, f)

@@ -19,11 +19,13 @@ from typing import Dict, List, Optional
 
 class MetricLogger:
     def __init__(self, log_dir: str = ".", enabled: bool = True,
-                 use_wandb: bool = False, run_name: str = "run"):
+                 use_wandb: bool = False, run_name: str = "run",
+                 to_file: bool = True):
         self.enabled = enabled
         self.records: List[Dict] = []
         self.summary: Dict = {}
-        self._path = os.path.join(log_dir, "metrics.jsonl") if enabled else None
+        self._path = os.path.join(log_dir, "metrics.jsonl") \
+            if (enabled and to_file) else None
         self._fh = None
         self._wandb = None
         if enabled and use_wandb:

@@ -1,0 +1,452 @@
+// feddrift MI355X (gfx950 / CDNA4) kernels
+//
+// Fused batched local training + evaluation sweeps for the MLP model family
+// (FeedForwardNN / LogisticRegression of the drift path).
+//
+// Design (not a port: the reference, microsoft/FedDrift, runs these loops as
+// per-model Python/torch eager code with CPU<->GPU model movement every
+// round — fedml_api/distributed/fedavg_ens/FedAvgEnsTrainer.py:47-95):
+//
+//  * mlp_train_kernel: ONE launch trains every (client, model) pair of the
+//    round. One workgroup per pair; the pair's weights and the gradient
+//    accumulator live in LDS for the entire E-step optimizer loop; each
+//    step stages its minibatch chunk through LDS, computes forward +
+//    backward with per-thread sample parallelism, reduces gradients with a
+//    per-thread ownership partition (no atomics), and applies the
+//    SGD/Adam(amsgrad, wd) update in-place. Adam state stays in HBM.
+//  * mlp_eval_kernel: ONE launch per accuracy/loss sweep (the K-models x
+//    C-clients matrices, prequential testing). One workgroup per data
+//    window; per-thread sample loop + LDS tree reduction; one global
+//    float64 atomicAdd per output slot.
+//
+// fp32 everywhere (the reference trains fp32; dtype parity is required for
+// the accuracy metric). Wavefront = 64; block = 256 threads.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <cmath>
+#include <vector>
+
+#define THREADS 256
+#define LDS_BUDGET_FLOATS (150 * 1024 / 4)
+
+#define KIND_FNN 0
+#define KIND_LR 1
+
+#define OPT_SGD 0
+#define OPT_ADAM 1
+
+// ---------------------------------------------------------------------------
+// training kernel
+// ---------------------------------------------------------------------------
+
+struct TrainArgs {
+  float* __restrict__ params;        // [R, P]
+  const int64_t* __restrict__ rows;  // [G]
+  const float* __restrict__ x;       // [N, D]
+  const int64_t* __restrict__ y;     // [N]
+  const int64_t* __restrict__ step_off;  // [G, E]
+  const int64_t* __restrict__ step_len;  // [G, E]
+  const float* __restrict__ x_mask;  // [G, D] or nullptr
+  float* __restrict__ m;             // [R, P] (adam) or nullptr
+  float* __restrict__ v;
+  float* __restrict__ vmax;
+  int* __restrict__ t;               // [R]
+  const float* __restrict__ lr;      // [R]
+  float wd;
+  int E, D, H, O, P, kind, opt, BC;
+};
+
+__device__ __forceinline__ float block_reduce_sum(float val, float* scratch) {
+  // wave shuffle reduce then cross-wave via LDS (4 waves / 256 threads)
+  for (int off = 32; off > 0; off >>= 1)
+    val += __shfl_down(val, off, 64);
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  if (lane == 0) scratch[wave] = val;
+  __syncthreads();
+  if (wave == 0) {
+    val = (lane < (blockDim.x >> 6)) ? scratch[lane] : 0.f;
+    for (int off = 2; off > 0; off >>= 1)
+      val += __shfl_down(val, off, 64);
+  }
+  __syncthreads();
+  return val;  // valid in thread 0
+}
+
+extern "C" __global__ __launch_bounds__(THREADS)
+void mlp_train_kernel(TrainArgs a) {
+  const int g = blockIdx.x;
+  const int64_t row = a.rows[g];
+  const int tid = threadIdx.x;
+
+  extern __shared__ __attribute__((aligned(16))) float lds[];
+  float* w = lds;              // [P]
+  float* grad = w + a.P;       // [P]
+  float* xb = grad + a.P;      // [BC, D]
+  float* act = xb + a.BC * a.D;      // [BC, H] (fnn activations)
+  float* dza = act + (a.kind == KIND_FNN ? a.BC * a.H : 0);  // [BC, H] dz1
+  float* dzo = dza + (a.kind == KIND_FNN ? a.BC * a.H : 0);  // [BC, O]
+
+  // stage weights
+  for (int p = tid; p < a.P; p += THREADS) w[p] = a.params[row * a.P + p];
+  __syncthreads();
+
+  const int HD = a.H * a.D;
+  const int OH = a.O * a.H;
+  const int OD = a.O * a.D;
+
+  for (int e = 0; e < a.E; ++e) {
+    const int64_t off = a.step_off[(int64_t)g * a.E + e];
+    const int n = (int)a.step_len[(int64_t)g * a.E + e];
+    if (n == 0) continue;  // reference skips the step entirely
+    const float inv_n = 1.0f / (float)n;
+
+    // zero gradient accumulator
+    for (int p = tid; p < a.P; p += THREADS) grad[p] = 0.f;
+    __syncthreads();
+
+    for (int c0 = 0; c0 < n; c0 += a.BC) {
+      const int bc = min(a.BC, n - c0);
+
+      // stage input chunk (optionally feature-masked: KUE)
+      for (int q = tid; q < bc * a.D; q += THREADS) {
+        const int i = q / a.D;
+        const int d = q % a.D;
+        float xv = a.x[(off + c0 + i) * a.D + d];
+        if (a.x_mask) xv *= a.x_mask[(int64_t)g * a.D + d];
+        xb[q] = xv;
+      }
+      __syncthreads();
+
+      // forward + output gradient per sample
+      for (int i = tid; i < bc; i += THREADS) {
+        const int64_t yi = a.y[off + c0 + i];
+        if (a.kind == KIND_FNN) {
+          // z1 = W1 x + b1 ; act = relu(z1)
+          for (int h = 0; h < a.H; ++h) {
+            float z = w[HD + h];
+            for (int d = 0; d < a.D; ++d) z += w[h * a.D + d] * xb[i * a.D + d];
+            act[i * a.H + h] = z > 0.f ? z : 0.f;
+          }
+          // z2 = W2 a + b2 -> dz2 = (softmax(z2) - onehot) / n
+          float zmax = -1e30f;
+          for (int o = 0; o < a.O; ++o) {
+            float z = w[HD + a.H + OH + o];
+            for (int h = 0; h < a.H; ++h)
+              z += w[HD + a.H + o * a.H + h] * act[i * a.H + h];
+            dzo[i * a.O + o] = z;
+            zmax = fmaxf(zmax, z);
+          }
+          float zsum = 0.f;
+          for (int o = 0; o < a.O; ++o) {
+            const float ez = __expf(dzo[i * a.O + o] - zmax);
+            dzo[i * a.O + o] = ez;
+            zsum += ez;
+          }
+          for (int o = 0; o < a.O; ++o) {
+            float sm = dzo[i * a.O + o] / zsum;
+            if (o == (int)yi) sm -= 1.f;
+            dzo[i * a.O + o] = sm * inv_n;
+          }
+          // dz1 = (W2^T dz2) * relu'(z1)
+          for (int h = 0; h < a.H; ++h) {
+            float s = 0.f;
+            for (int o = 0; o < a.O; ++o)
+              s += dzo[i * a.O + o] * w[HD + a.H + o * a.H + h];
+            dza[i * a.H + h] = act[i * a.H + h] > 0.f ? s : 0.f;
+          }
+        } else {  // LR: out = sigmoid(Wx+b); CE applied to the sigmoid
+          float p_[64];  // O <= 64 for the LR path
+          float pmax = -1e30f;
+          for (int o = 0; o < a.O; ++o) {
+            float z = w[OD + o];
+            for (int d = 0; d < a.D; ++d) z += w[o * a.D + d] * xb[i * a.D + d];
+            const float pv = 1.f / (1.f + __expf(-z));
+            p_[o] = pv;
+            pmax = fmaxf(pmax, pv);
+          }
+          float psum = 0.f;
+          for (int o = 0; o < a.O; ++o) psum += __expf(p_[o] - pmax);
+          for (int o = 0; o < a.O; ++o) {
+            float sm = __expf(p_[o] - pmax) / psum;
+            if (o == (int)yi) sm -= 1.f;
+            // dz = dp * sigmoid' = dp * p (1-p)
+            dzo[i * a.O + o] = sm * inv_n * p_[o] * (1.f - p_[o]);
+          }
+        }
+      }
+      __syncthreads();
+
+      // gradient accumulation: thread-owned parameter partition
+      if (a.kind == KIND_FNN) {
+        for (int p = tid; p < a.P; p += THREADS) {
+          float acc = 0.f;
+          if (p < HD) {
+            const int h = p / a.D, d = p % a.D;
+            for (int i = 0; i < bc; ++i)
+              acc += dza[i * a.H + h] * xb[i * a.D + d];
+          } else if (p < HD + a.H) {
+            const int h = p - HD;
+            for (int i = 0; i < bc; ++i) acc += dza[i * a.H + h];
+          } else if (p < HD + a.H + OH) {
+            const int q = p - HD - a.H;
+            const int o = q / a.H, h = q % a.H;
+            for (int i = 0; i < bc; ++i)
+              acc += dzo[i * a.O + o] * act[i * a.H + h];
+          } else {
+            const int o = p - HD - a.H - OH;
+            for (int i = 0; i < bc; ++i) acc += dzo[i * a.O + o];
+          }
+          grad[p] += acc;
+        }
+      } else {
+        for (int p = tid; p < a.P; p += THREADS) {
+          float acc = 0.f;
+          if (p < OD) {
+            const int o = p / a.D, d = p % a.D;
+            for (int i = 0; i < bc; ++i)
+              acc += dzo[i * a.O + o] * xb[i * a.D + d];
+          } else {
+            const int o = p - OD;
+            for (int i = 0; i < bc; ++i) acc += dzo[i * a.O + o];
+          }
+          grad[p] += acc;
+        }
+      }
+      __syncthreads();
+    }
+
+    // optimizer update (matches torch.optim exactly; see ops/mlp_torch.py)
+    if (a.opt == OPT_SGD) {
+      const float lr_ = a.lr[row];
+      for (int p = tid; p < a.P; p += THREADS) w[p] -= lr_ * grad[p];
+    } else {
+      if (tid == 0) a.t[row] += 1;
+      __syncthreads();
+      const int tnew = a.t[row];
+      const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
+      const float bc1 = 1.f - powf(b1, (float)tnew);
+      const float bc2 = 1.f - powf(b2, (float)tnew);
+      const float lr_ = a.lr[row];
+      for (int p = tid; p < a.P; p += THREADS) {
+        const int64_t gp = row * a.P + p;
+        const float gr = grad[p] + a.wd * w[p];
+        const float mn = b1 * a.m[gp] + (1.f - b1) * gr;
+        const float vn = b2 * a.v[gp] + (1.f - b2) * gr * gr;
+        a.m[gp] = mn;
+        a.v[gp] = vn;
+        const float vm = fmaxf(a.vmax[gp], vn);
+        a.vmax[gp] = vm;
+        const float denom = sqrtf(vm / bc2) + eps;
+        w[p] -= lr_ * (mn / bc1) / denom;
+      }
+    }
+    __syncthreads();
+  }
+
+  // write back trained weights
+  for (int p = tid; p < a.P; p += THREADS) a.params[row * a.P + p] = w[p];
+}
+
+// ---------------------------------------------------------------------------
+// evaluation kernel
+// ---------------------------------------------------------------------------
+
+struct EvalArgs {
+  const float* __restrict__ params;     // [M, P]
+  const int64_t* __restrict__ task_row; // [W]
+  const int64_t* __restrict__ task_id;  // [W]
+  const int64_t* __restrict__ off;      // [W]
+  const int64_t* __restrict__ len;      // [W]
+  const float* __restrict__ x;          // [N, D]
+  const int64_t* __restrict__ y;        // [N]
+  const float* __restrict__ x_mask;     // [W, D] or nullptr
+  double* __restrict__ correct;         // [T]
+  double* __restrict__ total;           // [T]
+  double* __restrict__ loss;            // [T]
+  double* __restrict__ mse;             // [T] or nullptr
+  int D, H, O, P, kind;
+};
+
+extern "C" __global__ __launch_bounds__(THREADS)
+void mlp_eval_kernel(EvalArgs a) {
+  const int wdx = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int64_t row = a.task_row[wdx];
+  const int64_t tsk = a.task_id[wdx];
+  const int64_t off = a.off[wdx];
+  const int n = (int)a.len[wdx];
+
+  extern __shared__ __attribute__((aligned(16))) float lds[];
+  float* w = lds;              // [P] (P fits: MLP family)
+  float* red = w + a.P;        // [8] reduction scratch
+
+  for (int p = tid; p < a.P; p += THREADS) w[p] = a.params[row * a.P + p];
+  __syncthreads();
+
+  const int HD = a.H * a.D;
+  const int OH = a.O * a.H;
+  const int OD = a.O * a.D;
+
+  float c_acc = 0.f, l_acc = 0.f, e_acc = 0.f;
+  for (int i = tid; i < n; i += THREADS) {
+    const int yi = (int)a.y[off + i];
+    float logits[64];
+    if (a.kind == KIND_FNN) {
+      // activations computed per-h on the fly (no [H] storage needed);
+      // x is read from global — the window is L1/L2 resident
+      for (int o = 0; o < a.O; ++o) logits[o] = w[HD + a.H + OH + o];
+      for (int h = 0; h < a.H; ++h) {
+        float z = w[HD + h];
+        for (int d = 0; d < a.D; ++d) {
+          float xd = a.x[(off + i) * a.D + d];
+          if (a.x_mask) xd *= a.x_mask[(int64_t)wdx * a.D + d];
+          z += w[h * a.D + d] * xd;
+        }
+        if (z > 0.f)
+          for (int o = 0; o < a.O; ++o)
+            logits[o] += w[HD + a.H + o * a.H + h] * z;
+      }
+    } else {
+      for (int o = 0; o < a.O; ++o) {
+        float z = w[OD + o];
+        for (int d = 0; d < a.D; ++d) {
+          float xd = a.x[(off + i) * a.D + d];
+          if (a.x_mask) xd *= a.x_mask[(int64_t)wdx * a.D + d];
+          z += w[o * a.D + d] * xd;
+        }
+        logits[o] = 1.f / (1.f + __expf(-z));
+      }
+    }
+    // argmax + CE(logits, y) + optional AUE mse
+    int best = 0;
+    float zmax = logits[0];
+    for (int o = 1; o < a.O; ++o)
+      if (logits[o] > zmax) { zmax = logits[o]; best = o; }
+    float zsum = 0.f;
+    for (int o = 0; o < a.O; ++o) zsum += __expf(logits[o] - zmax);
+    const float lse = logf(zsum) + zmax;
+    c_acc += (best == yi) ? 1.f : 0.f;
+    l_acc += lse - logits[yi];
+    if (a.mse) {
+      const float ptrue = __expf(logits[yi] - lse);
+      e_acc += (1.f - ptrue) * (1.f - ptrue);
+    }
+  }
+
+  float cs = block_reduce_sum(c_acc, red);
+  float ls = block_reduce_sum(l_acc, red);
+  float es = a.mse ? block_reduce_sum(e_acc, red) : 0.f;
+  if (tid == 0) {
+    atomicAdd(&a.correct[tsk], (double)cs);
+    atomicAdd(&a.total[tsk], (double)n);
+    atomicAdd(&a.loss[tsk], (double)ls);
+    if (a.mse) atomicAdd(&a.mse[tsk], (double)es);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static void check_f32_2d(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.dtype() == torch::kFloat32 &&
+              t.is_contiguous(), name, " must be contiguous f32 CUDA");
+}
+
+void train_fused_hip(torch::Tensor params, torch::Tensor rows,
+                     torch::Tensor x, torch::Tensor y,
+                     torch::Tensor step_off, torch::Tensor step_len,
+                     int64_t D, int64_t H, int64_t O, int64_t kind,
+                     c10::optional<torch::Tensor> x_mask,
+                     c10::optional<torch::Tensor> m,
+                     c10::optional<torch::Tensor> v,
+                     c10::optional<torch::Tensor> vmax,
+                     c10::optional<torch::Tensor> t,
+                     torch::Tensor lr, double wd) {
+  const int G = rows.size(0);
+  if (G == 0) return;
+  check_f32_2d(params, "params");
+  const int P = params.size(1);
+  const int E = step_off.size(1);
+  const bool adam = m.has_value();
+
+  // chunk size: fit 2P + BC*(D + 2H + O | D + O) in the LDS budget
+  const int per_sample = (kind == KIND_FNN) ? (D + 2 * H + O) : (D + O);
+  int BC = (LDS_BUDGET_FLOATS - 2 * P) / per_sample;
+  TORCH_CHECK(BC >= 1, "model too large for LDS-resident training path");
+  BC = std::min<int>(BC, 512);
+
+  const size_t lds_bytes =
+      (size_t)(2 * P + (size_t)BC * per_sample) * sizeof(float);
+
+  TrainArgs args;
+  args.params = params.data_ptr<float>();
+  args.rows = rows.data_ptr<int64_t>();
+  args.x = x.data_ptr<float>();
+  args.y = y.data_ptr<int64_t>();
+  args.step_off = step_off.data_ptr<int64_t>();
+  args.step_len = step_len.data_ptr<int64_t>();
+  args.x_mask = x_mask.has_value() ? x_mask->data_ptr<float>() : nullptr;
+  args.m = adam ? m->data_ptr<float>() : nullptr;
+  args.v = adam ? v->data_ptr<float>() : nullptr;
+  args.vmax = adam ? vmax->data_ptr<float>() : nullptr;
+  args.t = adam ? t->data_ptr<int>() : nullptr;
+  args.lr = lr.data_ptr<float>();
+  args.wd = (float)wd;
+  args.E = E; args.D = (int)D; args.H = (int)H; args.O = (int)O;
+  args.P = P; args.kind = (int)kind; args.opt = adam ? OPT_ADAM : OPT_SGD;
+  args.BC = BC;
+
+  hipLaunchKernelGGL(mlp_train_kernel, dim3(G), dim3(THREADS), lds_bytes,
+                     c10::hip::getCurrentHIPStream(), args);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "mlp_train_kernel launch");
+}
+
+std::vector<torch::Tensor> eval_tasks_hip(
+    torch::Tensor params, torch::Tensor x, torch::Tensor y,
+    torch::Tensor task_row, torch::Tensor task_id, torch::Tensor off,
+    torch::Tensor len, int64_t n_tasks, int64_t D, int64_t H, int64_t O,
+    int64_t kind, bool want_mse, c10::optional<torch::Tensor> x_mask) {
+  const int W = task_row.size(0);
+  auto optd = torch::TensorOptions().dtype(torch::kFloat64)
+                  .device(params.device());
+  auto correct = torch::zeros({n_tasks}, optd);
+  auto total = torch::zeros({n_tasks}, optd);
+  auto loss = torch::zeros({n_tasks}, optd);
+  auto mse = want_mse ? torch::zeros({n_tasks}, optd) : torch::Tensor();
+  if (W == 0)
+    return {correct, total, loss, mse};
+  const int P = params.size(1);
+
+  EvalArgs args;
+  args.params = params.data_ptr<float>();
+  args.task_row = task_row.data_ptr<int64_t>();
+  args.task_id = task_id.data_ptr<int64_t>();
+  args.off = off.data_ptr<int64_t>();
+  args.len = len.data_ptr<int64_t>();
+  args.x = x.data_ptr<float>();
+  args.y = y.data_ptr<int64_t>();
+  args.x_mask = x_mask.has_value() ? x_mask->data_ptr<float>() : nullptr;
+  args.correct = correct.data_ptr<double>();
+  args.total = total.data_ptr<double>();
+  args.loss = loss.data_ptr<double>();
+  args.mse = want_mse ? mse.data_ptr<double>() : nullptr;
+  args.D = (int)D; args.H = (int)H; args.O = (int)O; args.P = P;
+  args.kind = (int)kind;
+
+  const size_t lds_bytes = (size_t)(P + 8) * sizeof(float);
+  TORCH_CHECK(lds_bytes <= 150 * 1024, "model too large for eval LDS path");
+  hipLaunchKernelGGL(mlp_eval_kernel, dim3(W), dim3(THREADS), lds_bytes,
+                     c10::hip::getCurrentHIPStream(), args);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "mlp_eval_kernel launch");
+  return {correct, total, loss, mse};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("train_fused", &train_fused_hip, "fused batched MLP local training");
+  mod.def("eval_tasks", &eval_tasks_hip, "batched MLP accuracy/loss sweep");
+}

@@ -1,0 +1,60 @@
+"""GPU op module: hand-written CDNA4 HIP kernels for the hot paths.
+
+Same API as ops.mlp_torch. train_fused and eval_tasks (the per-round hot
+loops: all local training in one launch, all accuracy sweeps in one launch)
+run on the HIP kernels in ops/hip/feddrift_kernels.hip; the cold paths
+(ensemble-vote testing, KUE confusion matrices — a few times per
+iteration) currently run on torch GPU ops and will be ported next.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import torch
+
+from ..models.packed import MLPSpec
+from . import hip_loader, mlp_torch
+
+make_opt_state = mlp_torch.make_opt_state
+forward_logits = mlp_torch.forward_logits
+ens_vote_eval = mlp_torch.ens_vote_eval
+confusion_tasks = mlp_torch.confusion_tasks
+
+_KIND = {"fnn": 0, "lr": 1}
+
+
+def train_fused(spec: MLPSpec, params_all: torch.Tensor, rows: torch.Tensor,
+                x_arena: torch.Tensor, y_arena: torch.Tensor,
+                step_off: torch.Tensor, step_len: torch.Tensor, opt: Dict,
+                x_mask: Optional[torch.Tensor] = None) -> None:
+    if rows.numel() == 0:
+        return
+    mod = hip_loader.load()
+    adam = opt["kind"] == "adam"
+    mod.train_fused(
+        params_all, rows.contiguous(), x_arena, y_arena,
+        step_off.contiguous(), step_len.contiguous(),
+        spec.d, spec.h, spec.o, _KIND[spec.kind],
+        x_mask.contiguous() if x_mask is not None else None,
+        opt["m"] if adam else None,
+        opt["v"] if adam else None,
+        opt["vmax"] if adam else None,
+        opt["t"] if adam else None,
+        opt["lr"], float(opt.get("wd", 0.0)))
+
+
+def eval_tasks(spec: MLPSpec, params: torch.Tensor,
+               x_arena: torch.Tensor, y_arena: torch.Tensor,
+               task_row: torch.Tensor, task_id: torch.Tensor,
+               win_off: torch.Tensor, win_len: torch.Tensor, n_tasks: int,
+               want_mse: bool = False,
+               x_mask: Optional[torch.Tensor] = None):
+    mod = hip_loader.load()
+    correct, total, loss, mse = mod.eval_tasks(
+        params.contiguous(), x_arena, y_arena,
+        task_row.contiguous(), task_id.contiguous(),
+        win_off.contiguous(), win_len.contiguous(), n_tasks,
+        spec.d, spec.h, spec.o, _KIND[spec.kind], want_mse,
+        x_mask.contiguous() if x_mask is not None else None)
+    return correct, total, loss, (mse if want_mse else None)

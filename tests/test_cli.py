@@ -1,0 +1,57 @@
+"""CLI surface: prepare_data + per-iteration main_fedavg + bench contract."""
+
+import json
+import os
+import subprocess
+import sys
+
+import numpy as np
+
+REPO = os.path.abspath(os.path.join(os.path.dirname(__file__), ".."))
+
+
+def test_prepare_and_main_fedavg(tmp_path):
+    data = str(tmp_path / "data")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "scripts", "prepare_data.py"),
+         "--dataset", "sea", "--data_dir", data, "--sample_num", "200",
+         "--client_num_in_total", "4", "--train_iteration", "2",
+         "--change_points", "rand"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert os.path.exists(os.path.join(data, "sea", "client_0_iter_2.csv"))
+    assert os.path.exists(os.path.join(data, "changepoints", "rand.cp"))
+
+    for it in range(2):
+        r = subprocess.run(
+            [sys.executable, os.path.join(REPO, "scripts", "main_fedavg.py"),
+             "--dataset", "sea", "--data_dir", data, "--model", "fnn",
+             "--client_num_in_total", "4", "--client_num_per_round", "4",
+             "--batch_size", "200", "--comm_round", "3", "--epochs", "2",
+             "--total_train_iteration", "2", "--curr_train_iteration",
+             str(it), "--concept_num", "2", "--concept_drift_algo",
+             "softcluster", "--concept_drift_algo_arg", "H_A_C_1_10_0",
+             "--change_points", "rand", "--log_dir", str(tmp_path),
+             "--report_client", "0"],
+            capture_output=True, text=True, timeout=600, cwd=str(tmp_path))
+        assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
+        assert "avg Test/Acc" in r.stdout
+    assert os.path.exists(str(tmp_path / "model_params.pt"))
+
+
+def test_bench_contract():
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--steps", "3",
+         "--warmup", "1"],
+        capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-1500:]
+    line = [l for l in r.stdout.strip().splitlines()
+            if l.startswith("{")][-1]
+    out = json.loads(line)
+    for key in ["metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"]:
+        assert key in out
+    assert out["steps"] == 3
+    assert out["data"] == "synthetic"
+    assert out["value"] > 0

@@ -1,0 +1,128 @@
+"""Engine integration: full timelines on CPU, checkpoints, determinism,
+learning (accuracy recovers after drift)."""
+
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+from feddrift_amd.config import Config
+from feddrift_amd.comm import Communicator
+from feddrift_amd.engine.timeline import run_timeline
+
+
+def _write_data(tmp_path, n_clients=6, iters=4, drift_at=2):
+    from feddrift_amd.data.generators import generate_data
+    d = str(tmp_path)
+    os.makedirs(os.path.join(d, "changepoints"), exist_ok=True)
+    mat = np.zeros((iters + 1, n_clients), dtype=int)
+    mat[drift_at:, : n_clients // 2] = 1
+    np.savetxt(os.path.join(d, "changepoints", "T.cp"), mat, fmt="%u")
+    np.random.seed(0)
+    generate_data("sea", d, iters, n_clients, 0, 300, 0.0, 1, "T")
+    return d
+
+
+def _cfg(data_dir, tmp, algo, arg="", **kw):
+    base = dict(model="fnn", dataset="sea", data_dir=data_dir,
+                client_num_in_total=6, client_num_per_round=6,
+                batch_size=300, lr=0.01, epochs=5, comm_round=8,
+                total_train_iteration=4, concept_num=2,
+                concept_drift_algo=algo, concept_drift_algo_arg=arg,
+                change_points="T", dummy_arg=0, log_dir=str(tmp),
+                report_client=0)
+    base.update(kw)
+    return Config(**base)
+
+
+@pytest.fixture(scope="module")
+def sea_dir(tmp_path_factory):
+    return _write_data(tmp_path_factory.mktemp("sea"))
+
+
+def test_feddrift_learns_and_adapts(sea_dir, tmp_path):
+    cfg = _cfg(sea_dir, tmp_path, "softcluster", "H_A_C_1_10_0",
+               comm_round=20)
+    out = run_timeline(cfg)
+    # SEA Bayes accuracy is 0.9; FedDrift should exceed 0.8 on average after
+    # warm-up iterations and end well-adapted
+    assert out["per_iteration_test_acc"][-1] > 0.82
+    assert out["avg_test_acc"] > 0.75
+
+
+def test_checkpoint_layout(sea_dir, tmp_path):
+    cfg = _cfg(sea_dir, tmp_path, "softcluster", "H_A_C_1_10_0",
+               total_train_iteration=2)
+    run_timeline(cfg)
+    # reference-compatible files (FedAvgEnsServerManager.py:84-86 +
+    # sc_state pickling)
+    assert os.path.exists(os.path.join(str(tmp_path), "model_params.pt"))
+    assert os.path.exists(os.path.join(str(tmp_path), "sc_state.pkl"))
+    import torch
+    mp = torch.load(os.path.join(str(tmp_path), "model_params.pt"))
+    assert set(mp.keys()) == {0, 1}
+    assert "fc1.weight" in mp[0]
+
+
+def test_determinism_same_seed(sea_dir, tmp_path):
+    cfg1 = _cfg(sea_dir, tmp_path / "a", "softcluster", "mmacc_06",
+                total_train_iteration=3)
+    os.makedirs(str(tmp_path / "a"), exist_ok=True)
+    os.makedirs(str(tmp_path / "b"), exist_ok=True)
+    out1 = run_timeline(cfg1)
+    cfg2 = _cfg(sea_dir, tmp_path / "b", "softcluster", "mmacc_06",
+                total_train_iteration=3)
+    out2 = run_timeline(cfg2)
+    assert out1["per_iteration_test_acc"] == out2["per_iteration_test_acc"]
+
+
+def test_iteration_resume_matches_inprocess(sea_dir, tmp_path):
+    """Per-iteration invocation through checkpoint files must equal the
+    in-process timeline (the reference's per-mpirun operation model)."""
+    import dataclasses
+    from feddrift_amd.engine.fljob import FLJob
+    from feddrift_amd.engine.timeline import clean_state_files
+    from feddrift_amd.eval.metrics import MetricLogger
+
+    os.makedirs(str(tmp_path / "x"), exist_ok=True)
+    os.makedirs(str(tmp_path / "y"), exist_ok=True)
+    comm = Communicator()
+
+    cfgx = _cfg(sea_dir, tmp_path / "x", "softcluster", "H_A_C_1_10_0",
+                total_train_iteration=2)
+    outx = run_timeline(cfgx)
+
+    cfgy = _cfg(sea_dir, tmp_path / "y", "softcluster", "H_A_C_1_10_0",
+                total_train_iteration=2)
+    clean_state_files(cfgy)
+    accs = []
+    for it in range(2):
+        icfg = dataclasses.replace(cfgy, curr_train_iteration=it)
+        logger = MetricLogger(str(tmp_path / "y"), enabled=True,
+                              to_file=False)
+        job = FLJob(icfg, comm, logger)
+        job.run()
+        s = logger.series("Test/Acc")
+        accs.append(sum(s) / len(s))
+    assert np.allclose(accs, outx["per_iteration_test_acc"], atol=1e-12)
+
+
+@pytest.mark.parametrize("algo,arg", [
+    ("aue", ""), ("kue", ""), ("driftsurf", ""), ("ada", "win-1_round"),
+    ("exp", ""), ("mmacc", ""), ("softcluster", "hard"),
+    ("softcluster", "cfl_0.1_win-1"), ("single", "")])
+def test_algorithms_run_short(sea_dir, tmp_path, algo, arg):
+    cfg = _cfg(sea_dir, tmp_path, algo, arg, comm_round=4,
+               total_train_iteration=2)
+    out = run_timeline(cfg)
+    assert np.isfinite(out["avg_test_acc"])
+    assert out["avg_test_acc"] > 0.4
+
+
+def test_single_model_win1_beats_chance(sea_dir, tmp_path):
+    cfg = _cfg(sea_dir, tmp_path, "single", "", retrain_data="win-1",
+               comm_round=15)
+    out = run_timeline(cfg)
+    assert out["avg_test_acc"] > 0.7

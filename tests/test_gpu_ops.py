@@ -1,0 +1,144 @@
+"""GPU numerics: the HIP kernels must match the torch (fp32) reference
+implementation (ops/mlp_torch.py) — which itself matches the reference's
+eager loops exactly (see test_ops.py)."""
+
+import numpy as np
+import pytest
+import torch
+
+from feddrift_amd.models.packed import spec_for
+from feddrift_amd.ops import mlp_torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="needs ROCm GPU")
+
+
+def _setup(spec_kind, d, o, n=600, G=7, E=5, seed=0):
+    torch.manual_seed(seed)
+    spec = spec_for(spec_kind, d, o)
+    x = torch.rand(n, d) * 10
+    y = torch.randint(0, o, (n,))
+    params = torch.randn(G + 2, spec.n_params) * 0.3
+    rows = torch.randperm(G + 2)[:G]
+    # windows of varying lengths, incl. one zero-length (skip) step
+    off = torch.randint(0, n - 101, (G, E))
+    ln = torch.randint(1, 100, (G, E))
+    ln[1, 2] = 0
+    return spec, x, y, params, rows, off, ln
+
+
+@requires_gpu
+@pytest.mark.parametrize("kind,d,o", [("fnn", 3, 2), ("fnn", 8, 5),
+                                      ("lr", 6, 3)])
+@pytest.mark.parametrize("optname", ["adam", "sgd"])
+def test_hip_train_matches_torch(kind, d, o, optname):
+    from feddrift_amd.ops import mlp_hip
+    spec, x, y, params, rows, off, ln = _setup(kind, d, o)
+    dev = torch.device("cuda:0")
+
+    p_ref = params.clone()
+    opt_ref = mlp_torch.make_opt_state(optname, params.shape[0],
+                                       spec.n_params, 0.01, 0.001, "cpu")
+    mlp_torch.train_fused(spec, p_ref, rows, x, y, off, ln, opt_ref)
+
+    p_gpu = params.to(dev)
+    opt_gpu = mlp_torch.make_opt_state(optname, params.shape[0],
+                                       spec.n_params, 0.01, 0.001, dev)
+    mlp_hip.train_fused(spec, p_gpu, rows.to(dev), x.to(dev), y.to(dev),
+                        off.to(dev), ln.to(dev), opt_gpu)
+    torch.cuda.synchronize()
+
+    diff = (p_gpu.cpu() - p_ref).abs().max().item()
+    assert diff < 5e-5, diff
+    if optname == "adam":
+        assert torch.equal(opt_gpu["t"].cpu(), opt_ref["t"])
+        md = (opt_gpu["m"].cpu() - opt_ref["m"]).abs().max().item()
+        assert md < 5e-5, md
+
+
+@requires_gpu
+def test_hip_train_with_mask():
+    from feddrift_amd.ops import mlp_hip
+    spec, x, y, params, rows, off, ln = _setup("fnn", 5, 3)
+    dev = torch.device("cuda:0")
+    mask = (torch.rand(rows.shape[0], 5) > 0.4).float()
+
+    p_ref = params.clone()
+    opt_ref = mlp_torch.make_opt_state("adam", params.shape[0],
+                                       spec.n_params, 0.01, 0.001, "cpu")
+    mlp_torch.train_fused(spec, p_ref, rows, x, y, off, ln, opt_ref,
+                          x_mask=mask)
+
+    p_gpu = params.to(dev)
+    opt_gpu = mlp_torch.make_opt_state("adam", params.shape[0],
+                                       spec.n_params, 0.01, 0.001, dev)
+    mlp_hip.train_fused(spec, p_gpu, rows.to(dev), x.to(dev), y.to(dev),
+                        off.to(dev), ln.to(dev), opt_gpu, x_mask=mask.to(dev))
+    torch.cuda.synchronize()
+    assert (p_gpu.cpu() - p_ref).abs().max().item() < 5e-5
+
+
+@requires_gpu
+@pytest.mark.parametrize("kind,d,o", [("fnn", 3, 2), ("lr", 6, 3)])
+def test_hip_eval_matches_torch(kind, d, o):
+    from feddrift_amd.ops import mlp_hip
+    torch.manual_seed(3)
+    spec = spec_for(kind, d, o)
+    n = 2000
+    x = torch.rand(n, d) * 10
+    y = torch.randint(0, o, (n,))
+    params = torch.randn(4, spec.n_params) * 0.4
+    W = 30
+    task_row = torch.randint(0, 4, (W,))
+    task_id = torch.randint(0, 6, (W,))
+    off = torch.randint(0, n - 130, (W,))
+    ln = torch.randint(1, 128, (W,))
+
+    c0, t0, l0, m0 = mlp_torch.eval_tasks(spec, params, x, y, task_row,
+                                          task_id, off, ln, 6, want_mse=True)
+    dev = torch.device("cuda:0")
+    c1, t1, l1, m1 = mlp_hip.eval_tasks(
+        spec, params.to(dev), x.to(dev), y.to(dev), task_row.to(dev),
+        task_id.to(dev), off.to(dev), ln.to(dev), 6, want_mse=True)
+    torch.cuda.synchronize()
+    assert torch.equal(c1.cpu(), c0)      # correct counts are integral
+    assert torch.equal(t1.cpu(), t0)
+    assert (l1.cpu() - l0).abs().max().item() < 2e-2
+    assert (m1.cpu() - m0).abs().max().item() < 2e-2
+
+
+@requires_gpu
+def test_gpu_full_round_smoke():
+    import __graft_entry__
+    __graft_entry__.smoke()
+
+
+@requires_gpu
+def test_gpu_timeline_accuracy():
+    """Full FedDrift timeline on GPU with HIP kernels: must learn."""
+    import os
+    import tempfile
+    from feddrift_amd.config import Config
+    from feddrift_amd.data.generators import generate_data
+    from feddrift_amd.engine.timeline import run_timeline
+
+    with tempfile.TemporaryDirectory() as td:
+        d = os.path.join(td, "data")
+        os.makedirs(os.path.join(d, "changepoints"))
+        mat = np.zeros((4, 6), dtype=int)
+        mat[2:, :3] = 1
+        np.savetxt(os.path.join(d, "changepoints", "T.cp"), mat, fmt="%u")
+        np.random.seed(0)
+        generate_data("sea", d, 3, 6, 0, 300, 0.0, 1, "T")
+        cfg = Config(model="fnn", dataset="sea", data_dir=d,
+                     client_num_in_total=6, client_num_per_round=6,
+                     batch_size=300, lr=0.01, epochs=5, comm_round=15,
+                     total_train_iteration=3, concept_num=2,
+                     concept_drift_algo="softcluster",
+                     concept_drift_algo_arg="H_A_C_1_10_0",
+                     change_points="T", dummy_arg=0, log_dir=td,
+                     report_client=0, use_hip_kernels="always")
+        out = run_timeline(cfg)
+        assert out["per_iteration_test_acc"][-1] > 0.8, out
