@@ -127,33 +127,8 @@ class AlgoBase:
                           train_on_view: bool = True):
         """test_on_all_clients core: per-client _infer of the chosen model on
         the view train data (or current-iter all_data) + the test set."""
-        C = job.cfg.client_num_in_total
-        if train_model_per_client is None:
-            train_model_per_client = model_per_client
-        clients = range(C) if job.cfg.ci != 1 else range(1)
-        tl = TaskList()
-        for c in clients:
-            tid_tr = tl.new_task()
-            tid_te = tl.new_task()
-            if job.comm.owns_client(c):
-                mt = int(train_model_per_client[c])
-                me = int(model_per_client[c])
-                if train_on_view:
-                    if c in job.view_train_ref[mt]:
-                        tl.add_windows(tid_tr, mt,
-                                       job.view_train_ref[mt][c].windows)
-                else:
-                    tl.add_windows(tid_tr, mt,
-                                   job.all_ref[c][job.curr_iter].windows)
-                if c in job.test_ref:
-                    tl.add_windows(tid_te, me, job.test_ref[c].windows)
-        correct, total, loss, _ = job.run_eval(job.global_params, tl)
-        buf = torch.from_numpy(np.stack([correct, total, loss])).to(job.device)
-        job.comm.all_reduce_(buf)
-        correct, total, loss = buf.cpu().numpy()
-        tr = (correct[0::2], total[0::2], loss[0::2])
-        te = (correct[1::2], total[1::2], loss[1::2])
-        return tr, te
+        return job.client_eval(model_per_client, train_model_per_client,
+                               train_on_view=train_on_view)
 
 
 # ---------------------------------------------------------------------------
@@ -461,10 +436,9 @@ class AueAlgo(AlgoBase):
                 ids[(mi, c)] = tid
                 if job.comm.owns_client(c) and c in job.view_train_ref[0]:
                     tl.add_windows(tid, mi, job.view_train_ref[0][c].windows)
-        _, total, _, mse = job.run_eval(job.global_params, tl, want_mse=True)
-        buf = torch.from_numpy(np.stack([total, mse])).to(job.device)
-        job.comm.all_reduce_(buf)
-        total, mse = buf.cpu().numpy()
+        res = job.run_eval_dev(job.global_params, tl, want_mse=True)
+        job.comm.all_reduce_(res)
+        _, total, _, mse = res.cpu().numpy()
 
         if self.per_client:
             for c in range(C):

@@ -154,6 +154,7 @@ class FLJob:
         self.pick_rng = np.random.default_rng(
             (cfg.dummy_arg * 7919 + self.curr_iter) * 1009 + comm.rank)
 
+        self._eval_cache: Dict = {}
         self.algo.init_iteration(self)
 
     # ------------------------------------------------------------------
@@ -170,19 +171,34 @@ class FLJob:
                 self.global_params.unsqueeze(0).expand(nW, -1, -1)
                 .reshape(-1, self.spec.n_params))
 
-    def run_eval(self, params: torch.Tensor, tl: TaskList,
-                 want_mse: bool = False):
+    def eval_tensors(self, tl: TaskList):
+        """Upload one task list as device tensors (cacheable)."""
         dev = self.device
-        tr = torch.as_tensor(tl.task_row, dtype=torch.int64, device=dev)
-        ti = torch.as_tensor(tl.task_id, dtype=torch.int64, device=dev)
-        wo = torch.as_tensor(tl.off, dtype=torch.int64, device=dev)
-        wl = torch.as_tensor(tl.ln, dtype=torch.int64, device=dev)
+        idx = torch.as_tensor(
+            np.stack([np.asarray(tl.task_row, dtype=np.int64),
+                      np.asarray(tl.task_id, dtype=np.int64),
+                      np.asarray(tl.off, dtype=np.int64),
+                      np.asarray(tl.ln, dtype=np.int64)]), device=dev)
+        return idx
+
+    def run_eval_dev(self, params: torch.Tensor, tl: TaskList,
+                     want_mse: bool = False,
+                     idx: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Batched eval sweep; returns a stacked DEVICE tensor
+        [correct; total; loss (; mse)] x n_tasks (float64) — callers
+        all_reduce it and download once."""
+        if idx is None:
+            idx = self.eval_tensors(tl)
         c, t, l, m = self.backend.eval_tasks(
             self.spec, params, self.arena.x, self.arena.y,
-            tr, ti, wo, wl, tl.n_tasks, want_mse=want_mse)
-        out = [c.cpu().numpy(), t.cpu().numpy(), l.cpu().numpy()]
-        out.append(m.cpu().numpy() if m is not None else None)
-        return out
+            idx[0], idx[1], idx[2], idx[3], tl.n_tasks, want_mse=want_mse)
+        parts = [c, t, l] + ([m] if m is not None else [])
+        return torch.stack(parts)
+
+    def run_eval(self, params: torch.Tensor, tl: TaskList,
+                 want_mse: bool = False):
+        out = self.run_eval_dev(params, tl, want_mse).cpu().numpy()
+        return [out[0], out[1], out[2], out[3] if want_mse else None]
 
     def train(self, plan: TrainPlan) -> None:
         if plan.rows.size == 0:
@@ -215,12 +231,13 @@ class FLJob:
             partial[:, :P] = torch.einsum("wk,wkp->kp", w, reps)
             partial[:, P] = w.sum(dim=0)
         self.comm.all_reduce_(partial)
-        totals = partial[:, P].cpu().numpy()
-        for m in range(K):
-            if model_mask is not None and not model_mask[m]:
-                continue
-            if totals[m] > 0:
-                self.global_params[m] = partial[m, :P] / partial[m, P]
+        totals = partial[:, P]
+        upd = totals > 0
+        if model_mask is not None:
+            upd &= torch.as_tensor(model_mask, device=self.device)
+        newp = partial[:, :P] / totals.clamp(min=1e-30).unsqueeze(1)
+        self.global_params.copy_(
+            torch.where(upd.unsqueeze(1), newp, self.global_params))
         return totals
 
     def client_sampling(self, round_idx: int) -> np.ndarray:
@@ -274,13 +291,9 @@ class FLJob:
                 if self.comm.owns_client(c):
                     tl.add_windows(tid, m,
                                    self.all_ref[c][self.curr_iter].windows)
-        correct, total, _, _ = self.run_eval(self.global_params, tl)
-        cv = torch.zeros(2, tl.n_tasks, dtype=torch.float64)
-        cv[0] = torch.from_numpy(correct)
-        cv[1] = torch.from_numpy(total)
-        cv = cv.to(self.device)
-        self.comm.all_reduce_(cv)
-        cv = cv.cpu().numpy()
+        res = self.run_eval_dev(self.global_params, tl)
+        self.comm.all_reduce_(res)
+        cv = res.cpu().numpy()
         acc = np.zeros((len(model_rows), C))
         for (i, c), tid in ids.items():
             if cv[1][tid] != 0:
@@ -320,29 +333,50 @@ class FLJob:
                 acc[i][j] = correct[tid] / total[tid]
         return acc
 
-    # per-client eval of chosen model rows on train (curr iter) + test (t+1)
-    def client_eval(self, model_idx_per_client: np.ndarray):
-        """Returns per-client (train_correct, train_total, train_loss,
-        test_correct, test_total, test_loss) arrays, allreduced.
+    # per-client eval of chosen model rows on train + test (t+1) data
+    def client_eval(self, model_idx_per_client: np.ndarray,
+                    train_model_per_client: Optional[np.ndarray] = None,
+                    train_on_view: bool = False):
+        """Returns per-client (train_correct, train_total, train_loss) and
+        (test_correct, test_total, test_loss) arrays, allreduced.
 
-        ci==1 keeps only client 0 (reference CI shortcut,
-        FedAvgEnsAggregatorSoftCluster.py:259-264)."""
+        train data is the current-iteration all_data (softcluster,
+        FedAvgEnsAggregatorSoftCluster.py:227-231) or the per-model retrain
+        view (the other aggregators) when train_on_view=True.
+        ci==1 keeps only client 0 (reference CI shortcut, :259-264).
+        The task tensors are cached per model assignment — the assignment
+        changes only at clustering events, so steady-state rounds skip the
+        task-list rebuild + upload entirely."""
+        if train_model_per_client is None:
+            train_model_per_client = model_idx_per_client
+        key = (model_idx_per_client.tobytes(),
+               train_model_per_client.tobytes(), train_on_view)
+        cached = self._eval_cache.get(key)
         C = self.cfg.client_num_in_total
         clients = range(C) if self.cfg.ci != 1 else range(1)
-        tl = TaskList()
-        for c in clients:
-            tid_tr = tl.new_task()
-            tid_te = tl.new_task()
-            if self.comm.owns_client(c):
-                m = int(model_idx_per_client[c])
-                tl.add_windows(tid_tr, m,
-                               self.all_ref[c][self.curr_iter].windows)
-                if c in self.test_ref:
-                    tl.add_windows(tid_te, m, self.test_ref[c].windows)
-        correct, total, loss, _ = self.run_eval(self.global_params, tl)
-        buf = torch.from_numpy(np.stack([correct, total, loss])).to(self.device)
-        self.comm.all_reduce_(buf)
-        correct, total, loss = buf.cpu().numpy()
+        if cached is None:
+            tl = TaskList()
+            for c in clients:
+                tid_tr = tl.new_task()
+                tid_te = tl.new_task()
+                if self.comm.owns_client(c):
+                    mt = int(train_model_per_client[c])
+                    me = int(model_idx_per_client[c])
+                    if train_on_view:
+                        if c in self.view_train_ref[mt]:
+                            tl.add_windows(tid_tr, mt,
+                                           self.view_train_ref[mt][c].windows)
+                    else:
+                        tl.add_windows(tid_tr, mt,
+                                       self.all_ref[c][self.curr_iter].windows)
+                    if c in self.test_ref:
+                        tl.add_windows(tid_te, me, self.test_ref[c].windows)
+            cached = (tl, self.eval_tensors(tl))
+            self._eval_cache[key] = cached
+        tl, idx = cached
+        res = self.run_eval_dev(self.global_params, tl, idx=idx)
+        self.comm.all_reduce_(res)
+        correct, total, loss = res.cpu().numpy()
         n = len(list(clients))
         tr = (correct[0::2][:n], total[0::2][:n], loss[0::2][:n])
         te = (correct[1::2][:n], total[1::2][:n], loss[1::2][:n])

@@ -109,61 +109,77 @@ void mlp_train_kernel(TrainArgs a) {
     __syncthreads();
 
     for (int c0 = 0; c0 < n; c0 += a.BC) {
-      const int bc = min(a.BC, n - c0);
+      const int bc = (a.BC < n - c0) ? a.BC : (n - c0);
 
-      // stage input chunk (optionally feature-masked: KUE)
+      // phase A: stage input chunk (optionally feature-masked: KUE)
       for (int q = tid; q < bc * a.D; q += THREADS) {
         const int i = q / a.D;
-        const int d = q % a.D;
+        const int d = q - i * a.D;
         float xv = a.x[(off + c0 + i) * a.D + d];
         if (a.x_mask) xv *= a.x_mask[(int64_t)g * a.D + d];
         xb[q] = xv;
       }
       __syncthreads();
 
-      // forward + output gradient per sample
-      for (int i = tid; i < bc; i += THREADS) {
-        const int64_t yi = a.y[off + c0 + i];
-        if (a.kind == KIND_FNN) {
-          // z1 = W1 x + b1 ; act = relu(z1)
-          for (int h = 0; h < a.H; ++h) {
-            float z = w[HD + h];
-            for (int d = 0; d < a.D; ++d) z += w[h * a.D + d] * xb[i * a.D + d];
-            act[i * a.H + h] = z > 0.f ? z : 0.f;
-          }
-          // z2 = W2 a + b2 -> dz2 = (softmax(z2) - onehot) / n
+      if (a.kind == KIND_FNN) {
+        // phase B: z1/act, one work item per (sample, hidden unit) —
+        // keeps all 256 threads busy and the dependent chains short
+        // (the original sample-serial h-loop left the workgroup
+        // latency-bound at ~10 active threads for SEA-sized models)
+        for (int q = tid; q < bc * a.H; q += THREADS) {
+          const int i = q / a.H;
+          const int h = q - i * a.H;
+          float z = w[HD + h];
+          for (int d = 0; d < a.D; ++d)
+            z += w[h * a.D + d] * xb[i * a.D + d];
+          act[q] = z > 0.f ? z : 0.f;
+        }
+        __syncthreads();
+
+        // phase C: z2 + softmax + dz2 per sample (O is small)
+        for (int i = tid; i < bc; i += THREADS) {
+          const int yi = (int)a.y[off + c0 + i];
+          float z2[64];
           float zmax = -1e30f;
           for (int o = 0; o < a.O; ++o) {
             float z = w[HD + a.H + OH + o];
             for (int h = 0; h < a.H; ++h)
               z += w[HD + a.H + o * a.H + h] * act[i * a.H + h];
-            dzo[i * a.O + o] = z;
+            z2[o] = z;
             zmax = fmaxf(zmax, z);
           }
           float zsum = 0.f;
           for (int o = 0; o < a.O; ++o) {
-            const float ez = __expf(dzo[i * a.O + o] - zmax);
-            dzo[i * a.O + o] = ez;
-            zsum += ez;
+            z2[o] = __expf(z2[o] - zmax);
+            zsum += z2[o];
           }
           for (int o = 0; o < a.O; ++o) {
-            float sm = dzo[i * a.O + o] / zsum;
-            if (o == (int)yi) sm -= 1.f;
+            float sm = z2[o] / zsum;
+            if (o == yi) sm -= 1.f;
             dzo[i * a.O + o] = sm * inv_n;
           }
-          // dz1 = (W2^T dz2) * relu'(z1)
-          for (int h = 0; h < a.H; ++h) {
-            float s = 0.f;
-            for (int o = 0; o < a.O; ++o)
-              s += dzo[i * a.O + o] * w[HD + a.H + o * a.H + h];
-            dza[i * a.H + h] = act[i * a.H + h] > 0.f ? s : 0.f;
-          }
-        } else {  // LR: out = sigmoid(Wx+b); CE applied to the sigmoid
+        }
+        __syncthreads();
+
+        // phase D: dz1 per (sample, hidden unit)
+        for (int q = tid; q < bc * a.H; q += THREADS) {
+          const int i = q / a.H;
+          const int h = q - i * a.H;
+          float s = 0.f;
+          for (int o = 0; o < a.O; ++o)
+            s += dzo[i * a.O + o] * w[HD + a.H + o * a.H + h];
+          dza[q] = act[q] > 0.f ? s : 0.f;
+        }
+        __syncthreads();
+      } else {  // LR: out = sigmoid(Wx+b); CE applied to the sigmoid
+        for (int i = tid; i < bc; i += THREADS) {
+          const int yi = (int)a.y[off + c0 + i];
           float p_[64];  // O <= 64 for the LR path
           float pmax = -1e30f;
           for (int o = 0; o < a.O; ++o) {
             float z = w[OD + o];
-            for (int d = 0; d < a.D; ++d) z += w[o * a.D + d] * xb[i * a.D + d];
+            for (int d = 0; d < a.D; ++d)
+              z += w[o * a.D + d] * xb[i * a.D + d];
             const float pv = 1.f / (1.f + __expf(-z));
             p_[o] = pv;
             pmax = fmaxf(pmax, pv);
@@ -172,48 +188,52 @@ void mlp_train_kernel(TrainArgs a) {
           for (int o = 0; o < a.O; ++o) psum += __expf(p_[o] - pmax);
           for (int o = 0; o < a.O; ++o) {
             float sm = __expf(p_[o] - pmax) / psum;
-            if (o == (int)yi) sm -= 1.f;
+            if (o == yi) sm -= 1.f;
             // dz = dp * sigmoid' = dp * p (1-p)
             dzo[i * a.O + o] = sm * inv_n * p_[o] * (1.f - p_[o]);
           }
         }
+        __syncthreads();
       }
-      __syncthreads();
 
-      // gradient accumulation: thread-owned parameter partition
-      if (a.kind == KIND_FNN) {
-        for (int p = tid; p < a.P; p += THREADS) {
+      // phase E: gradient accumulation. NSUB threads share each parameter
+      // entry, each summing a sample stride, then one LDS atomicAdd —
+      // P*NSUB work items keep the block busy for small P
+      {
+        int nsub = (2 * THREADS) / (a.P > 0 ? a.P : 1);
+        nsub = nsub < 1 ? 1 : (nsub > 32 ? 32 : nsub);
+        for (int q = tid; q < a.P * nsub; q += THREADS) {
+          const int p = q % a.P;
+          const int sub = q / a.P;
           float acc = 0.f;
-          if (p < HD) {
-            const int h = p / a.D, d = p % a.D;
-            for (int i = 0; i < bc; ++i)
-              acc += dza[i * a.H + h] * xb[i * a.D + d];
-          } else if (p < HD + a.H) {
-            const int h = p - HD;
-            for (int i = 0; i < bc; ++i) acc += dza[i * a.H + h];
-          } else if (p < HD + a.H + OH) {
-            const int q = p - HD - a.H;
-            const int o = q / a.H, h = q % a.H;
-            for (int i = 0; i < bc; ++i)
-              acc += dzo[i * a.O + o] * act[i * a.H + h];
+          if (a.kind == KIND_FNN) {
+            if (p < HD) {
+              const int h = p / a.D, d = p - (p / a.D) * a.D;
+              for (int i = sub; i < bc; i += nsub)
+                acc += dza[i * a.H + h] * xb[i * a.D + d];
+            } else if (p < HD + a.H) {
+              const int h = p - HD;
+              for (int i = sub; i < bc; i += nsub) acc += dza[i * a.H + h];
+            } else if (p < HD + a.H + OH) {
+              const int qq = p - HD - a.H;
+              const int o = qq / a.H, h = qq - (qq / a.H) * a.H;
+              for (int i = sub; i < bc; i += nsub)
+                acc += dzo[i * a.O + o] * act[i * a.H + h];
+            } else {
+              const int o = p - HD - a.H - OH;
+              for (int i = sub; i < bc; i += nsub) acc += dzo[i * a.O + o];
+            }
           } else {
-            const int o = p - HD - a.H - OH;
-            for (int i = 0; i < bc; ++i) acc += dzo[i * a.O + o];
+            if (p < OD) {
+              const int o = p / a.D, d = p - (p / a.D) * a.D;
+              for (int i = sub; i < bc; i += nsub)
+                acc += dzo[i * a.O + o] * xb[i * a.D + d];
+            } else {
+              const int o = p - OD;
+              for (int i = sub; i < bc; i += nsub) acc += dzo[i * a.O + o];
+            }
           }
-          grad[p] += acc;
-        }
-      } else {
-        for (int p = tid; p < a.P; p += THREADS) {
-          float acc = 0.f;
-          if (p < OD) {
-            const int o = p / a.D, d = p % a.D;
-            for (int i = 0; i < bc; ++i)
-              acc += dzo[i * a.O + o] * xb[i * a.D + d];
-          } else {
-            const int o = p - OD;
-            for (int i = 0; i < bc; ++i) acc += dzo[i * a.O + o];
-          }
-          grad[p] += acc;
+          atomicAdd(&grad[p], acc);
         }
       }
       __syncthreads();

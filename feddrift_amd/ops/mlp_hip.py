@@ -44,6 +44,32 @@ def train_fused(spec: MLPSpec, params_all: torch.Tensor, rows: torch.Tensor,
         opt["lr"], float(opt.get("wd", 0.0)))
 
 
+_EVAL_CHUNK = 128
+
+
+def _chunk_windows(task_row, task_id, win_off, win_len, x_mask):
+    """Split long windows into <=128-sample pieces: more workgroups, shorter
+    per-thread latency chains (the kernel accumulates per task via
+    atomicAdd, so chunking is free)."""
+    if win_len.numel() == 0 or int(win_len.max()) <= _EVAL_CHUNK:
+        return task_row, task_id, win_off, win_len, x_mask
+    nch = (win_len + (_EVAL_CHUNK - 1)) // _EVAL_CHUNK
+    task_row = task_row.repeat_interleave(nch)
+    task_id = task_id.repeat_interleave(nch)
+    base_off = win_off.repeat_interleave(nch)
+    base_len = win_len.repeat_interleave(nch)
+    if x_mask is not None and x_mask.dim() == 2:
+        x_mask = x_mask.repeat_interleave(nch, dim=0)
+    # position of each chunk within its window
+    csum = torch.cumsum(nch, 0)
+    start = torch.repeat_interleave(csum - nch, nch)
+    pos = torch.arange(task_row.numel(), device=task_row.device) - start
+    off = base_off + pos * _EVAL_CHUNK
+    ln = torch.minimum(base_len - pos * _EVAL_CHUNK,
+                       torch.full_like(base_len, _EVAL_CHUNK))
+    return task_row, task_id, off, ln, x_mask
+
+
 def eval_tasks(spec: MLPSpec, params: torch.Tensor,
                x_arena: torch.Tensor, y_arena: torch.Tensor,
                task_row: torch.Tensor, task_id: torch.Tensor,
@@ -51,6 +77,8 @@ def eval_tasks(spec: MLPSpec, params: torch.Tensor,
                want_mse: bool = False,
                x_mask: Optional[torch.Tensor] = None):
     mod = hip_loader.load()
+    task_row, task_id, win_off, win_len, x_mask = _chunk_windows(
+        task_row, task_id, win_off, win_len, x_mask)
     correct, total, loss, mse = mod.eval_tasks(
         params.contiguous(), x_arena, y_arena,
         task_row.contiguous(), task_id.contiguous(),
