@@ -58,7 +58,6 @@ def build_dataset(n_clients: int, seed: int) -> DriftDataset:
 
 
 def one_round(job: FLJob, r: int, client_idx) -> None:
-    job.sync_replicas()
     plan = job.algo.plan(job, r, client_idx)
     job.train(plan)
     job.algo.aggregate(job, r, plan, client_idx)

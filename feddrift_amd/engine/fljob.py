@@ -69,15 +69,19 @@ class TaskList:
         self.n_tasks += 1
         return self.n_tasks - 1
 
+    CHUNK = 128  # split long windows: more workgroups, short latency chains
+
     def add_windows(self, task_id: int, row: int,
                     windows: Sequence[Tuple[int, int]]) -> None:
         for o, l in windows:
-            if l <= 0:
-                continue
-            self.task_row.append(row)
-            self.task_id.append(task_id)
-            self.off.append(o)
-            self.ln.append(l)
+            while l > 0:
+                ln = min(l, self.CHUNK)
+                self.task_row.append(row)
+                self.task_id.append(task_id)
+                self.off.append(o)
+                self.ln.append(ln)
+                o += ln
+                l -= ln
 
 
 class FLJob:
@@ -155,6 +159,8 @@ class FLJob:
             (cfg.dummy_arg * 7919 + self.curr_iter) * 1009 + comm.rank)
 
         self._eval_cache: Dict = {}
+        self._partial: Optional[torch.Tensor] = None
+        self._partial_fused = False
         self.algo.init_iteration(self)
 
     # ------------------------------------------------------------------
@@ -201,43 +207,84 @@ class FLJob:
         return [out[0], out[1], out[2], out[3] if want_mse else None]
 
     def train(self, plan: TrainPlan) -> None:
-        if plan.rows.size == 0:
-            return
+        """All local training of this round in ONE fused launch. On the HIP
+        path the kernel also (a) stages each pair's initial weights straight
+        from the global model row (fusing the server->client broadcast) and
+        (b) accumulates the weighted aggregation partial sums on its way out
+        — so the whole round's compute is train-kernel + all_reduce +
+        apply-kernel."""
+        K, P = self.n_models, self.spec.n_params
         dev = self.device
-        self.backend.train_fused(
-            self.spec, self.replicas,
-            torch.as_tensor(plan.rows, dtype=torch.int64, device=dev),
-            self.arena.x, self.arena.y,
-            torch.as_tensor(plan.step_off, dtype=torch.int64, device=dev),
-            torch.as_tensor(plan.step_len, dtype=torch.int64, device=dev),
-            self.opt, x_mask=plan.x_mask)
+        hip = self.backend is not ops.mlp_torch
+        if self._partial is None or self._partial.shape[0] != K:
+            self._partial = torch.zeros(K, P + 1, device=dev)
+        if hip:
+            self._partial.zero_()
+        if plan.rows.size == 0:
+            self._partial_fused = hip
+            if not hip:
+                self.sync_replicas()
+            return
+        rows_t = torch.as_tensor(plan.rows, dtype=torch.int64, device=dev)
+        off_t = torch.as_tensor(plan.step_off, dtype=torch.int64, device=dev)
+        len_t = torch.as_tensor(plan.step_len, dtype=torch.int64, device=dev)
+        if hip:
+            wi = plan.rows // K
+            mo = (plan.rows % K).astype(np.int32)
+            sw = plan.sample_num[wi, plan.rows % K].astype(np.float32)
+            self.backend.train_fused(
+                self.spec, self.replicas, rows_t, self.arena.x, self.arena.y,
+                off_t, len_t, self.opt, x_mask=plan.x_mask,
+                in_params=self.global_params,
+                model_of=torch.as_tensor(mo, device=dev),
+                sample_w=torch.as_tensor(sw, device=dev),
+                partial=self._partial)
+            self._partial_fused = True
+        else:
+            self.sync_replicas()
+            self.backend.train_fused(
+                self.spec, self.replicas, rows_t, self.arena.x, self.arena.y,
+                off_t, len_t, self.opt, x_mask=plan.x_mask)
+            self._partial_fused = False
 
     def aggregate(self, plan: TrainPlan,
-                  model_mask: Optional[np.ndarray] = None) -> np.ndarray:
-        """Per-model sample-weighted averaging across all workers.
+                  model_mask: Optional[np.ndarray] = None) -> torch.Tensor:
+        """Per-model sample-weighted averaging across all workers: one
+        all_reduce of the fused [K, P+1] partial-sum tensor + one apply op.
 
         model_mask[m]=False skips the model entirely (its global params are
         left untouched), matching the softcluster skip rule
         (FedAvgEnsAggregatorSoftCluster.py:151-153). Returns total weights
-        per model. Models with zero total weight are skipped
-        (:167-169)."""
+        per model (device tensor). Models with zero total weight are
+        skipped (:167-169)."""
         K, P = self.n_models, self.spec.n_params
-        partial = torch.zeros(K, P + 1, device=self.device)
-        nW = len(self.owned_workers)
-        if nW:
-            w = torch.as_tensor(plan.sample_num, dtype=torch.float32,
-                                device=self.device)          # [nW, K]
-            reps = self.replicas.reshape(nW, K, P)
-            partial[:, :P] = torch.einsum("wk,wkp->kp", w, reps)
-            partial[:, P] = w.sum(dim=0)
+        partial = self._partial
+        if not self._partial_fused:
+            nW = len(self.owned_workers)
+            partial.zero_()
+            if nW:
+                w = torch.as_tensor(plan.sample_num, dtype=torch.float32,
+                                    device=self.device)      # [nW, K]
+                reps = self.replicas.reshape(nW, K, P)
+                partial[:, :P] = torch.einsum("wk,wkp->kp", w, reps)
+                partial[:, P] = w.sum(dim=0)
         self.comm.all_reduce_(partial)
         totals = partial[:, P]
-        upd = totals > 0
+        mask_t = None
         if model_mask is not None:
-            upd &= torch.as_tensor(model_mask, device=self.device)
-        newp = partial[:, :P] / totals.clamp(min=1e-30).unsqueeze(1)
-        self.global_params.copy_(
-            torch.where(upd.unsqueeze(1), newp, self.global_params))
+            mask_t = torch.as_tensor(
+                np.ascontiguousarray(model_mask, dtype=np.uint8),
+                device=self.device)
+        if self.backend is not ops.mlp_torch:
+            from ..ops import mlp_hip
+            mlp_hip.apply_aggregate(self.global_params, partial, mask_t)
+        else:
+            upd = totals > 0
+            if mask_t is not None:
+                upd &= mask_t.bool()
+            newp = partial[:, :P] / totals.clamp(min=1e-30).unsqueeze(1)
+            self.global_params.copy_(
+                torch.where(upd.unsqueeze(1), newp, self.global_params))
         return totals
 
     def client_sampling(self, round_idx: int) -> np.ndarray:
@@ -443,7 +490,6 @@ class FLJob:
     def run(self) -> None:
         client_idx = self.client_sampling(0)
         for r in range(self.cfg.comm_round):
-            self.sync_replicas()
             plan = self.algo.plan(self, r, client_idx)
             self.train(plan)
             self.algo.aggregate(self, r, plan, client_idx)

@@ -43,7 +43,11 @@
 // ---------------------------------------------------------------------------
 
 struct TrainArgs {
-  float* __restrict__ params;        // [R, P]
+  float* __restrict__ params;        // [R, P] replicas (out)
+  const float* __restrict__ in_params;  // [K, P] global models (in), or null
+  const int* __restrict__ model_of;  // [G] model per pair (with in_params)
+  const float* __restrict__ sample_w;   // [G] aggregation weight, or null
+  float* __restrict__ partial;       // [K, P+1] weighted sums, or null
   const int64_t* __restrict__ rows;  // [G]
   const float* __restrict__ x;       // [N, D]
   const int64_t* __restrict__ y;     // [N]
@@ -90,8 +94,15 @@ void mlp_train_kernel(TrainArgs a) {
   float* dza = act + (a.kind == KIND_FNN ? a.BC * a.H : 0);  // [BC, H] dz1
   float* dzo = dza + (a.kind == KIND_FNN ? a.BC * a.H : 0);  // [BC, O]
 
-  // stage weights
-  for (int p = tid; p < a.P; p += THREADS) w[p] = a.params[row * a.P + p];
+  // stage weights: straight from the GLOBAL model row (replaces the
+  // reference's server->client model broadcast and the engine's
+  // replica-sync copy — the round starts here)
+  if (a.in_params) {
+    const int64_t src = (int64_t)a.model_of[g] * a.P;
+    for (int p = tid; p < a.P; p += THREADS) w[p] = a.in_params[src + p];
+  } else {
+    for (int p = tid; p < a.P; p += THREADS) w[p] = a.params[row * a.P + p];
+  }
   __syncthreads();
 
   const int HD = a.H * a.D;
@@ -267,8 +278,37 @@ void mlp_train_kernel(TrainArgs a) {
     __syncthreads();
   }
 
-  // write back trained weights
+  // write back trained weights + fused aggregation partial sums
+  // (device-scope atomics; the weighted per-model average replaces the
+  // reference's pickled-state_dict triple loop,
+  // FedAvgEnsAggregatorSoftCluster.py:148-195)
   for (int p = tid; p < a.P; p += THREADS) a.params[row * a.P + p] = w[p];
+  if (a.partial && a.sample_w) {
+    const float sw = a.sample_w[g];
+    if (sw > 0.f) {
+      const int64_t pbase = (int64_t)a.model_of[g] * (a.P + 1);
+      for (int p = tid; p < a.P; p += THREADS)
+        atomicAdd(&a.partial[pbase + p], sw * w[p]);
+      if (tid == 0) atomicAdd(&a.partial[pbase + a.P], sw);
+    }
+  }
+}
+
+// apply the aggregated average: models with positive total weight get
+// partial/total, others keep their parameters (skip rules:
+// FedAvgEnsAggregatorSoftCluster.py:151-169)
+extern "C" __global__ __launch_bounds__(THREADS)
+void apply_aggregate_kernel(float* __restrict__ global_params,
+                            const float* __restrict__ partial,
+                            const unsigned char* __restrict__ mask,
+                            int K, int P) {
+  const int m = blockIdx.y;
+  const float tot = partial[(int64_t)m * (P + 1) + P];
+  if (tot <= 0.f || (mask && !mask[m])) return;
+  const float inv = 1.0f / tot;
+  for (int p = blockIdx.x * THREADS + threadIdx.x; p < P;
+       p += gridDim.x * THREADS)
+    global_params[(int64_t)m * P + p] = partial[(int64_t)m * (P + 1) + p] * inv;
 }
 
 // ---------------------------------------------------------------------------
@@ -386,7 +426,11 @@ void train_fused_hip(torch::Tensor params, torch::Tensor rows,
                      c10::optional<torch::Tensor> v,
                      c10::optional<torch::Tensor> vmax,
                      c10::optional<torch::Tensor> t,
-                     torch::Tensor lr, double wd) {
+                     torch::Tensor lr, double wd,
+                     c10::optional<torch::Tensor> in_params,
+                     c10::optional<torch::Tensor> model_of,
+                     c10::optional<torch::Tensor> sample_w,
+                     c10::optional<torch::Tensor> partial) {
   const int G = rows.size(0);
   if (G == 0) return;
   check_f32_2d(params, "params");
@@ -405,6 +449,12 @@ void train_fused_hip(torch::Tensor params, torch::Tensor rows,
 
   TrainArgs args;
   args.params = params.data_ptr<float>();
+  args.in_params = in_params.has_value() ? in_params->data_ptr<float>()
+                                         : nullptr;
+  args.model_of = model_of.has_value() ? model_of->data_ptr<int>() : nullptr;
+  args.sample_w = sample_w.has_value() ? sample_w->data_ptr<float>()
+                                       : nullptr;
+  args.partial = partial.has_value() ? partial->data_ptr<float>() : nullptr;
   args.rows = rows.data_ptr<int64_t>();
   args.x = x.data_ptr<float>();
   args.y = y.data_ptr<int64_t>();
@@ -466,7 +516,24 @@ std::vector<torch::Tensor> eval_tasks_hip(
   return {correct, total, loss, mse};
 }
 
+void apply_aggregate_hip(torch::Tensor global_params, torch::Tensor partial,
+                         c10::optional<torch::Tensor> mask) {
+  const int K = global_params.size(0);
+  const int P = global_params.size(1);
+  const int bx = std::min(64, (P + THREADS - 1) / THREADS);
+  hipLaunchKernelGGL(apply_aggregate_kernel, dim3(bx, K), dim3(THREADS), 0,
+                     c10::hip::getCurrentHIPStream(),
+                     global_params.data_ptr<float>(),
+                     partial.data_ptr<float>(),
+                     mask.has_value() ? mask->data_ptr<unsigned char>()
+                                      : nullptr,
+                     K, P);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "apply_aggregate launch");
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("train_fused", &train_fused_hip, "fused batched MLP local training");
   mod.def("eval_tasks", &eval_tasks_hip, "batched MLP accuracy/loss sweep");
+  mod.def("apply_aggregate", &apply_aggregate_hip,
+          "masked weighted-average model update");
 }

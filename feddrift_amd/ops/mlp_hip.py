@@ -27,7 +27,14 @@ _KIND = {"fnn": 0, "lr": 1}
 def train_fused(spec: MLPSpec, params_all: torch.Tensor, rows: torch.Tensor,
                 x_arena: torch.Tensor, y_arena: torch.Tensor,
                 step_off: torch.Tensor, step_len: torch.Tensor, opt: Dict,
-                x_mask: Optional[torch.Tensor] = None) -> None:
+                x_mask: Optional[torch.Tensor] = None,
+                in_params: Optional[torch.Tensor] = None,
+                model_of: Optional[torch.Tensor] = None,
+                sample_w: Optional[torch.Tensor] = None,
+                partial: Optional[torch.Tensor] = None) -> None:
+    """in_params/model_of: stage initial weights straight from the global
+    model rows (fuses the round's model broadcast into the launch);
+    sample_w/partial: fused weighted aggregation partial sums."""
     if rows.numel() == 0:
         return
     mod = hip_loader.load()
@@ -41,33 +48,13 @@ def train_fused(spec: MLPSpec, params_all: torch.Tensor, rows: torch.Tensor,
         opt["v"] if adam else None,
         opt["vmax"] if adam else None,
         opt["t"] if adam else None,
-        opt["lr"], float(opt.get("wd", 0.0)))
+        opt["lr"], float(opt.get("wd", 0.0)),
+        in_params, model_of, sample_w, partial)
 
 
-_EVAL_CHUNK = 128
-
-
-def _chunk_windows(task_row, task_id, win_off, win_len, x_mask):
-    """Split long windows into <=128-sample pieces: more workgroups, shorter
-    per-thread latency chains (the kernel accumulates per task via
-    atomicAdd, so chunking is free)."""
-    if win_len.numel() == 0 or int(win_len.max()) <= _EVAL_CHUNK:
-        return task_row, task_id, win_off, win_len, x_mask
-    nch = (win_len + (_EVAL_CHUNK - 1)) // _EVAL_CHUNK
-    task_row = task_row.repeat_interleave(nch)
-    task_id = task_id.repeat_interleave(nch)
-    base_off = win_off.repeat_interleave(nch)
-    base_len = win_len.repeat_interleave(nch)
-    if x_mask is not None and x_mask.dim() == 2:
-        x_mask = x_mask.repeat_interleave(nch, dim=0)
-    # position of each chunk within its window
-    csum = torch.cumsum(nch, 0)
-    start = torch.repeat_interleave(csum - nch, nch)
-    pos = torch.arange(task_row.numel(), device=task_row.device) - start
-    off = base_off + pos * _EVAL_CHUNK
-    ln = torch.minimum(base_len - pos * _EVAL_CHUNK,
-                       torch.full_like(base_len, _EVAL_CHUNK))
-    return task_row, task_id, off, ln, x_mask
+def apply_aggregate(global_params: torch.Tensor, partial: torch.Tensor,
+                    mask: Optional[torch.Tensor] = None) -> None:
+    hip_loader.load().apply_aggregate(global_params, partial, mask)
 
 
 def eval_tasks(spec: MLPSpec, params: torch.Tensor,
@@ -76,9 +63,10 @@ def eval_tasks(spec: MLPSpec, params: torch.Tensor,
                win_off: torch.Tensor, win_len: torch.Tensor, n_tasks: int,
                want_mse: bool = False,
                x_mask: Optional[torch.Tensor] = None):
+    # NOTE: callers that build task lists on the host chunk long windows
+    # there (engine TaskList, cached per iteration) — more workgroups,
+    # shorter latency chains, zero per-call device work.
     mod = hip_loader.load()
-    task_row, task_id, win_off, win_len, x_mask = _chunk_windows(
-        task_row, task_id, win_off, win_len, x_mask)
     correct, total, loss, mse = mod.eval_tasks(
         params.contiguous(), x_arena, y_arena,
         task_row.contiguous(), task_id.contiguous(),

@@ -142,3 +142,59 @@ def test_gpu_timeline_accuracy():
                      report_client=0, use_hip_kernels="always")
         out = run_timeline(cfg)
         assert out["per_iteration_test_acc"][-1] > 0.8, out
+
+
+@requires_gpu
+def test_hip_fused_broadcast_and_partial():
+    """The in_params/partial fusion (stage from global rows + weighted
+    aggregation sums in the train launch) must equal the unfused sequence:
+    replica sync -> train -> einsum."""
+    from feddrift_amd.ops import mlp_hip
+    torch.manual_seed(2)
+    spec = spec_for("fnn", 3, 2)
+    dev = torch.device("cuda:0")
+    K, nW, E, n = 3, 4, 4, 500
+    P = spec.n_params
+    x = (torch.rand(n, 3) * 10).to(dev)
+    y = torch.randint(0, 2, (n,)).to(dev)
+    glob = (torch.randn(K, P) * 0.3).to(dev)
+    G = nW * K
+    rows = torch.arange(G, device=dev)
+    model_of = (rows % K).to(torch.int32)
+    sample_w = torch.rand(G, device=dev) + 0.5
+    off = torch.randint(0, n - 101, (G, E)).to(dev)
+    ln = torch.randint(1, 100, (G, E)).to(dev)
+
+    # reference: sync replicas then train then einsum
+    reps_ref = glob[rows % K].clone()
+    opt_ref = mlp_torch.make_opt_state("adam", G, P, 0.01, 0.001, dev)
+    mlp_torch.train_fused(spec, reps_ref, rows, x, y, off, ln, opt_ref)
+    part_ref = torch.zeros(K, P + 1, device=dev)
+    for g in range(G):
+        part_ref[g % K, :P] += sample_w[g] * reps_ref[g]
+        part_ref[g % K, P] += sample_w[g]
+
+    reps = torch.zeros(G, P, device=dev)
+    opt = mlp_torch.make_opt_state("adam", G, P, 0.01, 0.001, dev)
+    partial = torch.zeros(K, P + 1, device=dev)
+    mlp_hip.train_fused(spec, reps, rows, x, y, off, ln, opt,
+                        in_params=glob, model_of=model_of,
+                        sample_w=sample_w, partial=partial)
+    torch.cuda.synchronize()
+    assert (reps - reps_ref).abs().max().item() < 5e-5
+    assert (partial - part_ref).abs().max().item() < 3e-4
+
+    # apply: average models, verify against manual
+    glob2 = glob.clone()
+    mlp_hip.apply_aggregate(glob2, partial, None)
+    torch.cuda.synchronize()
+    want = part_ref[:, :P] / part_ref[:, P:P + 1]
+    assert (glob2 - want).abs().max().item() < 3e-4
+
+    # masked apply keeps masked-out rows
+    glob3 = glob.clone()
+    mask = torch.tensor([1, 0, 1], dtype=torch.uint8, device=dev)
+    mlp_hip.apply_aggregate(glob3, partial, mask)
+    torch.cuda.synchronize()
+    assert torch.equal(glob3[1], glob[1])
+    assert (glob3[0] - want[0]).abs().max().item() < 3e-4
