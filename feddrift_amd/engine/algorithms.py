@@ -27,7 +27,7 @@ from ..drift.driftsurf import DriftSurfState
 from ..drift.kue import KueState
 from ..drift.mmacc import MultiModelAccState
 from ..drift.softcluster import SoftClusterState
-from .fljob import FLJob, TaskList, TrainPlan
+from .fljob import FLJob, TaskList, TrainPlan, build_template
 
 
 def make(cfg: Config):
@@ -93,34 +93,35 @@ class AlgoBase:
         pass
 
     # ---- shared helpers ----
+    _tmpl = None
+    _tmpl_key = None
+
     def default_plan(self, job: FLJob, client_idx: np.ndarray,
                      lr: Optional[float] = None) -> TrainPlan:
         """Reference FedAvgEnsTrainer semantics: per model, E random batch
         picks from its retrain view; num_samples = the view's per-client
-        sample count (FedAvgEnsTrainer.py:47-95)."""
-        E = job.cfg.epochs
-        K = job.n_models
-        rows, offs, lens = [], [], []
-        nW = len(job.owned_workers)
-        sample = np.zeros((nW, K))
+        sample count (FedAvgEnsTrainer.py:47-95). The plan structure is
+        cached per client assignment; only the picks are drawn per round."""
         if lr is not None:
             job.opt["lr"].fill_(lr)
-        for wi, w in enumerate(job.owned_workers):
-            c = int(client_idx[w])
-            for m in range(K):
-                n = job.views[m].train_n(c)
-                sample[wi, m] = n
-                if n == 0:
-                    continue
-                wins = job.view_train_ref[m][c].windows
-                picks = job.pick_rng.integers(0, len(wins), size=E)
-                rows.append(job.row(wi, m))
-                offs.append([wins[p][0] for p in picks])
-                lens.append([wins[p][1] for p in picks])
-        return TrainPlan(np.asarray(rows, dtype=np.int64),
-                         np.asarray(offs, dtype=np.int64).reshape(-1, E),
-                         np.asarray(lens, dtype=np.int64).reshape(-1, E),
-                         sample)
+        key = client_idx.tobytes()
+        if self._tmpl is None or self._tmpl_key != key:
+            K = job.n_models
+            nW = len(job.owned_workers)
+            sample = np.zeros((nW, K))
+            pairs = []
+            for wi, w in enumerate(job.owned_workers):
+                c = int(client_idx[w])
+                for m in range(K):
+                    n = job.views[m].train_n(c)
+                    sample[wi, m] = n
+                    if n == 0:
+                        continue
+                    pairs.append((job.row(wi, m),
+                                  job.view_train_ref[m][c].windows))
+            self._tmpl = build_template(pairs, sample)
+            self._tmpl_key = key
+        return self._tmpl.draw(job.pick_rng, job.cfg.epochs)
 
     def client_eval_views(self, job: FLJob, model_per_client: np.ndarray,
                           train_model_per_client: Optional[np.ndarray] = None,
@@ -173,6 +174,8 @@ class Hooks:
 # CFL (cfl_*), softmax, gmm, geni
 # ---------------------------------------------------------------------------
 class SoftClusterAlgo(AlgoBase):
+    _mask = None
+
     def build_views(self, job: FLJob) -> list:
         # views are win-1 placeholders; the trainer uses all_data
         # (reference SoftCluster_data_loader:1334-1341)
@@ -265,58 +268,65 @@ class SoftClusterAlgo(AlgoBase):
 
         job.save_state_pickle("sc_state.pkl", st)
 
+    def invalidate_plan(self) -> None:
+        """Weights changed mid-iteration (hard-r recluster / CFL split):
+        rebuild the plan structure next round."""
+        self._tmpl = None
+        self._mask = None
+
     def plan(self, job: FLJob, round_idx: int,
              client_idx: np.ndarray) -> TrainPlan:
         """TrainerSoftCluster: pairs weighted by sc_weights x iteration batch
         counts; unit-weight fast path pools batches across selected
         iterations (FedAvgEnsTrainerSoftCluster.py:63-135). NOTE the
         reference weighs by len(all_local_data[t]) = the BATCH count of
-        iteration t, not the sample count — preserved."""
-        st = self.state
-        E = job.cfg.epochs
-        K = job.n_models
-        T = job.curr_iter + 1
-        rows, offs, lens = [], [], []
-        nW = len(job.owned_workers)
-        sample = np.zeros((nW, K))
-        w_iter = st.get_weights()
-        for wi, w in enumerate(job.owned_workers):
-            c = int(client_idx[w])
-            for m in range(K):
-                if not np.any(w_iter[job.curr_iter][m]):
-                    continue
-                unnorm = np.array([
-                    w_iter[t][m][c] * len(job.all_ref[c][t].windows)
-                    for t in range(T)])
-                tot = unnorm.sum()
-                if tot == 0:
-                    continue
-                sample[wi, m] = tot
-                pool = []
-                for t in range(T):
-                    if unnorm[t] > 0:
-                        pool.extend(job.all_ref[c][t].windows)
-                picks = job.pick_rng.integers(0, len(pool), size=E)
-                rows.append(job.row(wi, m))
-                offs.append([pool[p][0] for p in picks])
-                lens.append([pool[p][1] for p in picks])
-        return TrainPlan(np.asarray(rows, dtype=np.int64),
-                         np.asarray(offs, dtype=np.int64).reshape(-1, E),
-                         np.asarray(lens, dtype=np.int64).reshape(-1, E),
-                         sample)
+        iteration t, not the sample count — preserved. Structure cached;
+        only the random picks are drawn per round."""
+        key = client_idx.tobytes()
+        if self._tmpl is None or self._tmpl_key != key:
+            st = self.state
+            K = job.n_models
+            T = job.curr_iter + 1
+            nW = len(job.owned_workers)
+            sample = np.zeros((nW, K))
+            pairs = []
+            w_iter = st.get_weights()
+            active = [bool(np.any(w_iter[job.curr_iter][m]))
+                      for m in range(K)]
+            for wi, w in enumerate(job.owned_workers):
+                c = int(client_idx[w])
+                for m in range(K):
+                    if not active[m]:
+                        continue
+                    unnorm = np.array([
+                        w_iter[t][m][c] * len(job.all_ref[c][t].windows)
+                        for t in range(T)])
+                    tot = unnorm.sum()
+                    if tot == 0:
+                        continue
+                    sample[wi, m] = tot
+                    pool = []
+                    for t in range(T):
+                        if unnorm[t] > 0:
+                            pool.extend(job.all_ref[c][t].windows)
+                    pairs.append((job.row(wi, m), pool))
+            self._tmpl = build_template(pairs, sample)
+            self._tmpl_key = key
+            self._mask = np.asarray(active)
+        return self._tmpl.draw(job.pick_rng, job.cfg.epochs)
 
     def aggregate(self, job: FLJob, round_idx: int, plan: TrainPlan,
                   client_idx: np.ndarray) -> None:
         st = self.state
         if "cfl" in job.cfg.concept_drift_algo_arg:
             if self._cfl_round(job, round_idx, plan, client_idx):
+                self.invalidate_plan()
                 return  # split: skip this round's aggregation (:141-148)
-        mask = np.array([bool(np.any(st.get_weights()[job.curr_iter][m]))
-                         for m in range(job.n_models)])
-        job.aggregate(plan, model_mask=mask)
+        job.aggregate(plan, model_mask=self._mask)
         if job.cfg.concept_drift_algo_arg == "hard-r":
             acc = job.train_acc_matrix_rows(list(range(job.n_models)))
             st.cluster(Hooks(job), acc, job.curr_iter, round_idx + 1)
+            self.invalidate_plan()
 
     def _cfl_round(self, job: FLJob, round_idx: int, plan: TrainPlan,
                    client_idx: np.ndarray) -> bool:
@@ -826,15 +836,20 @@ class KueAlgo(AlgoBase):
         return torch.as_tensor(
             self.state.get_masks().astype(np.float32), device=job.device)
 
+    _row_mask = None
+
     def plan(self, job: FLJob, round_idx: int,
              client_idx: np.ndarray) -> TrainPlan:
         plan = self.default_plan(job, client_idx)
         if plan.rows.size:
-            masks = self._masks_tensor(job)
-            model_of_row = torch.as_tensor(plan.rows % job.n_models,
-                                           dtype=torch.int64,
-                                           device=job.device)
-            plan.x_mask = masks[model_of_row]
+            if self._row_mask is None or \
+                    self._row_mask.shape[0] != plan.rows.size:
+                masks = self._masks_tensor(job)
+                model_of_row = torch.as_tensor(plan.rows % job.n_models,
+                                               dtype=torch.int64,
+                                               device=job.device)
+                self._row_mask = masks[model_of_row].contiguous()
+            plan.x_mask = self._row_mask
         return plan
 
     def aggregate(self, job: FLJob, round_idx: int, plan: TrainPlan,

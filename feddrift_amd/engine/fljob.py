@@ -44,6 +44,50 @@ from .arena import DeviceArena, SegRef
 
 
 @dataclass
+class PlanTemplate:
+    """Cached structure of a round's training plan: which pairs train, from
+    which batch pools, with which aggregation weights. Only the random batch
+    picks change between rounds — drawing them is one vectorized RNG call
+    (the reference draws np.random.choice per model per step per worker,
+    FedAvgEnsTrainer.py:67; same distribution)."""
+    rows: np.ndarray         # [G]
+    sample_num: np.ndarray   # [nW_local, K]
+    pool_start: np.ndarray   # [G] into pool_off/pool_len
+    pool_size: np.ndarray    # [G]
+    pool_off: np.ndarray     # [sum pools]
+    pool_len: np.ndarray     # [sum pools]
+
+    def draw(self, rng: np.random.Generator, epochs: int) -> "TrainPlan":
+        G = len(self.rows)
+        if G == 0:
+            return TrainPlan(self.rows, np.zeros((0, epochs), np.int64),
+                             np.zeros((0, epochs), np.int64), self.sample_num)
+        u = rng.random((G, epochs))
+        pick = (u * self.pool_size[:, None]).astype(np.int64)
+        idx = self.pool_start[:, None] + pick
+        return TrainPlan(self.rows, self.pool_off[idx], self.pool_len[idx],
+                         self.sample_num)
+
+
+def build_template(pairs, sample_num) -> PlanTemplate:
+    """pairs: list of (row, windows); windows = [(off, len), ...] pools."""
+    rows, starts, sizes, offs, lens = [], [], [], [], []
+    for row, wins in pairs:
+        rows.append(row)
+        starts.append(len(offs))
+        sizes.append(len(wins))
+        for o, l in wins:
+            offs.append(o)
+            lens.append(l)
+    return PlanTemplate(np.asarray(rows, dtype=np.int64),
+                        sample_num,
+                        np.asarray(starts, dtype=np.int64),
+                        np.asarray(sizes, dtype=np.int64),
+                        np.asarray(offs, dtype=np.int64),
+                        np.asarray(lens, dtype=np.int64))
+
+
+@dataclass
 class TrainPlan:
     """Task arrays for one round of local training on this rank."""
     rows: np.ndarray            # [G] row ids into the replica buffer

@@ -80,6 +80,63 @@ __device__ __forceinline__ float block_reduce_sum(float val, float* scratch) {
   return val;  // valid in thread 0
 }
 
+__device__ __forceinline__ void stage_weights(const TrainArgs& a, int g,
+                                              int64_t row, float* w) {
+  if (a.in_params) {
+    const int64_t src = (int64_t)a.model_of[g] * a.P;
+    for (int p = threadIdx.x; p < a.P; p += THREADS)
+      w[p] = a.in_params[src + p];
+  } else {
+    for (int p = threadIdx.x; p < a.P; p += THREADS)
+      w[p] = a.params[row * a.P + p];
+  }
+  __syncthreads();
+}
+
+__device__ __forceinline__ void opt_update(const TrainArgs& a, int64_t row,
+                                           float* w, const float* grad) {
+  const int tid = threadIdx.x;
+  if (a.opt == OPT_SGD) {
+    const float lr_ = a.lr[row];
+    for (int p = tid; p < a.P; p += THREADS) w[p] -= lr_ * grad[p];
+    return;
+  }
+  if (tid == 0) a.t[row] += 1;
+  __syncthreads();
+  const int tnew = a.t[row];
+  const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
+  const float bc1 = 1.f - powf(b1, (float)tnew);
+  const float bc2 = 1.f - powf(b2, (float)tnew);
+  const float lr_ = a.lr[row];
+  for (int p = tid; p < a.P; p += THREADS) {
+    const int64_t gp = row * a.P + p;
+    const float gr = grad[p] + a.wd * w[p];
+    const float mn = b1 * a.m[gp] + (1.f - b1) * gr;
+    const float vn = b2 * a.v[gp] + (1.f - b2) * gr * gr;
+    a.m[gp] = mn;
+    a.v[gp] = vn;
+    const float vm = fmaxf(a.vmax[gp], vn);
+    a.vmax[gp] = vm;
+    const float denom = sqrtf(vm / bc2) + eps;
+    w[p] -= lr_ * (mn / bc1) / denom;
+  }
+}
+
+__device__ __forceinline__ void write_back(const TrainArgs& a, int g,
+                                           int64_t row, const float* w) {
+  const int tid = threadIdx.x;
+  for (int p = tid; p < a.P; p += THREADS) a.params[row * a.P + p] = w[p];
+  if (a.partial && a.sample_w) {
+    const float sw = a.sample_w[g];
+    if (sw > 0.f) {
+      const int64_t pbase = (int64_t)a.model_of[g] * (a.P + 1);
+      for (int p = tid; p < a.P; p += THREADS)
+        atomicAdd(&a.partial[pbase + p], sw * w[p]);
+      if (tid == 0) atomicAdd(&a.partial[pbase + a.P], sw);
+    }
+  }
+}
+
 extern "C" __global__ __launch_bounds__(THREADS)
 void mlp_train_kernel(TrainArgs a) {
   const int g = blockIdx.x;
@@ -97,13 +154,7 @@ void mlp_train_kernel(TrainArgs a) {
   // stage weights: straight from the GLOBAL model row (replaces the
   // reference's server->client model broadcast and the engine's
   // replica-sync copy — the round starts here)
-  if (a.in_params) {
-    const int64_t src = (int64_t)a.model_of[g] * a.P;
-    for (int p = tid; p < a.P; p += THREADS) w[p] = a.in_params[src + p];
-  } else {
-    for (int p = tid; p < a.P; p += THREADS) w[p] = a.params[row * a.P + p];
-  }
-  __syncthreads();
+  stage_weights(a, g, row, w);
 
   const int HD = a.H * a.D;
   const int OH = a.O * a.H;
@@ -251,47 +302,176 @@ void mlp_train_kernel(TrainArgs a) {
     }
 
     // optimizer update (matches torch.optim exactly; see ops/mlp_torch.py)
-    if (a.opt == OPT_SGD) {
-      const float lr_ = a.lr[row];
-      for (int p = tid; p < a.P; p += THREADS) w[p] -= lr_ * grad[p];
-    } else {
-      if (tid == 0) a.t[row] += 1;
-      __syncthreads();
-      const int tnew = a.t[row];
-      const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
-      const float bc1 = 1.f - powf(b1, (float)tnew);
-      const float bc2 = 1.f - powf(b2, (float)tnew);
-      const float lr_ = a.lr[row];
-      for (int p = tid; p < a.P; p += THREADS) {
-        const int64_t gp = row * a.P + p;
-        const float gr = grad[p] + a.wd * w[p];
-        const float mn = b1 * a.m[gp] + (1.f - b1) * gr;
-        const float vn = b2 * a.v[gp] + (1.f - b2) * gr * gr;
-        a.m[gp] = mn;
-        a.v[gp] = vn;
-        const float vm = fmaxf(a.vmax[gp], vn);
-        a.vmax[gp] = vm;
-        const float denom = sqrtf(vm / bc2) + eps;
-        w[p] -= lr_ * (mn / bc1) / denom;
-      }
-    }
+    opt_update(a, row, w, grad);
     __syncthreads();
   }
 
-  // write back trained weights + fused aggregation partial sums
-  // (device-scope atomics; the weighted per-model average replaces the
-  // reference's pickled-state_dict triple loop,
-  // FedAvgEnsAggregatorSoftCluster.py:148-195)
-  for (int p = tid; p < a.P; p += THREADS) a.params[row * a.P + p] = w[p];
-  if (a.partial && a.sample_w) {
-    const float sw = a.sample_w[g];
-    if (sw > 0.f) {
-      const int64_t pbase = (int64_t)a.model_of[g] * (a.P + 1);
-      for (int p = tid; p < a.P; p += THREADS)
-        atomicAdd(&a.partial[pbase + p], sw * w[p]);
-      if (tid == 0) atomicAdd(&a.partial[pbase + a.P], sw);
+  write_back(a, g, row, w);
+}
+
+// ---------------------------------------------------------------------------
+// specialized small-MLP training kernel: the whole per-sample pipeline and
+// the per-thread gradient accumulators live in REGISTERS (compile-time
+// D/H/O, fully unrolled); gradients reduce by wave shuffles + one LDS
+// atomic per wave. This is the SEA/SINE/CIRCLE flagship path — the generic
+// kernel above stays latency-bound on LDS round trips for these shapes.
+// ---------------------------------------------------------------------------
+
+template <int TD, int TH, int TO, int KIND>
+__global__ __launch_bounds__(THREADS)
+void mlp_train_small_kernel(TrainArgs a) {
+  constexpr int TP = (KIND == KIND_FNN)
+                         ? (TH * TD + TH + TO * TH + TO)
+                         : (TO * TD + TO);
+  constexpr int HD = TH * TD;
+  constexpr int OH = TO * TH;
+  const int g = blockIdx.x;
+  const int64_t row = a.rows[g];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+
+  __shared__ __attribute__((aligned(16))) float w[TP];
+  __shared__ __attribute__((aligned(16))) float grad[TP];
+  __shared__ float msk[TD > 0 ? TD : 1];
+
+  stage_weights(a, g, row, w);
+  if (a.x_mask && tid < TD) msk[tid] = a.x_mask[(int64_t)g * TD + tid];
+  __syncthreads();
+
+  for (int e = 0; e < a.E; ++e) {
+    const int64_t off = a.step_off[(int64_t)g * a.E + e];
+    const int n = (int)a.step_len[(int64_t)g * a.E + e];
+    if (n == 0) continue;
+    const float inv_n = 1.0f / (float)n;
+
+    for (int p = tid; p < TP; p += THREADS) grad[p] = 0.f;
+    __syncthreads();
+
+    float gacc[TP];
+#pragma unroll
+    for (int p = 0; p < TP; ++p) gacc[p] = 0.f;
+
+    for (int i = tid; i < n; i += THREADS) {
+      float x[TD];
+#pragma unroll
+      for (int d = 0; d < TD; ++d) {
+        x[d] = a.x[(off + i) * TD + d];
+        if (a.x_mask) x[d] *= msk[d];
+      }
+      const int yi = (int)a.y[off + i];
+      if constexpr (KIND == KIND_FNN) {
+        float act[TH], dzo[TO];
+#pragma unroll
+        for (int h = 0; h < TH; ++h) {
+          float z = w[HD + h];
+#pragma unroll
+          for (int d = 0; d < TD; ++d) z += w[h * TD + d] * x[d];
+          act[h] = z > 0.f ? z : 0.f;
+        }
+        float zmax = -1e30f;
+#pragma unroll
+        for (int o = 0; o < TO; ++o) {
+          float z = w[HD + TH + OH + o];
+#pragma unroll
+          for (int h = 0; h < TH; ++h) z += w[HD + TH + o * TH + h] * act[h];
+          dzo[o] = z;
+          zmax = fmaxf(zmax, z);
+        }
+        float zsum = 0.f;
+#pragma unroll
+        for (int o = 0; o < TO; ++o) {
+          dzo[o] = __expf(dzo[o] - zmax);
+          zsum += dzo[o];
+        }
+#pragma unroll
+        for (int o = 0; o < TO; ++o) {
+          float sm = dzo[o] / zsum;
+          if (o == yi) sm -= 1.f;
+          dzo[o] = sm * inv_n;
+        }
+        float dza[TH];
+#pragma unroll
+        for (int h = 0; h < TH; ++h) {
+          float s = 0.f;
+#pragma unroll
+          for (int o = 0; o < TO; ++o) s += dzo[o] * w[HD + TH + o * TH + h];
+          dza[h] = act[h] > 0.f ? s : 0.f;
+        }
+#pragma unroll
+        for (int h = 0; h < TH; ++h) {
+#pragma unroll
+          for (int d = 0; d < TD; ++d) gacc[h * TD + d] += dza[h] * x[d];
+          gacc[HD + h] += dza[h];
+        }
+#pragma unroll
+        for (int o = 0; o < TO; ++o) {
+#pragma unroll
+          for (int h = 0; h < TH; ++h)
+            gacc[HD + TH + o * TH + h] += dzo[o] * act[h];
+          gacc[HD + TH + OH + o] += dzo[o];
+        }
+      } else {  // LR
+        float p_[TO], dzo[TO];
+        float pmax = -1e30f;
+#pragma unroll
+        for (int o = 0; o < TO; ++o) {
+          float z = w[TO * TD + o];
+#pragma unroll
+          for (int d = 0; d < TD; ++d) z += w[o * TD + d] * x[d];
+          p_[o] = 1.f / (1.f + __expf(-z));
+          pmax = fmaxf(pmax, p_[o]);
+        }
+        float psum = 0.f;
+#pragma unroll
+        for (int o = 0; o < TO; ++o) psum += __expf(p_[o] - pmax);
+#pragma unroll
+        for (int o = 0; o < TO; ++o) {
+          float sm = __expf(p_[o] - pmax) / psum;
+          if (o == yi) sm -= 1.f;
+          dzo[o] = sm * inv_n * p_[o] * (1.f - p_[o]);
+        }
+#pragma unroll
+        for (int o = 0; o < TO; ++o) {
+#pragma unroll
+          for (int d = 0; d < TD; ++d) gacc[o * TD + d] += dzo[o] * x[d];
+          gacc[TO * TD + o] += dzo[o];
+        }
+      }
     }
+
+    // reduce: wave shuffles then one LDS atomic per (wave, entry)
+#pragma unroll
+    for (int p = 0; p < TP; ++p) {
+      float v = gacc[p];
+      for (int s = 32; s > 0; s >>= 1) v += __shfl_down(v, s, 64);
+      if (lane == 0) atomicAdd(&grad[p], v);
+    }
+    __syncthreads();
+
+    opt_update(a, row, w, grad);
+    __syncthreads();
   }
+
+  write_back(a, g, row, w);
+}
+
+#define SMALL_SHAPE_LIST(X)      \
+  X(3, 6, 2, KIND_FNN)           \
+  X(2, 4, 2, KIND_FNN)           \
+  X(4, 8, 2, KIND_FNN)           \
+  X(3, 0, 2, KIND_LR)            \
+  X(2, 0, 2, KIND_LR)
+
+static bool launch_small(const TrainArgs& args, int G, hipStream_t stream) {
+#define TRY_SHAPE(SD, SH, SO, SK)                                         \
+  if (args.kind == SK && args.D == SD && args.H == SH && args.O == SO) {  \
+    hipLaunchKernelGGL((mlp_train_small_kernel<SD, SH, SO, SK>), dim3(G), \
+                       dim3(THREADS), 0, stream, args);                   \
+    return true;                                                          \
+  }
+  SMALL_SHAPE_LIST(TRY_SHAPE)
+#undef TRY_SHAPE
+  return false;
 }
 
 // apply the aggregated average: models with positive total weight get
@@ -471,8 +651,10 @@ void train_fused_hip(torch::Tensor params, torch::Tensor rows,
   args.P = P; args.kind = (int)kind; args.opt = adam ? OPT_ADAM : OPT_SGD;
   args.BC = BC;
 
-  hipLaunchKernelGGL(mlp_train_kernel, dim3(G), dim3(THREADS), lds_bytes,
-                     c10::hip::getCurrentHIPStream(), args);
+  if (!launch_small(args, G, c10::hip::getCurrentHIPStream())) {
+    hipLaunchKernelGGL(mlp_train_kernel, dim3(G), dim3(THREADS), lds_bytes,
+                       c10::hip::getCurrentHIPStream(), args);
+  }
   TORCH_CHECK(hipGetLastError() == hipSuccess, "mlp_train_kernel launch");
 }
 
