@@ -312,7 +312,8 @@ class SoftClusterAlgo(AlgoBase):
                     pairs.append((job.row(wi, m), pool))
             self._tmpl = build_template(pairs, sample)
             self._tmpl_key = key
-            self._mask = np.asarray(active)
+            self._mask = torch.as_tensor(
+                np.asarray(active, dtype=np.uint8), device=job.device)
         return self._tmpl.draw(job.pick_rng, job.cfg.epochs)
 
     def aggregate(self, job: FLJob, round_idx: int, plan: TrainPlan,

@@ -61,12 +61,13 @@ class PlanTemplate:
         G = len(self.rows)
         if G == 0:
             return TrainPlan(self.rows, np.zeros((0, epochs), np.int64),
-                             np.zeros((0, epochs), np.int64), self.sample_num)
+                             np.zeros((0, epochs), np.int64), self.sample_num,
+                             template=self)
         u = rng.random((G, epochs))
         pick = (u * self.pool_size[:, None]).astype(np.int64)
         idx = self.pool_start[:, None] + pick
         return TrainPlan(self.rows, self.pool_off[idx], self.pool_len[idx],
-                         self.sample_num)
+                         self.sample_num, template=self)
 
 
 def build_template(pairs, sample_num) -> PlanTemplate:
@@ -97,6 +98,7 @@ class TrainPlan:
     # sample counts, per-algorithm — reference passes them as num_samples)
     sample_num: np.ndarray      # [nW_local, K]
     x_mask: Optional[torch.Tensor] = None   # [G, D] per-pair input mask (KUE)
+    template: Optional["PlanTemplate"] = None
 
 
 class TaskList:
@@ -239,11 +241,9 @@ class FLJob:
         all_reduce it and download once."""
         if idx is None:
             idx = self.eval_tensors(tl)
-        c, t, l, m = self.backend.eval_tasks(
+        return self.backend.eval_tasks_stacked(
             self.spec, params, self.arena.x, self.arena.y,
             idx[0], idx[1], idx[2], idx[3], tl.n_tasks, want_mse=want_mse)
-        parts = [c, t, l] + ([m] if m is not None else [])
-        return torch.stack(parts)
 
     def run_eval(self, params: torch.Tensor, tl: TaskList,
                  want_mse: bool = False):
@@ -269,23 +269,34 @@ class FLJob:
             if not hip:
                 self.sync_replicas()
             return
-        rows_t = torch.as_tensor(plan.rows, dtype=torch.int64, device=dev)
         off_t = torch.as_tensor(plan.step_off, dtype=torch.int64, device=dev)
         len_t = torch.as_tensor(plan.step_len, dtype=torch.int64, device=dev)
         if hip:
-            wi = plan.rows // K
-            mo = (plan.rows % K).astype(np.int32)
-            sw = plan.sample_num[wi, plan.rows % K].astype(np.float32)
+            # per-template constants (rows / model / aggregation weights)
+            # are uploaded once and reused every round
+            tmpl = plan.template
+            cache = getattr(tmpl, "_dev", None) if tmpl is not None else None
+            if cache is None:
+                wi = plan.rows // K
+                mo = (plan.rows % K).astype(np.int32)
+                sw = plan.sample_num[wi, plan.rows % K].astype(np.float32)
+                cache = (
+                    torch.as_tensor(plan.rows, dtype=torch.int64, device=dev),
+                    torch.as_tensor(mo, device=dev),
+                    torch.as_tensor(sw, device=dev))
+                if tmpl is not None:
+                    tmpl._dev = cache
+            rows_t, mo_t, sw_t = cache
             self.backend.train_fused(
                 self.spec, self.replicas, rows_t, self.arena.x, self.arena.y,
                 off_t, len_t, self.opt, x_mask=plan.x_mask,
                 in_params=self.global_params,
-                model_of=torch.as_tensor(mo, device=dev),
-                sample_w=torch.as_tensor(sw, device=dev),
+                model_of=mo_t, sample_w=sw_t,
                 partial=self._partial)
             self._partial_fused = True
         else:
             self.sync_replicas()
+            rows_t = torch.as_tensor(plan.rows, dtype=torch.int64, device=dev)
             self.backend.train_fused(
                 self.spec, self.replicas, rows_t, self.arena.x, self.arena.y,
                 off_t, len_t, self.opt, x_mask=plan.x_mask)
@@ -316,9 +327,12 @@ class FLJob:
         totals = partial[:, P]
         mask_t = None
         if model_mask is not None:
-            mask_t = torch.as_tensor(
-                np.ascontiguousarray(model_mask, dtype=np.uint8),
-                device=self.device)
+            if isinstance(model_mask, torch.Tensor):
+                mask_t = model_mask
+            else:
+                mask_t = torch.as_tensor(
+                    np.ascontiguousarray(model_mask, dtype=np.uint8),
+                    device=self.device)
         if self.backend is not ops.mlp_torch:
             from ..ops import mlp_hip
             mlp_hip.apply_aggregate(self.global_params, partial, mask_t)

@@ -332,10 +332,23 @@ void mlp_train_small_kernel(TrainArgs a) {
 
   __shared__ __attribute__((aligned(16))) float w[TP];
   __shared__ __attribute__((aligned(16))) float grad[TP];
+  // Adam state LDS-resident for the whole E-step loop (HBM round trips
+  // per optimizer step removed; staged back at the end)
+  __shared__ __attribute__((aligned(16))) float sm_[TP], sv_[TP], svm_[TP];
   __shared__ float msk[TD > 0 ? TD : 1];
+  __shared__ int tstep;
 
   stage_weights(a, g, row, w);
   if (a.x_mask && tid < TD) msk[tid] = a.x_mask[(int64_t)g * TD + tid];
+  const bool adam = (a.opt == OPT_ADAM);
+  if (adam) {
+    for (int p = tid; p < TP; p += THREADS) {
+      sm_[p] = a.m[row * TP + p];
+      sv_[p] = a.v[row * TP + p];
+      svm_[p] = a.vmax[row * TP + p];
+    }
+    if (tid == 0) tstep = a.t[row];
+  }
   __syncthreads();
 
   for (int e = 0; e < a.E; ++e) {
@@ -448,10 +461,40 @@ void mlp_train_small_kernel(TrainArgs a) {
     }
     __syncthreads();
 
-    opt_update(a, row, w, grad);
+    // optimizer step, all state in LDS (numerics match ops/mlp_torch.py)
+    if (!adam) {
+      const float lr_ = a.lr[row];
+      for (int p = tid; p < TP; p += THREADS) w[p] -= lr_ * grad[p];
+    } else {
+      if (tid == 0) tstep += 1;
+      __syncthreads();
+      const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
+      const float bc1 = 1.f - powf(b1, (float)tstep);
+      const float bc2 = 1.f - powf(b2, (float)tstep);
+      const float lr_ = a.lr[row];
+      for (int p = tid; p < TP; p += THREADS) {
+        const float gr = grad[p] + a.wd * w[p];
+        const float mn = b1 * sm_[p] + (1.f - b1) * gr;
+        const float vn = b2 * sv_[p] + (1.f - b2) * gr * gr;
+        sm_[p] = mn;
+        sv_[p] = vn;
+        const float vm = fmaxf(svm_[p], vn);
+        svm_[p] = vm;
+        const float denom = sqrtf(vm / bc2) + eps;
+        w[p] -= lr_ * (mn / bc1) / denom;
+      }
+    }
     __syncthreads();
   }
 
+  if (adam) {
+    for (int p = tid; p < TP; p += THREADS) {
+      a.m[row * TP + p] = sm_[p];
+      a.v[row * TP + p] = sv_[p];
+      a.vmax[row * TP + p] = svm_[p];
+    }
+    if (tid == 0) a.t[row] = tstep;
+  }
   write_back(a, g, row, w);
 }
 
@@ -658,7 +701,7 @@ void train_fused_hip(torch::Tensor params, torch::Tensor rows,
   TORCH_CHECK(hipGetLastError() == hipSuccess, "mlp_train_kernel launch");
 }
 
-std::vector<torch::Tensor> eval_tasks_hip(
+torch::Tensor eval_tasks_hip(
     torch::Tensor params, torch::Tensor x, torch::Tensor y,
     torch::Tensor task_row, torch::Tensor task_id, torch::Tensor off,
     torch::Tensor len, int64_t n_tasks, int64_t D, int64_t H, int64_t O,
@@ -666,12 +709,11 @@ std::vector<torch::Tensor> eval_tasks_hip(
   const int W = task_row.size(0);
   auto optd = torch::TensorOptions().dtype(torch::kFloat64)
                   .device(params.device());
-  auto correct = torch::zeros({n_tasks}, optd);
-  auto total = torch::zeros({n_tasks}, optd);
-  auto loss = torch::zeros({n_tasks}, optd);
-  auto mse = want_mse ? torch::zeros({n_tasks}, optd) : torch::Tensor();
+  // one [rows, n_tasks] output buffer: correct/total/loss(/mse) are its
+  // contiguous rows — no post-hoc stack/cat
+  auto out = torch::zeros({want_mse ? 4 : 3, n_tasks}, optd);
   if (W == 0)
-    return {correct, total, loss, mse};
+    return out;
   const int P = params.size(1);
 
   EvalArgs args;
@@ -683,10 +725,11 @@ std::vector<torch::Tensor> eval_tasks_hip(
   args.x = x.data_ptr<float>();
   args.y = y.data_ptr<int64_t>();
   args.x_mask = x_mask.has_value() ? x_mask->data_ptr<float>() : nullptr;
-  args.correct = correct.data_ptr<double>();
-  args.total = total.data_ptr<double>();
-  args.loss = loss.data_ptr<double>();
-  args.mse = want_mse ? mse.data_ptr<double>() : nullptr;
+  double* base = out.data_ptr<double>();
+  args.correct = base;
+  args.total = base + n_tasks;
+  args.loss = base + 2 * n_tasks;
+  args.mse = want_mse ? base + 3 * n_tasks : nullptr;
   args.D = (int)D; args.H = (int)H; args.O = (int)O; args.P = P;
   args.kind = (int)kind;
 
@@ -695,7 +738,7 @@ std::vector<torch::Tensor> eval_tasks_hip(
   hipLaunchKernelGGL(mlp_eval_kernel, dim3(W), dim3(THREADS), lds_bytes,
                      c10::hip::getCurrentHIPStream(), args);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "mlp_eval_kernel launch");
-  return {correct, total, loss, mse};
+  return out;
 }
 
 void apply_aggregate_hip(torch::Tensor global_params, torch::Tensor partial,

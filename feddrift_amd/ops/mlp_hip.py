@@ -66,11 +66,24 @@ def eval_tasks(spec: MLPSpec, params: torch.Tensor,
     # NOTE: callers that build task lists on the host chunk long windows
     # there (engine TaskList, cached per iteration) — more workgroups,
     # shorter latency chains, zero per-call device work.
+    out = eval_tasks_stacked(spec, params, x_arena, y_arena, task_row,
+                             task_id, win_off, win_len, n_tasks, want_mse,
+                             x_mask)
+    return out[0], out[1], out[2], (out[3] if want_mse else None)
+
+
+def eval_tasks_stacked(spec: MLPSpec, params: torch.Tensor,
+                       x_arena: torch.Tensor, y_arena: torch.Tensor,
+                       task_row: torch.Tensor, task_id: torch.Tensor,
+                       win_off: torch.Tensor, win_len: torch.Tensor,
+                       n_tasks: int, want_mse: bool = False,
+                       x_mask: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """[3 or 4, n_tasks] float64: correct/total/loss(/mse) rows in ONE
+    contiguous buffer (ready for all_reduce, no stack)."""
     mod = hip_loader.load()
-    correct, total, loss, mse = mod.eval_tasks(
+    return mod.eval_tasks(
         params.contiguous(), x_arena, y_arena,
         task_row.contiguous(), task_id.contiguous(),
         win_off.contiguous(), win_len.contiguous(), n_tasks,
         spec.d, spec.h, spec.o, _KIND[spec.kind], want_mse,
         x_mask.contiguous() if x_mask is not None else None)
-    return correct, total, loss, (mse if want_mse else None)

@@ -322,3 +322,18 @@ def confusion_tasks(spec: MLPSpec,
         0, flat.reshape(-1),
         torch.ones_like(flat, dtype=torch.float64).reshape(-1))
     return A
+
+
+@torch.no_grad()
+def eval_tasks_stacked(spec: MLPSpec, params: torch.Tensor,
+                       x_arena: torch.Tensor, y_arena: torch.Tensor,
+                       task_row: torch.Tensor, task_id: torch.Tensor,
+                       win_off: torch.Tensor, win_len: torch.Tensor,
+                       n_tasks: int, want_mse: bool = False,
+                       x_mask: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """[3 or 4, n_tasks] float64 stacked result (see mlp_hip)."""
+    c, t, l, m = eval_tasks(spec, params, x_arena, y_arena, task_row,
+                            task_id, win_off, win_len, n_tasks,
+                            want_mse=want_mse, x_mask=x_mask)
+    parts = [c, t, l] + ([m] if m is not None else [])
+    return torch.stack(parts)
