@@ -84,11 +84,51 @@ def sample_mnist(n: int, concept: int, rng: np.random.Generator) -> np.ndarray:
     return np.concatenate([x, y[:, None]], axis=1)
 
 
-_SAMPLERS = {"sea": sample_sea, "sine": sample_sine, "circle": sample_circle,
-             "MNIST": sample_mnist}
+# synthetic stand-ins for the image benchmark configs (no network access):
+# class-conditional Gaussian images with label-swap drift concepts, shaped
+# like CIFAR-10 (3x32x32, ResNet-18 config) and FederatedEMNIST (28x28,
+# 62 classes, CNN ensemble config)
+_PROTO_CACHE = {}
 
-FEATURE_NUM = {"sea": 3, "sine": 2, "circle": 2, "MNIST": 784}
-CLASS_NUM = {"sea": 2, "sine": 2, "circle": 2, "MNIST": 10}
+
+def _prototypes(n_classes: int, dim: int, seed: int) -> np.ndarray:
+    key = (n_classes, dim, seed)
+    if key not in _PROTO_CACHE:
+        prng = np.random.default_rng(seed)
+        _PROTO_CACHE[key] = prng.random((n_classes, dim)) * 0.8
+    return _PROTO_CACHE[key]
+
+
+def _sample_image_like(n: int, concept: int, rng: np.random.Generator,
+                       n_classes: int, dim: int, seed: int) -> np.ndarray:
+    proto = _prototypes(n_classes, dim, seed)
+    y = rng.integers(0, n_classes, size=n).astype(np.float64)
+    x = proto[y.astype(int)] + rng.normal(0.0, 0.25, size=(n, dim))
+    x = np.clip(x, 0.0, 1.0)
+    if concept != 0:
+        a, b = _MNIST_SWAPS.get(concept, (1, 2))
+        ya, yb = y == a, y == b
+        y[ya] = b
+        y[yb] = a
+    return np.concatenate([x, y[:, None]], axis=1)
+
+
+def sample_cifar(n, concept, rng):
+    return _sample_image_like(n, concept, rng, 10, 3072, 1414)
+
+
+def sample_femnist(n, concept, rng):
+    return _sample_image_like(n, concept, rng, 62, 784, 1703)
+
+
+_SAMPLERS = {"sea": sample_sea, "sine": sample_sine, "circle": sample_circle,
+             "MNIST": sample_mnist, "cifar": sample_cifar,
+             "femnist": sample_femnist}
+
+FEATURE_NUM = {"sea": 3, "sine": 2, "circle": 2, "MNIST": 784,
+               "cifar": 3072, "femnist": 784}
+CLASS_NUM = {"sea": 2, "sine": 2, "circle": 2, "MNIST": 10,
+             "cifar": 10, "femnist": 62}
 
 _SEA_COLS = ["f1", "f2", "f3", "label"]
 

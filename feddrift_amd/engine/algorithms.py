@@ -495,12 +495,10 @@ class AueAlgo(AlgoBase):
             if not job.comm.owns_client(c) or c not in job.test_ref:
                 continue
             w = self.ens_weights[c] if self.per_client else self.ens_weights
-            cc, tt = job.backend.ens_vote_eval(
-                job.spec, job.global_params,
+            cc, tt = job.ens_vote_eval(
                 torch.as_tensor(w[:M], dtype=torch.float32,
                                 device=job.device),
-                job.arena.x, job.arena.y, job.test_ref[c].windows,
-                mode="hard")
+                job.test_ref[c].windows, mode="hard")
             tec[i], tet[i] = cc, tt
         buf = torch.from_numpy(np.stack(
             [trc, trt, trl, tec, tet])).to(job.device)
@@ -536,24 +534,17 @@ class DriftSurfAlgo(AlgoBase):
         """run_ds_algo at data-load time (DriftSurf_data_loader:269-314):
         score the pickled models on the newest global batch (CPU eval; runs
         once per iteration, data is host-side at this point)."""
-        from ..ops import mlp_torch
-        spec = job.spec
 
         def score(key: str) -> float:
             flat = self.state.models.get(key)
             if flat is None:
                 return 0.0
-            params = torch.from_numpy(np.asarray(flat, dtype=np.float32)) \
-                .reshape(1, -1)
             correct = total = 0.0
             for c, seg in view.train.items():
                 if seg.n == 0:
                     continue
-                x = torch.from_numpy(seg.x)
-                y = torch.from_numpy(seg.y)
-                logits = mlp_torch.forward_logits(spec, params,
-                                                  x.unsqueeze(0))
-                correct += (logits.argmax(-1).squeeze(0) == y).sum().item()
+                logits = job.forward_flat(np.asarray(flat), seg.x)
+                correct += float((logits.argmax(-1) == seg.y).sum())
                 total += seg.n
             return correct / total if total else 0.0
 
@@ -644,8 +635,6 @@ class MultiModelAlgo(AlgoBase):
     def _run_select(self, job: FLJob, newest_view) -> None:
         """run_model_select with per-(model, client) scoring on the newest
         local batch (reference FedAvgEnsDataLoader.py:350-390)."""
-        from ..ops import mlp_torch
-        spec = job.spec
         cache: Dict[tuple, float] = {}
 
         def score(m: int, c: int) -> float:
@@ -657,12 +646,8 @@ class MultiModelAlgo(AlgoBase):
             seg = newest_view.train[c]
             if seg.n == 0:
                 return 0.0
-            params = torch.from_numpy(np.asarray(flat, dtype=np.float32)) \
-                .reshape(1, -1)
-            logits = mlp_torch.forward_logits(
-                spec, params, torch.from_numpy(seg.x).unsqueeze(0))
-            acc = float((logits.argmax(-1).squeeze(0) ==
-                         torch.from_numpy(seg.y)).sum().item()) / seg.n
+            logits = job.forward_flat(np.asarray(flat), seg.x)
+            acc = float((logits.argmax(-1) == seg.y).sum()) / seg.n
             cache[(m, c)] = acc
             return acc
 
@@ -870,8 +855,7 @@ class KueAlgo(AlgoBase):
             for c in range(job.cfg.client_num_in_total):
                 if job.comm.owns_client(c) and c in job.view_train_ref[m]:
                     tl.add_windows(tid, m, job.view_train_ref[m][c].windows)
-            A = job.backend.confusion_tasks(
-                job.spec, job.global_params, job.arena.x, job.arena.y,
+            A = job.confusion(
                 torch.as_tensor(tl.task_row, dtype=torch.int64,
                                 device=job.device),
                 torch.as_tensor(tl.task_id, dtype=torch.int64,
@@ -921,11 +905,9 @@ class KueAlgo(AlgoBase):
         for i, c in enumerate(clients):
             if not job.comm.owns_client(c) or c not in job.test_ref:
                 continue
-            cc, tt = job.backend.ens_vote_eval(
-                job.spec, job.global_params,
+            cc, tt = job.ens_vote_eval(
                 torch.as_tensor(w, dtype=torch.float32, device=job.device),
-                job.arena.x, job.arena.y, job.test_ref[c].windows,
-                mode="soft", masks=masks)
+                job.test_ref[c].windows, mode="soft", masks=masks)
             tec[i], tet[i] = cc, tt
         buf = torch.from_numpy(np.stack(
             [trc, trt, trl, tec, tet])).to(job.device)

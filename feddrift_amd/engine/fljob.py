@@ -150,9 +150,26 @@ class FLJob:
         self.curr_iter = cfg.curr_train_iteration
         self.dataset = dataset or DriftDataset(cfg.data_dir, cfg.dataset,
                                                cfg.client_num_in_total)
-        self.spec = packed.spec_for(cfg.model, self.dataset.feature_num,
-                                    self.dataset.class_num)
-        self.packer = packed.PackedMLP(self.spec)
+        # two model paths: the MLP family runs on the fused HIP kernels /
+        # batched torch ops; convolutional models (cnn / resnet) run as one
+        # live torch module per rank over flat state rows (MIOpen convs)
+        self.is_module_path = cfg.model not in ("lr", "fnn")
+        if self.is_module_path:
+            from ..models.generic_packer import ModulePacker
+            proto0 = zoo.create_model(cfg.model, self.dataset.class_num,
+                                      self.dataset.feature_num)
+            self.packer = ModulePacker(proto0)
+            self.spec = None
+            from ..ops.module_engine import ModuleEngine
+            self.mod_engine = ModuleEngine(proto0, self.packer, self.device)
+        else:
+            self.spec = packed.spec_for(cfg.model, self.dataset.feature_num,
+                                        self.dataset.class_num)
+            self.packer = packed.PackedMLP(self.spec)
+            self.mod_engine = None
+
+        self.n_params = self.packer.n_params if self.is_module_path \
+            else self.spec.n_params
 
         from . import algorithms
         self.algo = algorithms.make(cfg)
@@ -182,7 +199,7 @@ class FLJob:
         self.arena.freeze()
 
         # model bank: K global rows + canonical init row
-        P = self.spec.n_params
+        P = self.n_params
         proto = zoo.create_model(cfg.model, self.dataset.class_num,
                                  self.dataset.feature_num)
         self.init_flat = self.packer.flatten(proto.state_dict()).to(self.device)
@@ -195,9 +212,13 @@ class FLJob:
         self.owned_workers = comm.owned_workers(self.n_workers)
         nW = len(self.owned_workers)
         self.replicas = torch.zeros(nW * self.n_models, P, device=self.device)
-        self.opt = ops.mlp_torch.make_opt_state(
-            cfg.client_optimizer, nW * self.n_models, P, cfg.lr, cfg.wd,
-            self.device)
+        if self.is_module_path:
+            self.opt = self.mod_engine.make_opt_state(
+                cfg.client_optimizer, nW * self.n_models, cfg.lr, cfg.wd)
+        else:
+            self.opt = ops.mlp_torch.make_opt_state(
+                cfg.client_optimizer, nW * self.n_models, P, cfg.lr, cfg.wd,
+                self.device)
 
         # per-rank batch-pick RNG (statistical parity; reference draws on
         # each worker process's own global np RNG)
@@ -221,7 +242,7 @@ class FLJob:
         if nW:
             self.replicas.copy_(
                 self.global_params.unsqueeze(0).expand(nW, -1, -1)
-                .reshape(-1, self.spec.n_params))
+                .reshape(-1, self.n_params))
 
     def eval_tensors(self, tl: TaskList):
         """Upload one task list as device tensors (cacheable)."""
@@ -241,6 +262,11 @@ class FLJob:
         all_reduce it and download once."""
         if idx is None:
             idx = self.eval_tensors(tl)
+        if self.is_module_path:
+            return self.mod_engine.eval_tasks_stacked(
+                params, idx[0], idx[1], idx[2], idx[3], tl.n_tasks,
+                want_mse=want_mse, x_arena=self.arena.x,
+                y_arena=self.arena.y)
         return self.backend.eval_tasks_stacked(
             self.spec, params, self.arena.x, self.arena.y,
             idx[0], idx[1], idx[2], idx[3], tl.n_tasks, want_mse=want_mse)
@@ -250,6 +276,49 @@ class FLJob:
         out = self.run_eval_dev(params, tl, want_mse).cpu().numpy()
         return [out[0], out[1], out[2], out[3] if want_mse else None]
 
+    def ens_vote_eval(self, weights: torch.Tensor, windows, mode: str,
+                      masks: Optional[torch.Tensor] = None):
+        """Weighted-vote ensemble accuracy on one client's windows
+        (AUE/KUE testing) — dispatched per model path."""
+        if self.is_module_path:
+            return self.mod_engine.ens_vote_eval(
+                self.global_params, weights, self.arena.x, self.arena.y,
+                windows, mode=mode, masks=masks)
+        return self.backend.ens_vote_eval(
+            self.spec, self.global_params, weights, self.arena.x,
+            self.arena.y, windows, mode=mode, masks=masks)
+
+    def confusion(self, task_row, task_id, win_off, win_len, n_tasks: int,
+                  n_classes: int, x_mask: Optional[torch.Tensor] = None):
+        if self.is_module_path:
+            return self.mod_engine.confusion_tasks(
+                self.global_params, self.arena.x, self.arena.y, task_row,
+                task_id, win_off, win_len, n_tasks, n_classes,
+                x_mask=x_mask)
+        return self.backend.confusion_tasks(
+            self.spec, self.global_params, self.arena.x, self.arena.y,
+            task_row, task_id, win_off, win_len, n_tasks, n_classes,
+            x_mask=x_mask)
+
+    def forward_flat(self, flat: np.ndarray, x: np.ndarray) -> np.ndarray:
+        """Model output for a flat parameter vector on host data (used by
+        the DriftSurf / MultiModelAcc data-load-time scoring, which runs
+        before the device arena exists). CPU eval, once per iteration."""
+        xt = torch.from_numpy(np.ascontiguousarray(x, dtype=np.float32))
+        ft = torch.from_numpy(np.ascontiguousarray(flat, dtype=np.float32))
+        if self.is_module_path:
+            proto = zoo.create_model(self.cfg.model, self.dataset.class_num,
+                                     self.dataset.feature_num)
+            self.packer.load_into(proto, ft)
+            proto.eval()
+            with torch.no_grad():
+                return proto(xt).numpy()
+        from ..ops import mlp_torch
+        with torch.no_grad():
+            return mlp_torch.forward_logits(
+                self.spec, ft.reshape(1, -1), xt.unsqueeze(0)) \
+                .squeeze(0).numpy()
+
     def train(self, plan: TrainPlan) -> None:
         """All local training of this round in ONE fused launch. On the HIP
         path the kernel also (a) stages each pair's initial weights straight
@@ -257,17 +326,23 @@ class FLJob:
         (b) accumulates the weighted aggregation partial sums on its way out
         — so the whole round's compute is train-kernel + all_reduce +
         apply-kernel."""
-        K, P = self.n_models, self.spec.n_params
+        K, P = self.n_models, self.n_params
         dev = self.device
-        hip = self.backend is not ops.mlp_torch
+        hip = (not self.is_module_path) and self.backend is not ops.mlp_torch
         if self._partial is None or self._partial.shape[0] != K:
             self._partial = torch.zeros(K, P + 1, device=dev)
         if hip:
             self._partial.zero_()
         if plan.rows.size == 0:
             self._partial_fused = hip
-            if not hip:
+            if not hip and not self.is_module_path:
                 self.sync_replicas()
+            return
+        if self.is_module_path:
+            self.mod_engine.train(self.global_params, self.replicas, plan,
+                                  self.opt, self.arena.x, self.arena.y,
+                                  K, x_mask=plan.x_mask)
+            self._partial_fused = False
             return
         off_t = torch.as_tensor(plan.step_off, dtype=torch.int64, device=dev)
         len_t = torch.as_tensor(plan.step_len, dtype=torch.int64, device=dev)
@@ -312,7 +387,7 @@ class FLJob:
         (FedAvgEnsAggregatorSoftCluster.py:151-153). Returns total weights
         per model (device tensor). Models with zero total weight are
         skipped (:167-169)."""
-        K, P = self.n_models, self.spec.n_params
+        K, P = self.n_models, self.n_params
         partial = self._partial
         if not self._partial_fused:
             nW = len(self.owned_workers)

@@ -79,9 +79,20 @@ class CNN_DropOut(nn.Module):
 
 def reinitialize(model: nn.Module) -> None:
     torch.manual_seed(torch_seed)
+    from .resnet import FlatImageModel, ResNet
+    if isinstance(model, (ResNet, FlatImageModel)):
+        # deep models: recurse (the reference instead reloads pretrained
+        # torchvision weights here, utils.py:10-18 — unavailable offline)
+        for layer in model.modules():
+            if hasattr(layer, "reset_parameters"):
+                layer.reset_parameters()
+        return
     for layer in model.children():
         if hasattr(layer, "reset_parameters"):
             layer.reset_parameters()
+
+
+_IMAGE_SHAPES = {3072: (3, 32, 32), 150528: (3, 224, 224)}
 
 
 def create_model(model_name: str, output_dim: int, feature_dim: int) -> nn.Module:
@@ -90,10 +101,13 @@ def create_model(model_name: str, output_dim: int, feature_dim: int) -> nn.Modul
     elif model_name == "fnn":
         model = FeedForwardNN(feature_dim, output_dim, feature_dim * 2)
     elif model_name == "cnn":
-        model = CNN_DropOut()
+        model = CNN_DropOut(only_digits=(output_dim <= 10))
     elif model_name == "resnet":
-        from .resnet import resnet18
-        model = resnet18(num_classes=output_dim if output_dim > 2 else 1000)
+        from .resnet import FlatImageModel, resnet18
+        shape = _IMAGE_SHAPES.get(feature_dim)
+        if shape is None:
+            raise ValueError(f"no image shape for {feature_dim} features")
+        model = FlatImageModel(resnet18(num_classes=output_dim), shape)
     else:
         raise NameError(model_name)
     reinitialize(model)
