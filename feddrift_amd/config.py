@@ -60,6 +60,13 @@ class Config:
     log_dir: str = "."                   # where checkpoints/state files go
     bench_mode: int = 0                  # 1: synthetic steady-state (bench.py)
 
+    # robust aggregation (fedavg_robust / fedml_core robustness equivalent)
+    robust_norm_bound: float = 0.0       # >0: clip client updates to this L2
+    robust_noise: float = 0.0            # >0: Gaussian noise stddev on avg
+    # FedOpt server optimizer ('avg' = plain FedAvg replacement)
+    server_optimizer: str = "avg"
+    server_lr: float = 1.0
+
     def __post_init__(self):
         self.dataset_norm = "MNIST" if self.dataset.lower() == "mnist" else self.dataset
 

@@ -328,7 +328,9 @@ class FLJob:
         apply-kernel."""
         K, P = self.n_models, self.n_params
         dev = self.device
-        hip = (not self.is_module_path) and self.backend is not ops.mlp_torch
+        robust = self.cfg.robust_norm_bound > 0
+        hip = (not self.is_module_path) and self.backend is not ops.mlp_torch \
+            and not robust   # clipping happens between train and aggregate
         if self._partial is None or self._partial.shape[0] != K:
             self._partial = torch.zeros(K, P + 1, device=dev)
         if hip:
@@ -343,6 +345,7 @@ class FLJob:
                                   self.opt, self.arena.x, self.arena.y,
                                   K, x_mask=plan.x_mask)
             self._partial_fused = False
+            self._robust_clip(plan)
             return
         off_t = torch.as_tensor(plan.step_off, dtype=torch.int64, device=dev)
         len_t = torch.as_tensor(plan.step_len, dtype=torch.int64, device=dev)
@@ -376,6 +379,20 @@ class FLJob:
                 self.spec, self.replicas, rows_t, self.arena.x, self.arena.y,
                 off_t, len_t, self.opt, x_mask=plan.x_mask)
             self._partial_fused = False
+            self._robust_clip(plan)
+
+    def _robust_clip(self, plan: TrainPlan) -> None:
+        """Optional defense: clip client updates to an L2 ball before they
+        enter the aggregation (fedavg_robust equivalent,
+        fedml_core robustness/robust_aggregation.py:38)."""
+        if self.cfg.robust_norm_bound <= 0 or plan.rows.size == 0:
+            return
+        from ..comm.robust import robustify_replicas
+        rows_t = torch.as_tensor(plan.rows, dtype=torch.int64,
+                                 device=self.device)
+        mo = torch.as_tensor(plan.rows % self.n_models, device=self.device)
+        robustify_replicas(self.replicas, self.global_params, rows_t, mo,
+                           self.cfg.robust_norm_bound)
 
     def aggregate(self, plan: TrainPlan,
                   model_mask: Optional[np.ndarray] = None) -> torch.Tensor:
@@ -408,16 +425,37 @@ class FLJob:
                 mask_t = torch.as_tensor(
                     np.ascontiguousarray(model_mask, dtype=np.uint8),
                     device=self.device)
-        if self.backend is not ops.mlp_torch:
+        plain = (self.cfg.server_optimizer == "avg" and
+                 self.cfg.robust_noise <= 0)
+        if plain and not self.is_module_path and \
+                self.backend is not ops.mlp_torch:
             from ..ops import mlp_hip
             mlp_hip.apply_aggregate(self.global_params, partial, mask_t)
+            return totals
+        upd = totals > 0
+        if mask_t is not None:
+            upd &= mask_t.bool()
+        newp = partial[:, :P] / totals.clamp(min=1e-30).unsqueeze(1)
+        averaged = torch.where(upd.unsqueeze(1), newp, self.global_params)
+        if self.cfg.robust_noise > 0:
+            from ..comm.robust import add_noise
+            if getattr(self, "_noise_gen", None) is None:
+                # dedicated generator: every rank draws identical noise,
+                # keeping the lockstep-replicated control flow intact
+                self._noise_gen = torch.Generator(device=self.device)
+                self._noise_gen.manual_seed(self.cfg.dummy_arg + 999331)
+            noisy = add_noise(averaged, self.cfg.robust_noise,
+                              self._noise_gen)
+            averaged = torch.where(upd.unsqueeze(1), noisy, averaged)
+        if self.cfg.server_optimizer != "avg":
+            if getattr(self, "_server_opt", None) is None:
+                from .server_opt import ServerOptimizer
+                self._server_opt = ServerOptimizer(
+                    self.global_params, self.cfg.server_optimizer,
+                    self.cfg.server_lr)
+            self._server_opt.step(self.global_params, averaged, upd)
         else:
-            upd = totals > 0
-            if mask_t is not None:
-                upd &= mask_t.bool()
-            newp = partial[:, :P] / totals.clamp(min=1e-30).unsqueeze(1)
-            self.global_params.copy_(
-                torch.where(upd.unsqueeze(1), newp, self.global_params))
+            self.global_params.copy_(averaged)
         return totals
 
     def client_sampling(self, round_idx: int) -> np.ndarray:
