@@ -143,7 +143,7 @@ def test_robust_aggregation_end_to_end(tmp_path, sea_dataset_factory=None):
             ds.store.put(c, t, arr[:, :3], arr[:, 3])
     cfg = Config(model="fnn", dataset="sea", data_dir="/nonexistent",
                  client_num_in_total=4, client_num_per_round=4,
-                 batch_size=100, lr=0.01, epochs=3, comm_round=10,
+                 batch_size=100, lr=0.01, epochs=3, comm_round=30,
                  total_train_iteration=2, curr_train_iteration=1,
                  concept_num=2, concept_drift_algo="single",
                  log_dir=str(tmp_path), report_client=0,
