@@ -659,17 +659,25 @@ class FLJob:
     # the round loop
     # ------------------------------------------------------------------
     def run(self) -> None:
+        from .profiling import PhaseTimer
+        timer = PhaseTimer()
         client_idx = self.client_sampling(0)
         for r in range(self.cfg.comm_round):
-            plan = self.algo.plan(self, r, client_idx)
-            self.train(plan)
-            self.algo.aggregate(self, r, plan, client_idx)
-            self.algo.post_aggregate(self, r)
-            self.algo.test(self, r)
+            with timer.phase("plan"):
+                plan = self.algo.plan(self, r, client_idx)
+            with timer.phase("train"):
+                self.train(plan)
+            with timer.phase("aggregate"):
+                self.algo.aggregate(self, r, plan, client_idx)
+                self.algo.post_aggregate(self, r)
+            with timer.phase("test"):
+                self.algo.test(self, r)
             client_idx = self.client_sampling(r + 1)
         self.save_model_params()
         self.algo.finalize(self)
         self.comm.barrier()
+        if self.comm.is_root:
+            self.logger.set_summary("phase_times", timer.summary())
         self.logger.flush()
 
 
