@@ -24,11 +24,16 @@ ALGOS = [
     ("softcluster", "H_A_C_1_10_0"),    # FedDrift
     ("softcluster", "mmacc_06"),        # FedDrift-Eager
     ("softcluster", "hard"),            # IFCA
+    ("softcluster", "geni"),            # clustering oracle
     ("aue", ""),
+    ("auepc", ""),
     ("kue", ""),
     ("driftsurf", ""),
     ("ada", "win-1_round"),
     ("exp", ""),
+    ("lin", ""),
+    ("mmacc", ""),                      # FedDrift-Eager precursor
+    ("mmgeniex", ""),                   # oracle
     ("single", "win-1"),                # oblivious win-1
     ("single", "all"),                  # oblivious all
 ]
@@ -42,6 +47,7 @@ def main():
     p.add_argument("--iters", type=int, default=10)
     p.add_argument("--seed", type=int, default=0)
     p.add_argument("--algos", default="")   # comma filter, e.g. 'softcluster'
+    p.add_argument("--cp", default="A")      # change-point matrix name
     a = p.parse_args()
 
     comm = Communicator()
@@ -55,7 +61,7 @@ def main():
             if f.endswith(".cp"):
                 shutil.copy(os.path.join(repo_cp, f),
                             os.path.join(a.data_dir, "changepoints", f))
-        generate_data("sea", a.data_dir, a.iters, 10, 0, 100, 0.0, 1, "A")
+        generate_data("sea", a.data_dir, a.iters, 10, 0, 100, 0.0, 1, a.cp)
     comm.barrier()
 
     results = {}
@@ -73,7 +79,7 @@ def main():
             concept_drift_algo=algo,
             concept_drift_algo_arg=arg,
             retrain_data=arg if algo == "single" else "win-1",
-            change_points="A", dummy_arg=a.seed, sample_num=100,
+            change_points=a.cp, dummy_arg=a.seed, sample_num=100,
             log_dir=log_dir, report_client=0)
         t0 = time.time()
         out = run_timeline(cfg, comm)
