@@ -473,35 +473,43 @@ class AueAlgo(AlgoBase):
         if round_idx % 10 == 0 or round_idx > (job.cfg.comm_round - 10):
             self._update_ens_weights(job)
 
+    _test_lists = None
+
+    def _get_test_lists(self, job: FLJob):
+        if self._test_lists is None:
+            C = job.cfg.client_num_in_total
+            clients = range(C) if job.cfg.ci != 1 else range(1)
+            tl_tr = TaskList()
+            tl_te = TaskList()
+            for c in clients:
+                t1 = tl_tr.new_task()
+                t2 = tl_te.new_task()
+                if job.comm.owns_client(c):
+                    if c in job.view_train_ref[0]:
+                        tl_tr.add_windows(t1, 0,
+                                          job.view_train_ref[0][c].windows)
+                    if c in job.test_ref:
+                        tl_te.add_windows(t2, 0, job.test_ref[c].windows)
+            self._test_lists = (tl_tr, job.eval_tensors(tl_tr),
+                                tl_te, job.eval_tensors(tl_te))
+        return self._test_lists
+
     def test(self, job: FLJob, round_idx: int) -> None:
         cfg = job.cfg
         if not (round_idx % cfg.frequency_of_the_test == 0 or
                 round_idx == cfg.comm_round - 1):
             return
-        C = cfg.client_num_in_total
-        clients = range(C) if cfg.ci != 1 else range(1)
+        tl_tr, idx_tr, tl_te, idx_te = self._get_test_lists(job)
         # train: model 0 on view-0 train data (FedAvgEnsAggregatorAue.py:172)
-        tl = TaskList()
-        for c in clients:
-            tid = tl.new_task()
-            if job.comm.owns_client(c) and c in job.view_train_ref[0]:
-                tl.add_windows(tid, 0, job.view_train_ref[0][c].windows)
-        trc, trt, trl, _ = job.run_eval(job.global_params, tl)
-        # test: weighted-vote ensemble (:256-283)
-        tec = np.zeros(len(list(clients)))
-        tet = np.zeros(len(list(clients)))
+        res_tr = job.run_eval_dev(job.global_params, tl_tr, idx=idx_tr)
+        # test: weighted-vote ensemble, batched over all clients (:256-283)
         M = job.n_models
-        for i, c in enumerate(clients):
-            if not job.comm.owns_client(c) or c not in job.test_ref:
-                continue
-            w = self.ens_weights[c] if self.per_client else self.ens_weights
-            cc, tt = job.ens_vote_eval(
-                torch.as_tensor(w[:M], dtype=torch.float32,
-                                device=job.device),
-                job.test_ref[c].windows, mode="hard")
-            tec[i], tet[i] = cc, tt
-        buf = torch.from_numpy(np.stack(
-            [trc, trt, trl, tec, tet])).to(job.device)
+        w = self.ens_weights[:, :M] if self.per_client \
+            else self.ens_weights[:M]
+        res_te = job.ens_vote_multi(
+            torch.as_tensor(w, dtype=torch.float32, device=job.device),
+            tl_te, idx_te, mode="hard")
+        buf = torch.cat([res_tr, res_te])
         job.comm.all_reduce_(buf)
         trc, trt, trl, tec, tet = buf.cpu().numpy()
         job.log_round_stats(round_idx, (trc, trt, trl),
@@ -879,38 +887,45 @@ class KueAlgo(AlgoBase):
             plan.rows % job.n_models, dtype=torch.int64, device=job.device)
         return masks[model_of_row]
 
+    _test_lists = None
+
+    def _get_test_lists(self, job: FLJob):
+        if self._test_lists is None:
+            C = job.cfg.client_num_in_total
+            clients = range(C) if job.cfg.ci != 1 else range(1)
+            tl_tr = TaskList()
+            tl_te = TaskList()
+            for c in clients:
+                t1 = tl_tr.new_task()
+                t2 = tl_te.new_task()
+                if job.comm.owns_client(c):
+                    if c in job.view_train_ref[0]:
+                        tl_tr.add_windows(t1, 0,
+                                          job.view_train_ref[0][c].windows)
+                    if c in job.test_ref:
+                        tl_te.add_windows(t2, 0, job.test_ref[c].windows)
+            self._test_lists = (tl_tr, job.eval_tensors(tl_tr),
+                                tl_te, job.eval_tensors(tl_te))
+        return self._test_lists
+
     def test(self, job: FLJob, round_idx: int) -> None:
         cfg = job.cfg
         if not (round_idx % cfg.frequency_of_the_test == 0 or
                 round_idx == cfg.comm_round - 1):
             return
-        C = cfg.client_num_in_total
-        clients = range(C) if cfg.ci != 1 else range(1)
-        tl = TaskList()
-        for c in clients:
-            tid = tl.new_task()
-            if job.comm.owns_client(c) and c in job.view_train_ref[0]:
-                tl.add_windows(tid, 0, job.view_train_ref[0][c].windows)
-        trc, trt, trl, _ = job.run_eval(job.global_params, tl)
-        # soft-vote ensemble with masks, excluding the worst model
-        # (FedAvgEnsAggregatorKue.py:234-264)
+        tl_tr, idx_tr, tl_te, idx_te = self._get_test_lists(job)
+        res_tr = job.run_eval_dev(job.global_params, tl_tr, idx=idx_tr)
+        # soft-vote ensemble with masks, excluding the worst model, batched
+        # over all clients (FedAvgEnsAggregatorKue.py:234-264)
         masks = self._masks_tensor(job)
         include = np.array([m != self.state.get_worst_idx() and
                             self.ens_weights[m] > 0
                             for m in range(job.n_models)])
         w = np.where(include, self.ens_weights, 0.0)
-        nc = len(list(clients))
-        tec = np.zeros(nc)
-        tet = np.zeros(nc)
-        for i, c in enumerate(clients):
-            if not job.comm.owns_client(c) or c not in job.test_ref:
-                continue
-            cc, tt = job.ens_vote_eval(
-                torch.as_tensor(w, dtype=torch.float32, device=job.device),
-                job.test_ref[c].windows, mode="soft", masks=masks)
-            tec[i], tet[i] = cc, tt
-        buf = torch.from_numpy(np.stack(
-            [trc, trt, trl, tec, tet])).to(job.device)
+        res_te = job.ens_vote_multi(
+            torch.as_tensor(w, dtype=torch.float32, device=job.device),
+            tl_te, idx_te, mode="soft", masks=masks)
+        buf = torch.cat([res_tr, res_te])
         job.comm.all_reduce_(buf)
         trc, trt, trl, tec, tet = buf.cpu().numpy()
         job.log_round_stats(round_idx, (trc, trt, trl),

@@ -288,6 +288,34 @@ class FLJob:
             self.spec, self.global_params, weights, self.arena.x,
             self.arena.y, windows, mode=mode, masks=masks)
 
+    def ens_vote_multi(self, weights: torch.Tensor, tl: TaskList,
+                       idx: torch.Tensor, mode: str,
+                       masks: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Batched many-task ensemble vote ([2, n_tasks] correct/total)."""
+        if not self.is_module_path:
+            from ..ops import mlp_torch
+            return mlp_torch.ens_vote_multi(
+                self.spec, self.global_params, weights, self.arena.x,
+                self.arena.y, idx[1], idx[2], idx[3], tl.n_tasks,
+                mode=mode, masks=masks)
+        out = torch.zeros(2, tl.n_tasks, dtype=torch.float64,
+                          device=self.device)
+        tid = idx[1].cpu().numpy()
+        off = idx[2].cpu().numpy()
+        ln = idx[3].cpu().numpy()
+        for t in range(tl.n_tasks):
+            wins = [(int(o), int(l)) for o, l, ti in zip(off, ln, tid)
+                    if ti == t]
+            if not wins:
+                continue
+            w_t = weights[t] if weights.dim() == 2 else weights
+            c, n = self.mod_engine.ens_vote_eval(
+                self.global_params, w_t, self.arena.x, self.arena.y, wins,
+                mode=mode, masks=masks)
+            out[0, t] = c
+            out[1, t] = n
+        return out
+
     def confusion(self, task_row, task_id, win_off, win_len, n_tasks: int,
                   n_classes: int, x_mask: Optional[torch.Tensor] = None):
         if self.is_module_path:

@@ -337,6 +337,7 @@ void mlp_train_small_kernel(TrainArgs a) {
   __shared__ __attribute__((aligned(16))) float sm_[TP], sv_[TP], svm_[TP];
   __shared__ float msk[TD > 0 ? TD : 1];
   __shared__ int tstep;
+  __shared__ float b1pow, b2pow;   // beta^t maintained incrementally
 
   stage_weights(a, g, row, w);
   if (a.x_mask && tid < TD) msk[tid] = a.x_mask[(int64_t)g * TD + tid];
@@ -347,7 +348,11 @@ void mlp_train_small_kernel(TrainArgs a) {
       sv_[p] = a.v[row * TP + p];
       svm_[p] = a.vmax[row * TP + p];
     }
-    if (tid == 0) tstep = a.t[row];
+    if (tid == 0) {
+      tstep = a.t[row];
+      b1pow = powf(0.9f, (float)tstep);
+      b2pow = powf(0.999f, (float)tstep);
+    }
   }
   __syncthreads();
 
@@ -475,11 +480,15 @@ void mlp_train_small_kernel(TrainArgs a) {
       const float lr_ = a.lr[row];
       for (int p = tid; p < TP; p += THREADS) w[p] -= lr_ * grad[p];
     } else {
-      if (tid == 0) tstep += 1;
+      if (tid == 0) {
+        tstep += 1;
+        b1pow *= 0.9f;
+        b2pow *= 0.999f;
+      }
       __syncthreads();
       const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
-      const float bc1 = 1.f - powf(b1, (float)tstep);
-      const float bc2 = 1.f - powf(b2, (float)tstep);
+      const float bc1 = 1.f - b1pow;
+      const float bc2 = 1.f - b2pow;
       const float lr_ = a.lr[row];
       for (int p = tid; p < TP; p += THREADS) {
         const float gr = grad[p] + a.wd * w[p];

@@ -337,3 +337,52 @@ def eval_tasks_stacked(spec: MLPSpec, params: torch.Tensor,
                             want_mse=want_mse, x_mask=x_mask)
     parts = [c, t, l] + ([m] if m is not None else [])
     return torch.stack(parts)
+
+
+@torch.no_grad()
+def ens_vote_multi(spec: MLPSpec,
+                   params: torch.Tensor,        # [M, P]
+                   weights: torch.Tensor,       # [M] or [T, M]
+                   x_arena: torch.Tensor, y_arena: torch.Tensor,
+                   task_id: torch.Tensor,       # [W]
+                   win_off: torch.Tensor, win_len: torch.Tensor,
+                   n_tasks: int, mode: str = "hard",
+                   masks: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Batched weighted-vote ensemble accuracy over MANY tasks at once
+    (replaces per-client _infer_ens loops: FedAvgEnsAggregatorAue.py:256,
+    AuePc per-client weights :260, Kue soft vote :234). weights may be
+    per-task [T, M] (AUE-PC). Returns [2, n_tasks] = (correct, total)."""
+    dev = params.device
+    out = torch.zeros(2, n_tasks, dtype=torch.float64, device=dev)
+    W = task_id.shape[0]
+    if W == 0:
+        return out
+    M = params.shape[0]
+    bmax = int(win_len.max().item())
+    ar = torch.arange(bmax, device=dev).unsqueeze(0)
+    smask = ar < win_len.unsqueeze(1)
+    idx = win_off.unsqueeze(1) + torch.minimum(
+        ar, (win_len - 1).clamp(min=0).unsqueeze(1))
+    x = x_arena[idx.reshape(-1)].reshape(W, bmax, spec.d)
+    y = y_arena[idx.reshape(-1)].reshape(W, bmax)
+    if weights.dim() == 1:
+        wt = weights.unsqueeze(0).expand(W, M)
+    else:
+        wt = weights[task_id]                       # [W, M]
+    votes = torch.zeros(W, bmax, spec.o, device=dev)
+    for m in range(M):
+        wm = wt[:, m]
+        if float(wm.abs().max()) == 0.0:
+            continue
+        xm = x * masks[m] if masks is not None else x
+        logits = forward_logits(spec, params[m:m + 1].expand(W, -1), xm)
+        if mode == "hard":
+            votes.scatter_add_(
+                2, logits.argmax(-1, keepdim=True),
+                wm.reshape(W, 1, 1).expand(W, bmax, 1).contiguous())
+        else:
+            votes += wm.reshape(W, 1, 1) * torch.softmax(logits, -1)
+    corr = ((votes.argmax(-1) == y) & smask).sum(dim=1).double()
+    out[0].scatter_add_(0, task_id, corr)
+    out[1].scatter_add_(0, task_id, smask.sum(dim=1).double())
+    return out
