@@ -79,8 +79,10 @@ class CNN_DropOut(nn.Module):
 
 def reinitialize(model: nn.Module) -> None:
     torch.manual_seed(torch_seed)
+    from .cv_extra import CifarResNet, DenseNet, MobileNet
     from .resnet import FlatImageModel, ResNet
-    if isinstance(model, (ResNet, FlatImageModel)):
+    if isinstance(model, (ResNet, FlatImageModel, CifarResNet, DenseNet,
+                          MobileNet)):
         # deep models: recurse (the reference instead reloads pretrained
         # torchvision weights here, utils.py:10-18 — unavailable offline)
         for layer in model.modules():
@@ -102,12 +104,26 @@ def create_model(model_name: str, output_dim: int, feature_dim: int) -> nn.Modul
         model = FeedForwardNN(feature_dim, output_dim, feature_dim * 2)
     elif model_name == "cnn":
         model = CNN_DropOut(only_digits=(output_dim <= 10))
-    elif model_name == "resnet":
+    elif model_name in ("resnet", "resnet56", "resnet110", "resnet_gn",
+                        "mobilenet", "densenet"):
+        from .cv_extra import densenet121, MobileNet, resnet56, resnet110
         from .resnet import FlatImageModel, resnet18
         shape = _IMAGE_SHAPES.get(feature_dim)
         if shape is None:
             raise ValueError(f"no image shape for {feature_dim} features")
-        model = FlatImageModel(resnet18(num_classes=output_dim), shape)
+        backbones = {
+            "resnet": lambda: resnet18(num_classes=output_dim),
+            "resnet56": lambda: resnet56(num_classes=output_dim),
+            "resnet110": lambda: resnet110(num_classes=output_dim),
+            "resnet_gn": lambda: resnet56(num_classes=output_dim,
+                                          group_norm=True),
+            "mobilenet": lambda: MobileNet(num_classes=output_dim),
+            "densenet": lambda: densenet121(num_classes=output_dim),
+        }
+        model = FlatImageModel(backbones[model_name](), shape)
+    elif model_name == "rnn":
+        from .rnn import CharLSTM
+        model = CharLSTM(vocab_size=max(output_dim, 90))
     else:
         raise NameError(model_name)
     reinitialize(model)
