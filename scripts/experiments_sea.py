@@ -21,7 +21,9 @@ from feddrift_amd.data.generators import generate_data
 from feddrift_amd.engine.timeline import run_timeline
 
 ALGOS = [
-    ("softcluster", "H_A_C_1_10_0"),    # FedDrift
+    ("softcluster", "H_A_C_1_10_0"),    # FedDrift (delta=0.10)
+    ("softcluster", "H_A_C_1_06_0"),    # FedDrift (delta=0.06)
+    ("softcluster", "H_A_F_1_06_0"),    # FedDrift (per-client init)
     ("softcluster", "mmacc_06"),        # FedDrift-Eager
     ("softcluster", "hard"),            # IFCA
     ("softcluster", "geni"),            # clustering oracle
