@@ -73,11 +73,14 @@ def main():
         name = f"{algo}:{arg}" if arg else algo
         log_dir = os.path.join(a.data_dir, "run_" + name.replace(":", "_"))
         os.makedirs(log_dir, exist_ok=True)
+        # the F (per-client-init) FedDrift variant starts one model per
+        # client, so the ensemble cap equals the client count
+        k_cap = 10 if "_F_" in arg else 4
         cfg = Config(
             model="fnn", dataset="sea", data_dir=a.data_dir,
             client_num_in_total=10, client_num_per_round=10,
             batch_size=500, lr=0.01, epochs=5, comm_round=a.rounds,
-            total_train_iteration=a.iters, concept_num=4,
+            total_train_iteration=a.iters, concept_num=k_cap,
             concept_drift_algo=algo,
             concept_drift_algo_arg=arg,
             retrain_data=arg if algo == "single" else "win-1",
