@@ -152,6 +152,11 @@ class SoftClusterState:
         for c in range(self.client_num):
             hooks.log_client("Plurality/CL-{}", c,
                              self.get_test_model_idx(curr_iter, c), round_idx)
+            if "softmax" in self.cluster_alg:
+                hooks.log_client(
+                    "Weight-All/CL-{}", c,
+                    np.array2string(self.train_data_weights[curr_iter][:, c]),
+                    round_idx)
 
     def cluster_hard(self, acc_matrix: np.ndarray, curr_iter: int) -> None:
         self.train_data_weights[curr_iter] = np.zeros(

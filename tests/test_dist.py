@@ -80,3 +80,42 @@ def test_world2_matches_world1(tmp_path):
     # and later iterations within a small accuracy tolerance.
     assert abs(r1[0] - r2[0]) < 1e-6, (r1, r2)
     assert np.allclose(r1, r2, atol=0.02), (r1, r2)
+
+
+_WORKER_AUE = r"""
+import json, os, sys
+sys.path.insert(0, {repo!r})
+import numpy as np
+from feddrift_amd.comm import Communicator
+from feddrift_amd.config import Config
+from feddrift_amd.engine.timeline import run_timeline
+
+cfg = Config(model="fnn", dataset="sea", data_dir={data!r},
+             client_num_in_total=6, client_num_per_round=6,
+             batch_size=300, lr=0.01, epochs=5, comm_round=6,
+             total_train_iteration=3, concept_num=2, ensemble_window=3,
+             concept_drift_algo="aue",
+             change_points="T", dummy_arg=0, log_dir={log!r},
+             report_client=0)
+comm = Communicator()
+out = run_timeline(cfg, comm)
+if comm.is_root:
+    with open(os.path.join({log!r}, "result.json"), "w") as f:
+        json.dump(out["per_iteration_test_acc"], f)
+"""
+
+
+def test_world2_matches_world1_aue(tmp_path):
+    """AUE path (per-model views, ensemble-vote testing, MSE allreduce)
+    under client sharding."""
+    global _WORKER
+    data = _write_data(tmp_path)
+    saved = _WORKER
+    try:
+        globals()["_WORKER"] = _WORKER_AUE
+        r1 = _run(1, data, str(tmp_path / "a1"), 29614)
+        r2 = _run(2, data, str(tmp_path / "a2"), 29615)
+    finally:
+        globals()["_WORKER"] = saved
+    assert abs(r1[0] - r2[0]) < 1e-6, (r1, r2)
+    assert np.allclose(r1, r2, atol=0.02), (r1, r2)
