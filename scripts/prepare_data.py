@@ -33,6 +33,21 @@ def main():
     a = p.parse_args()
 
     np.random.seed(a.dummy_arg)
+
+    # seed the canonical change-point matrices (A..F, W..Z, R0..R9) into the
+    # data dir, like the reference's data/changepoints checkout
+    import shutil
+    repo_cp = os.path.join(os.path.dirname(os.path.abspath(__file__)), "..",
+                           "data", "changepoints")
+    dst_cp = os.path.join(a.data_dir, "changepoints")
+    os.makedirs(dst_cp, exist_ok=True)
+    if os.path.isdir(repo_cp):
+        for f in os.listdir(repo_cp):
+            if f.endswith(".cp") and not os.path.exists(
+                    os.path.join(dst_cp, f)):
+                shutil.copy(os.path.join(repo_cp, f),
+                            os.path.join(dst_cp, f))
+
     generate_data(a.dataset, a.data_dir, a.train_iteration,
                   a.client_num_in_total, a.drift_together, a.sample_num,
                   a.noise_prob, a.time_stretch, a.change_points)
