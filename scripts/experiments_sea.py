@@ -50,6 +50,7 @@ def main():
     p.add_argument("--seed", type=int, default=0)
     p.add_argument("--algos", default="")   # comma filter, e.g. 'softcluster'
     p.add_argument("--cp", default="A")      # change-point matrix name
+    p.add_argument("--dataset", default="sea")  # sea | sine | circle
     a = p.parse_args()
 
     comm = Communicator()
@@ -63,7 +64,7 @@ def main():
             if f.endswith(".cp"):
                 shutil.copy(os.path.join(repo_cp, f),
                             os.path.join(a.data_dir, "changepoints", f))
-        generate_data("sea", a.data_dir, a.iters, 10, 0, 100, 0.0, 1, a.cp)
+        generate_data(a.dataset, a.data_dir, a.iters, 10, 0, 100, 0.0, 1, a.cp)
     comm.barrier()
 
     results = {}
@@ -77,7 +78,7 @@ def main():
         # client, so the ensemble cap equals the client count
         k_cap = 10 if "_F_" in arg else 4
         cfg = Config(
-            model="fnn", dataset="sea", data_dir=a.data_dir,
+            model="fnn", dataset=a.dataset, data_dir=a.data_dir,
             client_num_in_total=10, client_num_per_round=10,
             batch_size=500, lr=0.01, epochs=5, comm_round=a.rounds,
             total_train_iteration=a.iters, concept_num=k_cap,
