@@ -59,10 +59,8 @@ struct TrainArgs {
   float* __restrict__ vmax;
   int* __restrict__ t;               // [R]
   const float* __restrict__ lr;      // [R]
-  float* __restrict__ grad_ws;       // [G, P] workspace (big-model path)
   float wd;
   int E, D, H, O, P, kind, opt, BC;
-  int big;                           // 1: weights+grads in HBM, not LDS
 };
 
 __device__ __forceinline__ float block_reduce_sum(float val, float* scratch) {
@@ -124,9 +122,10 @@ __device__ __forceinline__ void opt_update(const TrainArgs& a, int64_t row,
   }
 }
 
-__device__ __forceinline__ void accumulate_partial(const TrainArgs& a,
-                                                   int g, const float* w) {
+__device__ __forceinline__ void write_back(const TrainArgs& a, int g,
+                                           int64_t row, const float* w) {
   const int tid = threadIdx.x;
+  for (int p = tid; p < a.P; p += THREADS) a.params[row * a.P + p] = w[p];
   if (a.partial && a.sample_w) {
     const float sw = a.sample_w[g];
     if (sw > 0.f) {
@@ -138,13 +137,6 @@ __device__ __forceinline__ void accumulate_partial(const TrainArgs& a,
   }
 }
 
-__device__ __forceinline__ void write_back(const TrainArgs& a, int g,
-                                           int64_t row, const float* w) {
-  const int tid = threadIdx.x;
-  for (int p = tid; p < a.P; p += THREADS) a.params[row * a.P + p] = w[p];
-  accumulate_partial(a, g, w);
-}
-
 extern "C" __global__ __launch_bounds__(THREADS)
 void mlp_train_kernel(TrainArgs a) {
   const int g = blockIdx.x;
@@ -152,35 +144,17 @@ void mlp_train_kernel(TrainArgs a) {
   const int tid = threadIdx.x;
 
   extern __shared__ __attribute__((aligned(16))) float lds[];
-  float* w;
-  float* grad;
-  float* chunk;
-  if (a.big) {
-    // big-model path (e.g. FEMNIST-scale towers): weights live in the
-    // replica row and gradients in an HBM workspace (L2-served: the model
-    // row is re-read by every phase, the grad row is block-private);
-    // LDS holds only the per-chunk activations
-    w = a.params + row * a.P;
-    grad = a.grad_ws + (int64_t)g * a.P;
-    chunk = lds;
-    if (a.in_params) {
-      const int64_t src = (int64_t)a.model_of[g] * a.P;
-      for (int p = tid; p < a.P; p += THREADS) w[p] = a.in_params[src + p];
-    }
-    __syncthreads();
-  } else {
-    w = lds;                   // [P]
-    grad = w + a.P;            // [P]
-    chunk = grad + a.P;
-    // stage weights: straight from the GLOBAL model row (replaces the
-    // reference's server->client model broadcast and the engine's
-    // replica-sync copy — the round starts here)
-    stage_weights(a, g, row, w);
-  }
-  float* xb = chunk;                 // [BC, D]
+  float* w = lds;              // [P]
+  float* grad = w + a.P;       // [P]
+  float* xb = grad + a.P;      // [BC, D]
   float* act = xb + a.BC * a.D;      // [BC, H] (fnn activations)
   float* dza = act + (a.kind == KIND_FNN ? a.BC * a.H : 0);  // [BC, H] dz1
   float* dzo = dza + (a.kind == KIND_FNN ? a.BC * a.H : 0);  // [BC, O]
+
+  // stage weights: straight from the GLOBAL model row (replaces the
+  // reference's server->client model broadcast and the engine's
+  // replica-sync copy — the round starts here)
+  stage_weights(a, g, row, w);
 
   const int HD = a.H * a.D;
   const int OH = a.O * a.H;
@@ -332,10 +306,7 @@ void mlp_train_kernel(TrainArgs a) {
     __syncthreads();
   }
 
-  if (a.big)
-    accumulate_partial(a, g, w);   // w IS the replica row already
-  else
-    write_back(a, g, row, w);
+  write_back(a, g, row, w);
 }
 
 // ---------------------------------------------------------------------------
@@ -598,7 +569,7 @@ struct EvalArgs {
   double* __restrict__ total;           // [T]
   double* __restrict__ loss;            // [T]
   double* __restrict__ mse;             // [T] or nullptr
-  int D, H, O, P, kind, big;
+  int D, H, O, P, kind;
 };
 
 extern "C" __global__ __launch_bounds__(THREADS)
@@ -611,21 +582,11 @@ void mlp_eval_kernel(EvalArgs a) {
   const int n = (int)a.len[wdx];
 
   extern __shared__ __attribute__((aligned(16))) float lds[];
-  const float* w;
-  float* red;
-  if (a.big) {
-    // big models: weights read straight from HBM (L2-hot: the same model
-    // row is shared by many windows)
-    w = a.params + row * a.P;
-    red = lds;
-  } else {
-    float* wl = lds;           // [P]
-    red = wl + a.P;            // [8] reduction scratch
-    for (int p = tid; p < a.P; p += THREADS)
-      wl[p] = a.params[row * a.P + p];
-    __syncthreads();
-    w = wl;
-  }
+  float* w = lds;              // [P] (P fits: MLP family)
+  float* red = w + a.P;        // [8] reduction scratch
+
+  for (int p = tid; p < a.P; p += THREADS) w[p] = a.params[row * a.P + p];
+  __syncthreads();
 
   const int HD = a.H * a.D;
   const int OH = a.O * a.H;
@@ -718,27 +679,14 @@ void train_fused_hip(torch::Tensor params, torch::Tensor rows,
   const int E = step_off.size(1);
   const bool adam = m.has_value();
 
-  // chunk size: fit 2P + BC*(D + 2H + O | D + O) in the LDS budget;
-  // models beyond it keep weights/grads in HBM (big path) and use the
-  // whole budget for the chunk
+  // chunk size: fit 2P + BC*(D + 2H + O | D + O) in the LDS budget
   const int per_sample = (kind == KIND_FNN) ? (D + 2 * H + O) : (D + O);
-  int lds_model = 2 * P;
-  int big = 0;
-  int BC = (LDS_BUDGET_FLOATS - lds_model) / per_sample;
-  if (BC < 16) {
-    big = 1;
-    lds_model = 0;
-    BC = LDS_BUDGET_FLOATS / per_sample;
-  }
-  TORCH_CHECK(BC >= 1, "per-sample state too large for LDS");
+  int BC = (LDS_BUDGET_FLOATS - 2 * P) / per_sample;
+  TORCH_CHECK(BC >= 1, "model too large for LDS-resident training path");
   BC = std::min<int>(BC, 512);
 
-  torch::Tensor grad_ws;
-  if (big)
-    grad_ws = torch::empty({(int64_t)G, (int64_t)P}, params.options());
-
   const size_t lds_bytes =
-      (size_t)(lds_model + (size_t)BC * per_sample) * sizeof(float);
+      (size_t)(2 * P + (size_t)BC * per_sample) * sizeof(float);
 
   TrainArgs args;
   args.params = params.data_ptr<float>();
@@ -759,12 +707,10 @@ void train_fused_hip(torch::Tensor params, torch::Tensor rows,
   args.vmax = adam ? vmax->data_ptr<float>() : nullptr;
   args.t = adam ? t->data_ptr<int>() : nullptr;
   args.lr = lr.data_ptr<float>();
-  args.grad_ws = big ? grad_ws.data_ptr<float>() : nullptr;
   args.wd = (float)wd;
   args.E = E; args.D = (int)D; args.H = (int)H; args.O = (int)O;
   args.P = P; args.kind = (int)kind; args.opt = adam ? OPT_ADAM : OPT_SGD;
   args.BC = BC;
-  args.big = big;
 
   if (!launch_small(args, G, c10::hip::getCurrentHIPStream())) {
     hipLaunchKernelGGL(mlp_train_kernel, dim3(G), dim3(THREADS), lds_bytes,
@@ -804,9 +750,9 @@ torch::Tensor eval_tasks_hip(
   args.mse = want_mse ? base + 3 * n_tasks : nullptr;
   args.D = (int)D; args.H = (int)H; args.O = (int)O; args.P = P;
   args.kind = (int)kind;
-  args.big = ((size_t)(P + 8) * sizeof(float) > 150 * 1024) ? 1 : 0;
 
-  const size_t lds_bytes = args.big ? 64 : (size_t)(P + 8) * sizeof(float);
+  const size_t lds_bytes = (size_t)(P + 8) * sizeof(float);
+  TORCH_CHECK(lds_bytes <= 150 * 1024, "model too large for eval LDS path");
   hipLaunchKernelGGL(mlp_eval_kernel, dim3(W), dim3(THREADS), lds_bytes,
                      c10::hip::getCurrentHIPStream(), args);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "mlp_eval_kernel launch");

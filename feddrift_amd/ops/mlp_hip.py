@@ -23,11 +23,21 @@ confusion_tasks = mlp_torch.confusion_tasks
 
 _KIND = {"fnn": 0, "lr": 1}
 
-# Models whose weights+grads fit the 150 KiB LDS budget run fully
-# LDS-resident; larger ones (FEMNIST-scale towers) use the kernels'
-# big-model path (weights/grads in HBM, L2-served) — same launch, chosen
-# by the C++ host wrapper.
+# The train kernel keeps weights+grads (and per-sample chunk state) in LDS;
+# the eval kernel keeps weights in LDS. Models beyond these sizes run on
+# torch GPU ops (rocBLAS bmm) instead — a documented size-based dispatch,
+# not a silent fallback (the flagship drift models are far below the caps).
 LDS_BUDGET_FLOATS = 150 * 1024 // 4
+TRAIN_MAX_P = (LDS_BUDGET_FLOATS - 4096) // 2
+EVAL_MAX_P = LDS_BUDGET_FLOATS - 64
+
+
+def _fits_train(spec: MLPSpec) -> bool:
+    return spec.n_params <= TRAIN_MAX_P
+
+
+def _fits_eval(spec: MLPSpec) -> bool:
+    return spec.n_params <= EVAL_MAX_P
 
 
 def train_fused(spec: MLPSpec, params_all: torch.Tensor, rows: torch.Tensor,
@@ -42,6 +52,19 @@ def train_fused(spec: MLPSpec, params_all: torch.Tensor, rows: torch.Tensor,
     model rows (fuses the round's model broadcast into the launch);
     sample_w/partial: fused weighted aggregation partial sums."""
     if rows.numel() == 0:
+        return
+    if not _fits_train(spec):
+        if in_params is not None:
+            params_all[rows] = in_params[model_of.long()]
+        mlp_torch.train_fused(spec, params_all, rows, x_arena, y_arena,
+                              step_off, step_len, opt, x_mask=x_mask)
+        if partial is not None and sample_w is not None:
+            P = params_all.shape[1]
+            mo = model_of.long()
+            partial.index_add_(
+                0, mo,
+                torch.cat([sample_w.unsqueeze(1) * params_all[rows],
+                           sample_w.unsqueeze(1)], dim=1))
         return
     mod = hip_loader.load()
     adam = opt["kind"] == "adam"
@@ -86,6 +109,10 @@ def eval_tasks_stacked(spec: MLPSpec, params: torch.Tensor,
                        x_mask: Optional[torch.Tensor] = None) -> torch.Tensor:
     """[3 or 4, n_tasks] float64: correct/total/loss(/mse) rows in ONE
     contiguous buffer (ready for all_reduce, no stack)."""
+    if not _fits_eval(spec):
+        return mlp_torch.eval_tasks_stacked(
+            spec, params, x_arena, y_arena, task_row, task_id, win_off,
+            win_len, n_tasks, want_mse=want_mse, x_mask=x_mask)
     mod = hip_loader.load()
     return mod.eval_tasks(
         params.contiguous(), x_arena, y_arena,
