@@ -501,8 +501,7 @@ class AueAlgo(AlgoBase):
             return
         tl_tr, idx_tr, tl_te, idx_te = self._get_test_lists(job)
         # train: model 0 on view-0 train data (FedAvgEnsAggregatorAue.py:172)
-        res_tr = job.run_eval_dev(job.global_params, tl_tr, idx=idx_tr,
-                                  graph_key=(type(self).__name__, "test_tr"))
+        res_tr = job.run_eval_dev(job.global_params, tl_tr, idx=idx_tr)
         # test: weighted-vote ensemble, batched over all clients (:256-283)
         M = job.n_models
         w = self.ens_weights[:, :M] if self.per_client \
@@ -915,8 +914,7 @@ class KueAlgo(AlgoBase):
                 round_idx == cfg.comm_round - 1):
             return
         tl_tr, idx_tr, tl_te, idx_te = self._get_test_lists(job)
-        res_tr = job.run_eval_dev(job.global_params, tl_tr, idx=idx_tr,
-                                  graph_key=(type(self).__name__, "test_tr"))
+        res_tr = job.run_eval_dev(job.global_params, tl_tr, idx=idx_tr)
         # soft-vote ensemble with masks, excluding the worst model, batched
         # over all clients (FedAvgEnsAggregatorKue.py:234-264)
         masks = self._masks_tensor(job)

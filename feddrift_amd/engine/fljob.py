@@ -228,13 +228,6 @@ class FLJob:
         self._eval_cache: Dict = {}
         self._partial: Optional[torch.Tensor] = None
         self._partial_fused = False
-        # hipGraph capture of the steady-state per-round device work
-        # (train launch / eval sweep): replays replace the per-op
-        # host-dispatch gaps. Keyed by the cached plan template / eval
-        # task list; disabled automatically if capture fails.
-        self._graphs: Dict = {}
-        self._graph_ok = (self.device.type == "cuda" and
-                          cfg.use_hip_kernels != "never")
         self.algo.init_iteration(self)
 
     # ------------------------------------------------------------------
@@ -263,8 +256,7 @@ class FLJob:
 
     def run_eval_dev(self, params: torch.Tensor, tl: TaskList,
                      want_mse: bool = False,
-                     idx: Optional[torch.Tensor] = None,
-                     graph_key=None) -> torch.Tensor:
+                     idx: Optional[torch.Tensor] = None) -> torch.Tensor:
         """Batched eval sweep; returns a stacked DEVICE tensor
         [correct; total; loss (; mse)] x n_tasks (float64) — callers
         all_reduce it and download once."""
@@ -275,39 +267,9 @@ class FLJob:
                 params, idx[0], idx[1], idx[2], idx[3], tl.n_tasks,
                 want_mse=want_mse, x_arena=self.arena.x,
                 y_arena=self.arena.y)
-
-        def _sweep():
-            return self.backend.eval_tasks_stacked(
-                self.spec, params, self.arena.x, self.arena.y,
-                idx[0], idx[1], idx[2], idx[3], tl.n_tasks,
-                want_mse=want_mse)
-
-        # graph-replay the sweep when the caller names a stable key for a
-        # CACHED task list (params must be the global bank, whose storage
-        # is stable); the graph entry pins the idx tensors alive
-        if (graph_key is None or not self._graph_ok or
-                params.data_ptr() != self.global_params.data_ptr()):
-            return _sweep()
-        gkey = ("eval", graph_key, want_mse)
-        gr = self._graphs.get(gkey)
-        if gr is None:
-            warm = self._graphs.get(("warm",) + gkey, 0)
-            if warm < 2:
-                self._graphs[("warm",) + gkey] = warm + 1
-                return _sweep()
-            try:
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
-                    out = _sweep()
-                self._graphs[gkey] = (g, out, idx)
-                g.replay()
-                return out
-            except Exception:  # noqa: BLE001
-                self._graph_ok = False
-                return _sweep()
-        g, out, _pinned = gr
-        g.replay()
-        return out
+        return self.backend.eval_tasks_stacked(
+            self.spec, params, self.arena.x, self.arena.y,
+            idx[0], idx[1], idx[2], idx[3], tl.n_tasks, want_mse=want_mse)
 
     def run_eval(self, params: torch.Tensor, tl: TaskList,
                  want_mse: bool = False):
@@ -413,12 +375,8 @@ class FLJob:
             self._partial_fused = False
             self._robust_clip(plan)
             return
-        off_t = len_t = None
-        if not hip or not self._graph_ok:
-            off_t = torch.as_tensor(plan.step_off, dtype=torch.int64,
-                                    device=dev)
-            len_t = torch.as_tensor(plan.step_len, dtype=torch.int64,
-                                    device=dev)
+        off_t = torch.as_tensor(plan.step_off, dtype=torch.int64, device=dev)
+        len_t = torch.as_tensor(plan.step_len, dtype=torch.int64, device=dev)
         if hip:
             # per-template constants (rows / model / aggregation weights)
             # are uploaded once and reused every round
@@ -435,52 +393,12 @@ class FLJob:
                 if tmpl is not None:
                     tmpl._dev = cache
             rows_t, mo_t, sw_t = cache
-
-            def _launch(off_d, len_d):
-                self._partial.zero_()
-                self.backend.train_fused(
-                    self.spec, self.replicas, rows_t, self.arena.x,
-                    self.arena.y, off_d, len_d, self.opt,
-                    x_mask=plan.x_mask, in_params=self.global_params,
-                    model_of=mo_t, sample_w=sw_t, partial=self._partial)
-
-            gkey = ("train", id(tmpl), plan.step_off.shape)
-            if not self._graph_ok or tmpl is None:
-                if off_t is None:
-                    off_t = torch.as_tensor(plan.step_off,
-                                            dtype=torch.int64, device=dev)
-                    len_t = torch.as_tensor(plan.step_len,
-                                            dtype=torch.int64, device=dev)
-                _launch(off_t, len_t)
-            else:
-                gr = self._graphs.get(gkey)
-                if gr is None:
-                    off_t = torch.as_tensor(plan.step_off,
-                                            dtype=torch.int64, device=dev)
-                    len_t = torch.as_tensor(plan.step_len,
-                                            dtype=torch.int64, device=dev)
-                    # two eager warmups, then capture (hipGraph via torch)
-                    warm = self._graphs.get(("warm",) + gkey, 0)
-                    if warm < 2:
-                        self._graphs[("warm",) + gkey] = warm + 1
-                        _launch(off_t, len_t)
-                    else:
-                        off_buf = off_t.clone()
-                        len_buf = len_t.clone()
-                        try:
-                            g = torch.cuda.CUDAGraph()
-                            with torch.cuda.graph(g):
-                                _launch(off_buf, len_buf)
-                            self._graphs[gkey] = (g, off_buf, len_buf)
-                            g.replay()
-                        except Exception:  # noqa: BLE001
-                            self._graph_ok = False
-                            _launch(off_t, len_t)
-                else:
-                    g, off_buf, len_buf = gr
-                    off_buf.copy_(torch.from_numpy(plan.step_off))
-                    len_buf.copy_(torch.from_numpy(plan.step_len))
-                    g.replay()
+            self.backend.train_fused(
+                self.spec, self.replicas, rows_t, self.arena.x, self.arena.y,
+                off_t, len_t, self.opt, x_mask=plan.x_mask,
+                in_params=self.global_params,
+                model_of=mo_t, sample_w=sw_t,
+                partial=self._partial)
             self._partial_fused = True
         else:
             self.sync_replicas()
@@ -702,8 +620,7 @@ class FLJob:
             cached = (tl, self.eval_tensors(tl))
             self._eval_cache[key] = cached
         tl, idx = cached
-        res = self.run_eval_dev(self.global_params, tl, idx=idx,
-                                graph_key=("client_eval", key))
+        res = self.run_eval_dev(self.global_params, tl, idx=idx)
         self.comm.all_reduce_(res)
         correct, total, loss = res.cpu().numpy()
         n = len(list(clients))
