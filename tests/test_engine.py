@@ -110,9 +110,15 @@ def test_iteration_resume_matches_inprocess(sea_dir, tmp_path):
 
 
 @pytest.mark.parametrize("algo,arg", [
-    ("aue", ""), ("kue", ""), ("driftsurf", ""), ("ada", "win-1_round"),
-    ("exp", ""), ("mmacc", ""), ("softcluster", "hard"),
-    ("softcluster", "cfl_0.1_win-1"), ("single", "")])
+    ("aue", ""), ("auepc", ""), ("kue", ""), ("driftsurf", ""),
+    ("ada", "win-1_round"), ("ada", "all_iter"),
+    ("exp", ""), ("lin", ""), ("mmacc", ""), ("mmgeni", ""),
+    ("mmgeniex", ""), ("softcluster", "hard"),
+    ("softcluster", "hard-r"), ("softcluster", "softmax_0"),
+    ("softcluster", "gmm"), ("softcluster", "geni"),
+    ("softcluster", "cfl_0.1_win-1"), ("softcluster", "cfl_0.1_all"),
+    ("softclusterwin-1", "H_A_C_1_10_0"),
+    ("softclusterreset", "softmax_0"), ("single", "")])
 def test_algorithms_run_short(sea_dir, tmp_path, algo, arg):
     cfg = _cfg(sea_dir, tmp_path, algo, arg, comm_round=4,
                total_train_iteration=2)
