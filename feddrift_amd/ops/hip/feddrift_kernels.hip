@@ -21,6 +21,17 @@
 //
 // fp32 everywhere (the reference trains fp32; dtype parity is required for
 // the accuracy metric). Wavefront = 64; block = 256 threads.
+//
+// Why no MFMA here (a decision, not an omission): the drift-path models
+// are 3->6->2 / 2->4->2 MLPs — per pair the "GEMMs" are [batch,3]x[3,6]
+// and the K=batch gradient reductions produce [6,3] tiles. An
+// mfma_f32_16x16x4_f32 tile would be >90% padding on M=6/N=3 fragments
+// and the f32 MFMA rate equals the f32 VALU rate on gfx950, so matrix
+// cores buy nothing at these shapes; PMC shows the kernel is
+// LATENCY-bound (SQ_WAIT dominated), not FLOP-bound (profiles/README.md).
+// Where genuinely MFMA-shaped work exists — CNN/ResNet convolutions and
+// FEMNIST-scale tower GEMMs — the engine uses MIOpen / rocBLAS batched
+// GEMMs, which are the MFMA paths for library-shaped matmuls.
 
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
