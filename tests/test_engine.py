@@ -132,3 +132,31 @@ def test_single_model_win1_beats_chance(sea_dir, tmp_path):
                comm_round=15)
     out = run_timeline(cfg)
     assert out["avg_test_acc"] > 0.7
+
+
+def test_serving_from_checkpoint(sea_dir, tmp_path):
+    """DriftModelServer: load model_params.pt + sc_state.pkl and serve
+    per-client predictions routed to the client's cluster model."""
+    from feddrift_amd.engine.serve import DriftModelServer
+
+    cfg = _cfg(sea_dir, tmp_path, "softcluster", "H_A_C_1_10_0",
+               comm_round=15, total_train_iteration=3)
+    run_timeline(cfg)
+    srv = DriftModelServer(cfg, str(tmp_path))
+    x = np.random.default_rng(0).uniform(0, 10, size=(50, 3))
+    for c in range(cfg.client_num_in_total):
+        pred = srv.predict(c, x)
+        assert pred.shape == (50,)
+        assert set(np.unique(pred)).issubset({0, 1})
+    # routing should come from the drift state (valid model indices)
+    assert srv.route.shape == (cfg.client_num_in_total,)
+    assert srv.route.max() < srv.n_models
+    # served predictions broadly match the SEA rule on easy points
+    x_easy0 = np.column_stack([np.full(20, 5.0), np.full(20, 1.0),
+                               np.full(20, 1.0)])   # f2+f3=2 -> label 0
+    x_easy1 = np.column_stack([np.full(20, 5.0), np.full(20, 9.0),
+                               np.full(20, 9.0)])   # f2+f3=18 -> label 1
+    acc = np.mean([np.mean(srv.predict(c, x_easy0) == 0) +
+                   np.mean(srv.predict(c, x_easy1) == 1)
+                   for c in range(6)]) / 2
+    assert acc > 0.8
