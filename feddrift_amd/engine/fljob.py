@@ -160,8 +160,17 @@ class FLJob:
                                       self.dataset.feature_num)
             self.packer = ModulePacker(proto0)
             self.spec = None
-            from ..ops.module_engine import ModuleEngine
-            self.mod_engine = ModuleEngine(proto0, self.packer, self.device)
+            from ..ops.module_vmap import VmapEngine, vmap_compatible
+            if vmap_compatible(proto0):
+                # buffer-free modules (CNN_DropOut): all pairs train in one
+                # vmapped batched-autograd step (grouped MIOpen convs)
+                self.mod_engine = VmapEngine(proto0, self.packer,
+                                             self.device)
+            else:
+                # BN models (ResNet): stateful running stats -> sequential
+                from ..ops.module_engine import ModuleEngine
+                self.mod_engine = ModuleEngine(proto0, self.packer,
+                                               self.device)
         else:
             self.spec = packed.spec_for(cfg.model, self.dataset.feature_num,
                                         self.dataset.class_num)

@@ -113,3 +113,64 @@ def test_resnet_forward_backward():
     assert set(sd.keys()) == set(m.state_dict().keys())
     # buffers included (BN running stats), params subset marked
     assert p.n_train_params < p.n_params
+
+
+def test_vmap_engine_matches_sequential():
+    """Batched vmap training must match the sequential module engine for a
+    deterministic (dropout-free) module."""
+    from feddrift_amd.ops.module_vmap import VmapEngine, vmap_compatible
+    torch.manual_seed(3)
+    model = FeedForwardNN(4, 3, 8)
+    assert vmap_compatible(model)
+    packer = ModulePacker(model)
+    P = packer.n_params
+    dev = torch.device("cpu")
+
+    n, G, E, K = 240, 4, 3, 2
+    x = torch.rand(n, 4) * 5
+    y = torch.randint(0, 3, (n,))
+    glob = torch.stack([packer.flatten(FeedForwardNN(4, 3, 8).state_dict())
+                        for _ in range(K)])
+    offs = np.array([[0, 60, 120], [60, 0, 180], [120, 180, 0],
+                     [180, 60, 120]])
+    lens = np.full((G, E), 60)
+    lens[2, 1] = 0   # a skipped step
+    plan = TrainPlan(np.arange(G), offs, lens, np.ones((2, K)))
+
+    eng_v = VmapEngine(FeedForwardNN(4, 3, 8), ModulePacker(model), dev)
+    reps_v = torch.zeros(G, P)
+    opt_v = eng_v.make_opt_state("adam", G, 0.01, 0.001)
+    eng_v.train(glob.clone(), reps_v, plan, opt_v, x, y, n_models=K)
+
+    eng_s = ModuleEngine(FeedForwardNN(4, 3, 8), ModulePacker(model), dev)
+    reps_s = torch.zeros(G, P)
+    opt_s = eng_s.make_opt_state("adam", G, 0.01, 0.001)
+    # sequential engine's opt covers only trainable params = all here
+    eng_s.train(glob.clone(), reps_s, plan, opt_s, x, y, n_models=K)
+
+    assert torch.allclose(reps_v, reps_s, atol=1e-5), \
+        (reps_v - reps_s).abs().max()
+    assert torch.equal(opt_v["t"], opt_s["t"])
+
+
+def test_vmap_engine_eval_matches():
+    from feddrift_amd.ops.module_vmap import VmapEngine
+    torch.manual_seed(4)
+    model = FeedForwardNN(3, 2, 6)
+    packer = ModulePacker(model)
+    dev = torch.device("cpu")
+    eng_v = VmapEngine(FeedForwardNN(3, 2, 6), packer, dev)
+    eng_s = ModuleEngine(FeedForwardNN(3, 2, 6), packer, dev)
+    params = torch.stack([packer.flatten(FeedForwardNN(3, 2, 6).state_dict())
+                          for _ in range(3)])
+    x = torch.rand(400, 3) * 8
+    y = (x[:, 1] + x[:, 2] > 8).long()
+    tr = torch.tensor([0, 1, 2, 0])
+    ti = torch.tensor([0, 1, 2, 1])
+    wo = torch.tensor([0, 100, 200, 300])
+    wl = torch.tensor([100, 100, 100, 100])
+    a = eng_v.eval_tasks_stacked(params, tr, ti, wo, wl, 3, want_mse=True,
+                                 x_arena=x, y_arena=y)
+    b = eng_s.eval_tasks_stacked(params, tr, ti, wo, wl, 3, want_mse=True,
+                                 x_arena=x, y_arena=y)
+    assert torch.allclose(a, b, atol=1e-4), (a - b).abs().max()
