@@ -79,9 +79,26 @@ class VmapEngine:
         return make_opt_state(kind, n_rows, self.packer.n_params, lr, wd,
                               self.device)
 
+    # activation memory bound: vmapped conv activations scale with
+    # pairs x batch; chunk the pair dimension (pairs are independent, so
+    # chunking is numerically exact)
+    MAX_PAIRS = 256
+
     def train(self, global_params: torch.Tensor, replicas: torch.Tensor,
               plan, opt: Dict, x_arena: torch.Tensor, y_arena: torch.Tensor,
               n_models: int, x_mask: Optional[torch.Tensor] = None) -> None:
+        n_pairs = len(plan.rows)
+        if n_pairs > self.MAX_PAIRS:
+            import dataclasses
+            for g0 in range(0, n_pairs, self.MAX_PAIRS):
+                sl = slice(g0, g0 + self.MAX_PAIRS)
+                sub = dataclasses.replace(
+                    plan, rows=plan.rows[sl], step_off=plan.step_off[sl],
+                    step_len=plan.step_len[sl])
+                self.train(global_params, replicas, sub, opt, x_arena,
+                           y_arena, n_models,
+                           x_mask=x_mask[sl] if x_mask is not None else None)
+            return
         rows = torch.as_tensor(plan.rows, dtype=torch.int64,
                                device=self.device)
         G = rows.numel()
@@ -143,7 +160,7 @@ class VmapEngine:
                            want_mse: bool = False, x_arena=None,
                            y_arena=None,
                            x_mask: Optional[torch.Tensor] = None,
-                           max_windows: int = 256) -> torch.Tensor:
+                           max_samples: int = 16384) -> torch.Tensor:
         rowsn = 4 if want_mse else 3
         out = torch.zeros(rowsn, n_tasks, dtype=torch.float64,
                           device=self.device)
@@ -151,6 +168,10 @@ class VmapEngine:
         if W == 0:
             return out
         self.module.eval()
+        # bound activation memory: windows per sweep so that
+        # windows x batch <= max_samples
+        bmax_all = int(win_len.max())
+        max_windows = max(1, max_samples // max(1, bmax_all))
         for w0 in range(0, W, max_windows):
             tr = task_row[w0:w0 + max_windows]
             ti = task_id[w0:w0 + max_windows]

@@ -129,6 +129,22 @@ def main():
                      log_dir=tmp, report_client=0)
         run_cfg("femnist3400_aue_ensemble", cfg, ds, comm, results)
 
+    # -- config 5b: FEMNIST CNN ensemble (vmap-batched, 400 clients) -----
+    if "femnistcnn" not in a.skip:
+        n_cl = 400
+        iters = 3
+        ds = build_ds("femnist", sample_femnist, n_cl, iters, 100,
+                      lambda c, t: (c % 4) if t >= 2 else 0, seed=1)
+        cfg = Config(model="cnn", dataset="femnist",
+                     data_dir="/nonexistent",
+                     client_num_in_total=n_cl, client_num_per_round=n_cl,
+                     batch_size=100, lr=0.003, epochs=5,
+                     comm_round=8 if not a.full else 50,
+                     total_train_iteration=iters, concept_num=4,
+                     ensemble_window=4, concept_drift_algo="aue",
+                     log_dir=tmp, report_client=0)
+        run_cfg("femnist400_cnn_aue_vmap", cfg, ds, comm, results)
+
     if comm.is_root:
         os.makedirs(os.path.dirname(a.out), exist_ok=True)
         with open(a.out, "w") as f:
