@@ -132,14 +132,14 @@ def main():
     # -- config 5b: FEMNIST CNN ensemble (vmap-batched, 400 clients) -----
     if "femnistcnn" not in a.skip:
         n_cl = 400
-        iters = 3
+        iters = 2
         ds = build_ds("femnist", sample_femnist, n_cl, iters, 100,
                       lambda c, t: (c % 4) if t >= 2 else 0, seed=1)
         cfg = Config(model="cnn", dataset="femnist",
                      data_dir="/nonexistent",
                      client_num_in_total=n_cl, client_num_per_round=n_cl,
                      batch_size=100, lr=0.003, epochs=5,
-                     comm_round=8 if not a.full else 50,
+                     comm_round=30 if not a.full else 100,
                      total_train_iteration=iters, concept_num=4,
                      ensemble_window=4, concept_drift_algo="aue",
                      log_dir=tmp, report_client=0)
