@@ -3,7 +3,7 @@
 Two implementations with identical semantics:
   * ops.mlp_torch — vectorized pure-torch (device-agnostic). The numerics
     reference, and the CPU path for tests.
-  * ops.hip (libfeddrift_hip.so) — hand-written CDNA4 HIP kernels (gfx950):
+  * ops.hip (feddrift_hip.so (ops/hip/_build)) — hand-written CDNA4 HIP kernels (gfx950):
     the whole local-training phase of an FL round in one launch, the
     model x client accuracy sweep in one launch. Used on ROCm GPUs.
 
