@@ -121,3 +121,17 @@ def test_prequential_test_set():
     assert view.test[0].n == 40
     assert np.unique(view.test[0].x[:, 0]) == [3.0]
     assert view.train[0].n == 30
+
+
+def test_leaf_synthetic_generator():
+    from feddrift_amd.data.leaf_synthetic import generate_synthetic
+    data = generate_synthetic(0.5, 0.5, n_clients=8, dim=20, n_classes=5,
+                              seed=0)
+    assert len(data) == 8
+    for k, (x, y) in data.items():
+        assert x.shape[1] == 20 and x.shape[0] == len(y) >= 5
+        assert y.min() >= 0 and y.max() < 5
+    # heterogeneity: different clients see different label distributions
+    h0 = np.bincount(data[0][1], minlength=5) / len(data[0][1])
+    h1 = np.bincount(data[1][1], minlength=5) / len(data[1][1])
+    assert np.abs(h0 - h1).sum() > 0.1
