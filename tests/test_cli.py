@@ -55,3 +55,43 @@ def test_bench_contract():
     assert out["steps"] == 3
     assert out["data"] == "synthetic"
     assert out["value"] > 0
+
+
+def test_serve_api(tmp_path):
+    """REST serving wrapper over a trained checkpoint (FastAPI TestClient)."""
+    sys.path.insert(0, os.path.join(REPO, "scripts"))
+    from feddrift_amd.config import Config
+    from feddrift_amd.engine.timeline import run_timeline
+    from feddrift_amd.data.generators import generate_data
+    import serve_api
+
+    d = str(tmp_path / "data")
+    os.makedirs(os.path.join(d, "changepoints"))
+    np.random.seed(0)
+    mat = np.zeros((3, 4), dtype=int)
+    np.savetxt(os.path.join(d, "changepoints", "T.cp"), mat, fmt="%u")
+    generate_data("sea", d, 2, 4, 0, 150, 0.0, 1, "T")
+    cfg = Config(model="fnn", dataset="sea", data_dir=d,
+                 client_num_in_total=4, client_num_per_round=4,
+                 batch_size=150, comm_round=5, epochs=3,
+                 total_train_iteration=2, concept_num=2,
+                 concept_drift_algo="softcluster",
+                 concept_drift_algo_arg="H_A_C_1_10_0",
+                 change_points="T", log_dir=str(tmp_path),
+                 report_client=0)
+    run_timeline(cfg)
+
+    from feddrift_amd.engine.serve import DriftModelServer
+    from starlette.testclient import TestClient
+    srv = DriftModelServer(cfg, str(tmp_path))
+    app = serve_api.build_app(srv)
+    client = TestClient(app)
+    assert client.get("/health").json()["status"] == "ok"
+    r = client.post("/predict", json={"client": 1,
+                                      "x": [[5.0, 9.0, 9.0],
+                                            [5.0, 1.0, 1.0]]})
+    assert r.status_code == 200
+    body = r.json()
+    assert len(body["predictions"]) == 2
+    assert client.post("/predict", json={"client": 99,
+                                         "x": [[1, 2, 3]]}).status_code == 400
