@@ -881,12 +881,6 @@ class KueAlgo(AlgoBase):
         if job.curr_iter != 0:
             self.state.set_worst_idx(int(np.argmin(self.ens_weights)))
 
-    def plan_masks(self, job: FLJob, plan: TrainPlan) -> torch.Tensor:
-        masks = self._masks_tensor(job)
-        model_of_row = torch.as_tensor(
-            plan.rows % job.n_models, dtype=torch.int64, device=job.device)
-        return masks[model_of_row]
-
     _test_lists = None
 
     def _get_test_lists(self, job: FLJob):

@@ -285,18 +285,6 @@ class FLJob:
         out = self.run_eval_dev(params, tl, want_mse).cpu().numpy()
         return [out[0], out[1], out[2], out[3] if want_mse else None]
 
-    def ens_vote_eval(self, weights: torch.Tensor, windows, mode: str,
-                      masks: Optional[torch.Tensor] = None):
-        """Weighted-vote ensemble accuracy on one client's windows
-        (AUE/KUE testing) — dispatched per model path."""
-        if self.is_module_path:
-            return self.mod_engine.ens_vote_eval(
-                self.global_params, weights, self.arena.x, self.arena.y,
-                windows, mode=mode, masks=masks)
-        return self.backend.ens_vote_eval(
-            self.spec, self.global_params, weights, self.arena.x,
-            self.arena.y, windows, mode=mode, masks=masks)
-
     def ens_vote_multi(self, weights: torch.Tensor, tl: TaskList,
                        idx: torch.Tensor, mode: str,
                        masks: Optional[torch.Tensor] = None) -> torch.Tensor:

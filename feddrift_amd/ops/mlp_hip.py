@@ -3,8 +3,8 @@
 Same API as ops.mlp_torch. train_fused and eval_tasks (the per-round hot
 loops: all local training in one launch, all accuracy sweeps in one launch)
 run on the HIP kernels in ops/hip/feddrift_kernels.hip; the cold paths
-(ensemble-vote testing, KUE confusion matrices — a few times per
-iteration) currently run on torch GPU ops and will be ported next.
+(ensemble-vote testing, KUE confusion matrices — a handful of calls per
+iteration, batched across clients) run on torch GPU ops by design.
 """
 
 from __future__ import annotations
