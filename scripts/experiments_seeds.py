@@ -27,12 +27,13 @@ def main():
     p.add_argument("--rounds", type=int, default=200)
     p.add_argument("--algo", default="softcluster")
     p.add_argument("--arg", default="H_A_C_1_10_0")
+    p.add_argument("--dataset", default="sea")
     a = p.parse_args()
     comm = Communicator()
 
     accs = []
     for seed in range(a.seeds):
-        data_dir = f"/tmp/seed_{seed}"
+        data_dir = f"/tmp/seed_{a.dataset}_{seed}"
         if comm.is_root:
             import shutil
             repo_cp = os.path.join(os.path.dirname(__file__), "..", "data",
@@ -44,11 +45,11 @@ def main():
                     shutil.copy(os.path.join(repo_cp, f),
                                 os.path.join(data_dir, "changepoints", f))
             np.random.seed(seed)
-            generate_data("sea", data_dir, 10, 10, 0, 100, 0.0, 1, "A")
+            generate_data(a.dataset, data_dir, 10, 10, 0, 100, 0.0, 1, "A")
         comm.barrier()
         log_dir = os.path.join(data_dir, "run")
         os.makedirs(log_dir, exist_ok=True)
-        cfg = Config(model="fnn", dataset="sea", data_dir=data_dir,
+        cfg = Config(model="fnn", dataset=a.dataset, data_dir=data_dir,
                      client_num_in_total=10, client_num_per_round=10,
                      batch_size=500, lr=0.01, epochs=5,
                      comm_round=a.rounds, total_train_iteration=10,
