@@ -135,3 +135,25 @@ def test_leaf_synthetic_generator():
     h0 = np.bincount(data[0][1], minlength=5) / len(data[0][1])
     h1 = np.bincount(data[1][1], minlength=5) / len(data[1][1])
     assert np.abs(h0 - h1).sum() > 0.1
+
+
+def test_reference_style_csv_interop(tmp_path):
+    """CSV files written the reference's way (pandas to_csv, header row,
+    float label column — sea/data_loader.py:80-82, MNIST
+    data_loader_cont.py:85-88) load correctly through our RawStore."""
+    import pandas as pd
+    d = tmp_path / "sea"
+    d.mkdir()
+    df = pd.DataFrame({"f1": [1.5, 2.5], "f2": [3.5, 4.5],
+                       "f3": [5.5, 6.5], "label": [0.0, 1.0]})
+    df.to_csv(d / "client_0_iter_0.csv", index=False)
+    # MNIST-style: integer column names, float labels
+    arr = np.random.default_rng(0).random((3, 4))
+    df2 = pd.DataFrame(np.column_stack([arr, [0.0, 7.0, 3.0]]))
+    df2.to_csv(d / "client_1_iter_0.csv", index=False)
+
+    store = RawStore(str(tmp_path), "sea", 2)
+    x, y = store.get(0, 0)
+    assert x.shape == (2, 3) and y.tolist() == [0, 1]
+    x2, y2 = store.get(1, 0)
+    assert x2.shape == (3, 4) and y2.tolist() == [0, 7, 3]
