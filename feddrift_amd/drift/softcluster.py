@@ -531,6 +531,12 @@ class SoftClusterState:
     def get_test_model_idx(self, curr_iter: int, client_idx: int) -> int:
         return int(np.argmax(self.train_data_weights[curr_iter][:, client_idx]))
 
+    def get_test_model_idx_all(self, curr_iter: int) -> np.ndarray:
+        """Vectorized per-client test-model assignment (argmax of each
+        weight column) — the per-round per-client python loop is O(C) and
+        dominates host time at thousands of clients."""
+        return np.argmax(self.train_data_weights[curr_iter], axis=0)
+
     def get_weights(self) -> Dict[int, np.ndarray]:
         return self.train_data_weights
 

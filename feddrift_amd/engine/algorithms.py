@@ -370,9 +370,7 @@ class SoftClusterAlgo(AlgoBase):
         if round_idx % cfg.frequency_of_the_test == 0 or \
                 round_idx == cfg.comm_round - 1:
             st = self.state
-            C = cfg.client_num_in_total
-            mpc = np.array([st.get_test_model_idx(job.curr_iter, c)
-                            for c in range(C)])
+            mpc = st.get_test_model_idx_all(job.curr_iter)
             # softcluster tests train data on the CURRENT-iteration all_data
             # (FedAvgEnsAggregatorSoftCluster.py:227-231)
             tr, te = self.client_eval_views(job, mpc, train_on_view=False)
