@@ -660,6 +660,104 @@ void mlp_eval_kernel(EvalArgs a) {
   }
 }
 
+// specialized small-model eval kernel: fully unrolled per-sample pipeline,
+// x cached in registers (the generic kernel re-reads x[d] from global for
+// every hidden unit; at SEA shapes that is D*H loads per sample and the
+// kernel measures 87% SQ_WAIT) — same dispatch list as the train kernel
+template <int TD, int TH, int TO, int KIND>
+__global__ __launch_bounds__(THREADS)
+void mlp_eval_small_kernel(EvalArgs a) {
+  constexpr int TP = (KIND == KIND_FNN)
+                         ? (TH * TD + TH + TO * TH + TO)
+                         : (TO * TD + TO);
+  constexpr int HD = TH * TD;
+  constexpr int OH = TO * TH;
+  const int wdx = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int64_t row = a.task_row[wdx];
+  const int64_t tsk = a.task_id[wdx];
+  const int64_t off = a.off[wdx];
+  const int n = (int)a.len[wdx];
+
+  __shared__ __attribute__((aligned(16))) float w[TP];
+  __shared__ float red[8];
+  __shared__ float msk[TD];
+  for (int p = tid; p < TP; p += THREADS) w[p] = a.params[row * TP + p];
+  if (a.x_mask && tid < TD) msk[tid] = a.x_mask[(int64_t)wdx * TD + tid];
+  __syncthreads();
+
+  float c_acc = 0.f, l_acc = 0.f, e_acc = 0.f;
+  for (int i = tid; i < n; i += THREADS) {
+    float x[TD];
+#pragma unroll
+    for (int d = 0; d < TD; ++d) {
+      x[d] = a.x[(off + i) * TD + d];
+      if (a.x_mask) x[d] *= msk[d];
+    }
+    const int yi = (int)a.y[off + i];
+    float logits[TO];
+    if constexpr (KIND == KIND_FNN) {
+#pragma unroll
+      for (int o = 0; o < TO; ++o) logits[o] = w[HD + TH + OH + o];
+#pragma unroll
+      for (int h = 0; h < TH; ++h) {
+        float z = w[HD + h];
+#pragma unroll
+        for (int d = 0; d < TD; ++d) z += w[h * TD + d] * x[d];
+        z = z > 0.f ? z : 0.f;
+#pragma unroll
+        for (int o = 0; o < TO; ++o) logits[o] += w[HD + TH + o * TH + h] * z;
+      }
+    } else {
+#pragma unroll
+      for (int o = 0; o < TO; ++o) {
+        float z = w[TO * TD + o];
+#pragma unroll
+        for (int d = 0; d < TD; ++d) z += w[o * TD + d] * x[d];
+        logits[o] = 1.f / (1.f + __expf(-z));
+      }
+    }
+    int best = 0;
+    float zmax = logits[0];
+#pragma unroll
+    for (int o = 1; o < TO; ++o)
+      if (logits[o] > zmax) { zmax = logits[o]; best = o; }
+    float zsum = 0.f;
+#pragma unroll
+    for (int o = 0; o < TO; ++o) zsum += __expf(logits[o] - zmax);
+    const float lse = logf(zsum) + zmax;
+    c_acc += (best == yi) ? 1.f : 0.f;
+    l_acc += lse - logits[yi];
+    if (a.mse) {
+      const float ptrue = __expf(logits[yi] - lse);
+      e_acc += (1.f - ptrue) * (1.f - ptrue);
+    }
+  }
+
+  float cs = block_reduce_sum(c_acc, red);
+  float ls = block_reduce_sum(l_acc, red);
+  float es = a.mse ? block_reduce_sum(e_acc, red) : 0.f;
+  if (tid == 0) {
+    atomicAdd(&a.correct[tsk], (double)cs);
+    atomicAdd(&a.total[tsk], (double)n);
+    atomicAdd(&a.loss[tsk], (double)ls);
+    if (a.mse) atomicAdd(&a.mse[tsk], (double)es);
+  }
+}
+
+static bool launch_small_eval(const EvalArgs& args, int W,
+                              hipStream_t stream) {
+#define TRY_ESHAPE(SD, SH, SO, SK)                                          \
+  if (args.kind == SK && args.D == SD && args.H == SH && args.O == SO) {    \
+    hipLaunchKernelGGL((mlp_eval_small_kernel<SD, SH, SO, SK>), dim3(W),    \
+                       dim3(THREADS), 0, stream, args);                     \
+    return true;                                                            \
+  }
+  SMALL_SHAPE_LIST(TRY_ESHAPE)
+#undef TRY_ESHAPE
+  return false;
+}
+
 // ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
@@ -764,8 +862,10 @@ torch::Tensor eval_tasks_hip(
 
   const size_t lds_bytes = (size_t)(P + 8) * sizeof(float);
   TORCH_CHECK(lds_bytes <= 150 * 1024, "model too large for eval LDS path");
-  hipLaunchKernelGGL(mlp_eval_kernel, dim3(W), dim3(THREADS), lds_bytes,
-                     c10::hip::getCurrentHIPStream(), args);
+  if (!launch_small_eval(args, W, c10::hip::getCurrentHIPStream())) {
+    hipLaunchKernelGGL(mlp_eval_kernel, dim3(W), dim3(THREADS), lds_bytes,
+                       c10::hip::getCurrentHIPStream(), args);
+  }
   TORCH_CHECK(hipGetLastError() == hipSuccess, "mlp_eval_kernel launch");
   return out;
 }
