@@ -747,6 +747,13 @@ void mlp_eval_small_kernel(EvalArgs a) {
 
 static bool launch_small_eval(const EvalArgs& args, int W,
                               hipStream_t stream) {
+  // A/B escape hatch for benchmarking the generic vs templated eval path
+  // on the same box (FEDDRIFT_NO_SMALL_EVAL=1 forces the generic kernel).
+  static const bool disabled = [] {
+    const char* e = getenv("FEDDRIFT_NO_SMALL_EVAL");
+    return e && e[0] == '1';
+  }();
+  if (disabled) return false;
 #define TRY_ESHAPE(SD, SH, SO, SK)                                          \
   if (args.kind == SK && args.D == SD && args.H == SH && args.O == SO) {    \
     hipLaunchKernelGGL((mlp_eval_small_kernel<SD, SH, SO, SK>), dim3(W),    \
