@@ -18,7 +18,11 @@ def main():
     arm = "generic" if os.environ.get("FEDDRIFT_NO_SMALL_EVAL") == "1" \
         else "small"
     comm = Communicator()
-    for c, steps in [(10, 600), (200, 300), (3400, 25)]:
+    sizes = [(10, 600), (200, 300), (3400, 25)]
+    if len(sys.argv) > 1:                     # "clients:steps,clients:steps"
+        sizes = [tuple(int(v) for v in part.split(":"))
+                 for part in sys.argv[1].split(",")]
+    for c, steps in sizes:
         rps = run_scale(comm, c, steps=steps, warmup=max(10, steps // 5))
         print(f"AB {arm} clients={c} rps={rps:.1f}", flush=True)
 
