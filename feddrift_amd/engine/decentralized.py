@@ -73,6 +73,77 @@ class DecentralizedDSGD:
         return float((self.params - mean).norm(dim=1).max())
 
 
+class PushSumDSGD:
+    """Stochastic gradient push over a DIRECTED topology (reference
+    fedml_api/standalone/decentralized: push-sum gossip; asymmetric
+    out-neighbor graphs from AsymmetricTopologyManager).
+
+    Each worker keeps a numerator row x and a scalar weight w; the
+    de-biased estimate z = x / w is what local SGD steps update.  Both
+    x and w are mixed with a COLUMN-stochastic matrix each round, so
+    z converges to the network average even though the directed graph
+    is not doubly stochastic (the symmetric-mixing assumption DSGD
+    needs).  As with DSGD above, the whole per-neighbor push/receive
+    cycle of the reference is one [W, W] x [W, P] GEMM on rocBLAS.
+    """
+
+    def __init__(self, spec: MLPSpec, n_workers: int,
+                 topology: BaseTopologyManager, init_flat: torch.Tensor,
+                 x_arena: torch.Tensor, y_arena: torch.Tensor,
+                 windows_per_worker: List[List[Tuple[int, int]]],
+                 lr: float = 0.05, epochs: int = 1,
+                 device: Optional[torch.device] = None, seed: int = 0):
+        self.spec = spec
+        self.n = n_workers
+        self.device = device or x_arena.device
+        self.num = init_flat.unsqueeze(0).repeat(n_workers, 1) \
+            .to(self.device)
+        self.w = torch.ones(n_workers, device=self.device)
+        # column-stochastic mixing: node j splits its mass equally over
+        # its out-neighbors (incl. itself) -> column j of A sums to 1
+        out = torch.as_tensor(topology.topology, dtype=torch.float32,
+                              device=self.device)
+        out = (out > 0).float()
+        out.fill_diagonal_(1.0)
+        self.mix = out / out.sum(dim=0, keepdim=True)
+        self.x = x_arena
+        self.y = y_arena
+        self.windows = windows_per_worker
+        self.epochs = epochs
+        self.opt = mlp_torch.make_opt_state("sgd", n_workers,
+                                            spec.n_params, lr, 0.0,
+                                            self.device)
+        self.rng = np.random.default_rng(seed)
+
+    def round(self) -> None:
+        E = self.epochs
+        offs = np.zeros((self.n, E), dtype=np.int64)
+        lens = np.zeros((self.n, E), dtype=np.int64)
+        for k in range(self.n):
+            wins = self.windows[k]
+            picks = self.rng.integers(0, len(wins), size=E)
+            offs[k] = [wins[p][0] for p in picks]
+            lens[k] = [wins[p][1] for p in picks]
+        z = self.num / self.w.unsqueeze(1)
+        mlp_torch.train_fused(
+            self.spec, z, torch.arange(self.n, device=self.device),
+            self.x, self.y,
+            torch.as_tensor(offs, device=self.device),
+            torch.as_tensor(lens, device=self.device), self.opt)
+        # push step: re-scale the stepped estimate back into numerator
+        # space, then one column-stochastic mixing GEMM for both x and w
+        self.num = self.mix @ (z * self.w.unsqueeze(1))
+        self.w = self.mix @ self.w
+
+    def estimates(self) -> torch.Tensor:
+        return self.num / self.w.unsqueeze(1)
+
+    def consensus_distance(self) -> float:
+        z = self.estimates()
+        mean = z.mean(dim=0, keepdim=True)
+        return float((z - mean).norm(dim=1).max())
+
+
 class HierarchicalFL:
     """Two-level FedAvg: clients -> group aggregation every round,
     groups -> global aggregation every `group_comm_round` rounds
