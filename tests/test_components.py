@@ -109,6 +109,31 @@ def test_decentralized_dsgd_learns_and_converges():
     assert acc > 0.7, acc
 
 
+def test_pushsum_dsgd_directed_consensus():
+    from feddrift_amd.engine.decentralized import PushSumDSGD
+    spec = spec_for("fnn", 3, 2)
+    torch.manual_seed(0)
+    from feddrift_amd.models.zoo import FeedForwardNN
+    from feddrift_amd.models.packed import PackedMLP
+    init = PackedMLP(spec).flatten(FeedForwardNN(3, 2, 6).state_dict())
+    x, y, wins = _toy_data(6)
+    topo = AsymmetricTopologyManager(6, 2)       # directed out-neighbors
+    eng = PushSumDSGD(spec, 6, topo, init, x, y, wins, lr=0.05,
+                      epochs=2, device=torch.device("cpu"))
+    # mixing matrix is column-stochastic (push-sum requirement)
+    assert torch.allclose(eng.mix.sum(dim=0), torch.ones(6), atol=1e-6)
+    for _ in range(60):
+        eng.round()
+    # total weight mass is conserved by column-stochastic mixing
+    assert abs(eng.w.sum().item() - 6.0) < 1e-3
+    assert eng.consensus_distance() < 1.0
+    from feddrift_amd.ops import mlp_torch
+    logits = mlp_torch.forward_logits(spec, eng.estimates()[:1],
+                                      x.unsqueeze(0))
+    acc = (logits.squeeze(0).argmax(-1) == y).float().mean().item()
+    assert acc > 0.7, acc
+
+
 def test_hierarchical_fl_runs():
     spec = spec_for("fnn", 3, 2)
     torch.manual_seed(0)
