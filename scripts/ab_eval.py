@@ -22,9 +22,64 @@ def main():
     if len(sys.argv) > 1:                     # "clients:steps,clients:steps"
         sizes = [tuple(int(v) for v in part.split(":"))
                  for part in sys.argv[1].split(",")]
+    phases = os.environ.get("FEDDRIFT_AB_PHASES") == "1"
     for c, steps in sizes:
-        rps = run_scale(comm, c, steps=steps, warmup=max(10, steps // 5))
-        print(f"AB {arm} clients={c} rps={rps:.1f}", flush=True)
+        if phases:
+            _phase_breakdown(c, steps)
+        else:
+            rps = run_scale(comm, c, steps=steps, warmup=max(10, steps // 5))
+            print(f"AB {arm} clients={c} rps={rps:.1f}", flush=True)
+
+
+def _phase_breakdown(n_clients, steps):
+    """Where does the wall clock go at this scale? Times the four phases
+    of bench.one_round with a device sync after each (so the first sync
+    absorbs queued async work — read 'train' as dispatch+kernel)."""
+    import numpy as np
+    import torch
+
+    import bench
+    from feddrift_amd.comm import Communicator
+    from feddrift_amd.config import Config
+    from feddrift_amd.eval.metrics import MetricLogger
+    from feddrift_amd.engine.fljob import FLJob
+    from feddrift_amd.engine.profiling import PhaseTimer
+    from scripts.bench_sweep import run_scale as _unused  # noqa: F401
+
+    comm = Communicator()
+    cfg = Config(model="fnn", dataset="sea", data_dir="/nonexistent",
+                 client_num_in_total=n_clients,
+                 client_num_per_round=n_clients, batch_size=bench.SEQ,
+                 client_optimizer="adam", lr=0.01, epochs=bench.EPOCHS,
+                 comm_round=10 ** 9,
+                 total_train_iteration=bench.CURR_ITER + 1,
+                 curr_train_iteration=bench.CURR_ITER,
+                 concept_num=bench.N_MODELS,
+                 concept_drift_algo="softcluster",
+                 concept_drift_algo_arg="H_A_F_1_06_0", change_points="A",
+                 dummy_arg=0, report_client=0, bench_mode=1)
+    job = FLJob(cfg, comm, MetricLogger(enabled=True, to_file=False),
+                dataset=bench.build_dataset(n_clients, seed=1234))
+    idx = np.arange(n_clients)
+    for r in range(10):
+        bench.one_round(job, r, idx)
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    timer = PhaseTimer(sync_fn=torch.cuda.synchronize
+                       if torch.cuda.is_available() else None)
+    for r in range(10, 10 + steps):
+        with timer.phase("plan"):
+            plan = job.algo.plan(job, r, idx)
+        with timer.phase("train"):
+            job.train(plan)
+        with timer.phase("aggregate"):
+            job.algo.aggregate(job, r, plan, idx)
+            job.algo.post_aggregate(job, r)
+        with timer.phase("test"):
+            job.algo.test(job, r)
+    import json
+    print(f"PHASES clients={n_clients} "
+          + json.dumps(timer.summary()), flush=True)
 
 
 if __name__ == "__main__":
