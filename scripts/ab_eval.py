@@ -22,7 +22,7 @@ def main():
     if len(sys.argv) > 1:                     # "clients:steps,clients:steps"
         sizes = [tuple(int(v) for v in part.split(":"))
                  for part in sys.argv[1].split(",")]
-    phases = os.environ.get("FEDDRIFT_AB_PHASES") == "1"
+    phases = os.environ.get("FEDDRIFT_AB_PHASES") in ("1", "2")
     for c, steps in sizes:
         if phases:
             _phase_breakdown(c, steps)
@@ -80,6 +80,33 @@ def _phase_breakdown(n_clients, steps):
     import json
     print(f"PHASES clients={n_clients} "
           + json.dumps(timer.summary()), flush=True)
+
+    if os.environ.get("FEDDRIFT_AB_PHASES") == "2":
+        # sub-phase split of the test phase: device eval+sync vs D2H vs
+        # the algorithm's host-side bookkeeping around client_eval
+        import time as _time
+        sub = {"eval_dev": 0.0, "d2h": 0.0, "rest": 0.0}
+        orig = job.run_eval_dev
+
+        def timed_eval(*a, **k):
+            t0 = _time.perf_counter()
+            r = orig(*a, **k)
+            torch.cuda.synchronize()
+            sub["eval_dev"] += _time.perf_counter() - t0
+            t1 = _time.perf_counter()
+            _ = r.cpu()
+            sub["d2h"] += _time.perf_counter() - t1
+            return r
+
+        job.run_eval_dev = timed_eval
+        t0 = _time.perf_counter()
+        for r in range(100, 100 + steps):
+            job.algo.test(job, r)
+        total = _time.perf_counter() - t0
+        sub["rest"] = total - sub["eval_dev"] - sub["d2h"]
+        print(f"TESTSUB clients={n_clients} steps={steps} "
+              + json.dumps({k: round(v / steps * 1e3, 4)
+                            for k, v in sub.items()}), flush=True)
 
 
 if __name__ == "__main__":
