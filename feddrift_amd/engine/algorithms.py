@@ -185,9 +185,13 @@ class SoftClusterAlgo(AlgoBase):
     def _state_path(self, job):
         return job.ckpt_path("sc_state.pkl")
 
+    _mpc = None      # cached per-client test-model argmax (invalidated on
+                     # any clustering event; weights are static in between)
+
     def init_iteration(self, job: FLJob) -> None:
         cfg = job.cfg
         hooks = Hooks(job)
+        self._mpc = None
         if cfg.bench_mode:
             # synthetic steady-state: client c trains/tests model c % K at
             # every iteration — K active clusters, full per-round load,
@@ -273,6 +277,7 @@ class SoftClusterAlgo(AlgoBase):
         rebuild the plan structure next round."""
         self._tmpl = None
         self._mask = None
+        self._mpc = None
 
     def plan(self, job: FLJob, round_idx: int,
              client_idx: np.ndarray) -> TrainPlan:
@@ -370,7 +375,12 @@ class SoftClusterAlgo(AlgoBase):
         if round_idx % cfg.frequency_of_the_test == 0 or \
                 round_idx == cfg.comm_round - 1:
             st = self.state
-            mpc = st.get_test_model_idx_all(job.curr_iter)
+            mpc = self._mpc
+            if mpc is None:
+                # np.argmax over the [K, C] weight matrix costs >100 us at
+                # thousands of clients; the assignment only changes at
+                # clustering events, so cache it per iteration/recluster
+                mpc = self._mpc = st.get_test_model_idx_all(job.curr_iter)
             # softcluster tests train data on the CURRENT-iteration all_data
             # (FedAvgEnsAggregatorSoftCluster.py:227-231)
             tr, te = self.client_eval_views(job, mpc, train_on_view=False)
