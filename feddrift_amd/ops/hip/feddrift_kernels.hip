@@ -762,9 +762,10 @@ static bool launch_small_eval(const EvalArgs& args, int W,
   if (disabled) return false;
   // Same-box A/B (profiles/README.md): the register-pipelined kernel has
   // longer per-thread dependent chains than the generic unit-parallel one,
-  // so it only wins once the grid saturates the 256 CUs (measured
-  // crossover ~1600 workgroups; +14..16% at W>=11k, -6% at W~800).
-  if (W < 1536 && !forced) return false;
+  // so it only wins once the grid saturates the 256 CUs with enough waves.
+  // Measured (bench eval = 8 window-blocks/client): -6% at W=1600, tie at
+  // W=3200, +5% at W=6400 rising to +16% at W=27k -> gate at 4096.
+  if (W < 4096 && !forced) return false;
 #define TRY_ESHAPE(SD, SH, SO, SK)                                          \
   if (args.kind == SK && args.D == SD && args.H == SH && args.O == SO) {    \
     hipLaunchKernelGGL((mlp_eval_small_kernel<SD, SH, SO, SK>), dim3(W),    \

@@ -10,7 +10,7 @@ import torch
 
 # Latched once per process by the C++ dispatch: make the templated
 # small-eval kernel run even at tiny test grids (its production gate is
-# W >= 1536 workgroups), so the fnn(3,6,2) parity case below exercises it.
+# W >= 4096 workgroups), so the fnn(3,6,2) parity case below exercises it.
 # The lr(6,3) case is not in the template list and covers the generic
 # eval kernel in the same process.
 os.environ.setdefault("FEDDRIFT_FORCE_SMALL_EVAL", "1")
