@@ -93,12 +93,14 @@ __device__ __forceinline__ float block_reduce_sum(float val, float* scratch) {
 
 __device__ __forceinline__ void stage_weights(const TrainArgs& a, int g,
                                               int64_t row, float* w) {
+  // strided by blockDim.x: the templated small kernel also launches with
+  // 64-thread (single-wave) blocks
   if (a.in_params) {
     const int64_t src = (int64_t)a.model_of[g] * a.P;
-    for (int p = threadIdx.x; p < a.P; p += THREADS)
+    for (int p = threadIdx.x; p < a.P; p += blockDim.x)
       w[p] = a.in_params[src + p];
   } else {
-    for (int p = threadIdx.x; p < a.P; p += THREADS)
+    for (int p = threadIdx.x; p < a.P; p += blockDim.x)
       w[p] = a.params[row * a.P + p];
   }
   __syncthreads();
@@ -136,12 +138,12 @@ __device__ __forceinline__ void opt_update(const TrainArgs& a, int64_t row,
 __device__ __forceinline__ void write_back(const TrainArgs& a, int g,
                                            int64_t row, const float* w) {
   const int tid = threadIdx.x;
-  for (int p = tid; p < a.P; p += THREADS) a.params[row * a.P + p] = w[p];
+  for (int p = tid; p < a.P; p += blockDim.x) a.params[row * a.P + p] = w[p];
   if (a.partial && a.sample_w) {
     const float sw = a.sample_w[g];
     if (sw > 0.f) {
       const int64_t pbase = (int64_t)a.model_of[g] * (a.P + 1);
-      for (int p = tid; p < a.P; p += THREADS)
+      for (int p = tid; p < a.P; p += blockDim.x)
         atomicAdd(&a.partial[pbase + p], sw * w[p]);
       if (tid == 0) atomicAdd(&a.partial[pbase + a.P], sw);
     }
@@ -328,8 +330,8 @@ void mlp_train_kernel(TrainArgs a) {
 // kernel above stays latency-bound on LDS round trips for these shapes.
 // ---------------------------------------------------------------------------
 
-template <int TD, int TH, int TO, int KIND>
-__global__ __launch_bounds__(THREADS)
+template <int TD, int TH, int TO, int KIND, int BS>
+__global__ __launch_bounds__(BS)
 void mlp_train_small_kernel(TrainArgs a) {
   constexpr int TP = (KIND == KIND_FNN)
                          ? (TH * TD + TH + TO * TH + TO)
@@ -354,7 +356,7 @@ void mlp_train_small_kernel(TrainArgs a) {
   if (a.x_mask && tid < TD) msk[tid] = a.x_mask[(int64_t)g * TD + tid];
   const bool adam = (a.opt == OPT_ADAM);
   if (adam) {
-    for (int p = tid; p < TP; p += THREADS) {
+    for (int p = tid; p < TP; p += BS) {
       sm_[p] = a.m[row * TP + p];
       sv_[p] = a.v[row * TP + p];
       svm_[p] = a.vmax[row * TP + p];
@@ -373,14 +375,14 @@ void mlp_train_small_kernel(TrainArgs a) {
     if (n == 0) continue;
     const float inv_n = 1.0f / (float)n;
 
-    for (int p = tid; p < TP; p += THREADS) grad[p] = 0.f;
+    for (int p = tid; p < TP; p += BS) grad[p] = 0.f;
     __syncthreads();
 
     float gacc[TP];
 #pragma unroll
     for (int p = 0; p < TP; ++p) gacc[p] = 0.f;
 
-    for (int i = tid; i < n; i += THREADS) {
+    for (int i = tid; i < n; i += BS) {
       float x[TD];
 #pragma unroll
       for (int d = 0; d < TD; ++d) {
@@ -489,7 +491,7 @@ void mlp_train_small_kernel(TrainArgs a) {
     // optimizer step, all state in LDS (numerics match ops/mlp_torch.py)
     if (!adam) {
       const float lr_ = a.lr[row];
-      for (int p = tid; p < TP; p += THREADS) w[p] -= lr_ * grad[p];
+      for (int p = tid; p < TP; p += BS) w[p] -= lr_ * grad[p];
     } else {
       if (tid == 0) {
         tstep += 1;
@@ -501,7 +503,7 @@ void mlp_train_small_kernel(TrainArgs a) {
       const float bc1 = 1.f - b1pow;
       const float bc2 = 1.f - b2pow;
       const float lr_ = a.lr[row];
-      for (int p = tid; p < TP; p += THREADS) {
+      for (int p = tid; p < TP; p += BS) {
         const float gr = grad[p] + a.wd * w[p];
         const float mn = b1 * sm_[p] + (1.f - b1) * gr;
         const float vn = b2 * sv_[p] + (1.f - b2) * gr * gr;
@@ -517,7 +519,7 @@ void mlp_train_small_kernel(TrainArgs a) {
   }
 
   if (adam) {
-    for (int p = tid; p < TP; p += THREADS) {
+    for (int p = tid; p < TP; p += BS) {
       a.m[row * TP + p] = sm_[p];
       a.v[row * TP + p] = sv_[p];
       a.vmax[row * TP + p] = svm_[p];
@@ -535,11 +537,25 @@ void mlp_train_small_kernel(TrainArgs a) {
   X(2, 0, 2, KIND_LR)
 
 static bool launch_small(const TrainArgs& args, int G, hipStream_t stream) {
-#define TRY_SHAPE(SD, SH, SO, SK)                                         \
-  if (args.kind == SK && args.D == SD && args.H == SH && args.O == SO) {  \
-    hipLaunchKernelGGL((mlp_train_small_kernel<SD, SH, SO, SK>), dim3(G), \
-                       dim3(THREADS), 0, stream, args);                   \
-    return true;                                                          \
+  // Block-size dispatch: every wave of a block redundantly executes the
+  // TP x 6-level shuffle reduction, so once the grid alone can cover the
+  // chip (G single-wave blocks >> 256 CUs), 64-thread blocks cut the
+  // total reduction work 4x. Below that, 4-wave blocks win on occupancy.
+  static const int bs64_min = [] {
+    const char* e = getenv("FEDDRIFT_TRAIN_BS64_MIN_G");
+    return e ? atoi(e) : 8192;
+  }();
+  const bool bs64 = args.E > 0 && G >= bs64_min;
+#define TRY_SHAPE(SD, SH, SO, SK)                                          \
+  if (args.kind == SK && args.D == SD && args.H == SH && args.O == SO) {   \
+    if (bs64) {                                                            \
+      hipLaunchKernelGGL((mlp_train_small_kernel<SD, SH, SO, SK, 64>),     \
+                         dim3(G), dim3(64), 0, stream, args);              \
+    } else {                                                               \
+      hipLaunchKernelGGL((mlp_train_small_kernel<SD, SH, SO, SK, THREADS>),\
+                         dim3(G), dim3(THREADS), 0, stream, args);         \
+    }                                                                      \
+    return true;                                                           \
   }
   SMALL_SHAPE_LIST(TRY_SHAPE)
 #undef TRY_SHAPE
