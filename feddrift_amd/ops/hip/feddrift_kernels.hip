@@ -537,15 +537,18 @@ void mlp_train_small_kernel(TrainArgs a) {
   X(2, 0, 2, KIND_LR)
 
 static bool launch_small(const TrainArgs& args, int G, hipStream_t stream) {
-  // Block-size dispatch: every wave of a block redundantly executes the
-  // TP x 6-level shuffle reduction, so once the grid alone can cover the
-  // chip (G single-wave blocks >> 256 CUs), 64-thread blocks cut the
-  // total reduction work 4x. Below that, 4-wave blocks win on occupancy.
+  // Block-size choice: every wave of a block redundantly executes the
+  // TP x 6-level shuffle reduction, so single-wave (64-thread) blocks cut
+  // the total reduction work 4x. Measured same-box across grid sizes
+  // (profiles/README.md): +41% at G=3400, +52% at G=10000, and never
+  // slower even at G=100 (the per-pair critical path is the same serial
+  // E-step chain either way). Default BS=64 everywhere; the env knob
+  // remains as the benchmarking escape hatch (set huge to force 256).
   static const int bs64_min = [] {
     const char* e = getenv("FEDDRIFT_TRAIN_BS64_MIN_G");
-    return e ? atoi(e) : 8192;
+    return e ? atoi(e) : 0;
   }();
-  const bool bs64 = args.E > 0 && G >= bs64_min;
+  const bool bs64 = G >= bs64_min;
 #define TRY_SHAPE(SD, SH, SO, SK)                                          \
   if (args.kind == SK && args.D == SD && args.H == SH && args.O == SO) {   \
     if (bs64) {                                                            \
