@@ -683,8 +683,8 @@ void mlp_eval_kernel(EvalArgs a) {
 // x cached in registers (the generic kernel re-reads x[d] from global for
 // every hidden unit; at SEA shapes that is D*H loads per sample and the
 // kernel measures 87% SQ_WAIT) — same dispatch list as the train kernel
-template <int TD, int TH, int TO, int KIND>
-__global__ __launch_bounds__(THREADS)
+template <int TD, int TH, int TO, int KIND, int BS>
+__global__ __launch_bounds__(BS)
 void mlp_eval_small_kernel(EvalArgs a) {
   constexpr int TP = (KIND == KIND_FNN)
                          ? (TH * TD + TH + TO * TH + TO)
@@ -701,12 +701,12 @@ void mlp_eval_small_kernel(EvalArgs a) {
   __shared__ __attribute__((aligned(16))) float w[TP];
   __shared__ float red[8];
   __shared__ float msk[TD];
-  for (int p = tid; p < TP; p += THREADS) w[p] = a.params[row * TP + p];
+  for (int p = tid; p < TP; p += BS) w[p] = a.params[row * TP + p];
   if (a.x_mask && tid < TD) msk[tid] = a.x_mask[(int64_t)wdx * TD + tid];
   __syncthreads();
 
   float c_acc = 0.f, l_acc = 0.f, e_acc = 0.f;
-  for (int i = tid; i < n; i += THREADS) {
+  for (int i = tid; i < n; i += BS) {
     float x[TD];
 #pragma unroll
     for (int d = 0; d < TD; ++d) {
@@ -785,10 +785,14 @@ static bool launch_small_eval(const EvalArgs& args, int W,
   // Measured (bench eval = 8 window-blocks/client): -6% at W=1600, tie at
   // W=3200, +5% at W=6400 rising to +16% at W=27k -> gate at 4096.
   if (W < 4096 && !forced) return false;
+  // 64-thread (single-wave) blocks: windows are capped at 128 samples
+  // (TaskList.CHUNK), so 256-thread blocks leave half the lanes idle and
+  // run the block reduction across 4 waves; the gate above guarantees the
+  // grid is large enough to fill the CUs with single-wave blocks.
 #define TRY_ESHAPE(SD, SH, SO, SK)                                          \
   if (args.kind == SK && args.D == SD && args.H == SH && args.O == SO) {    \
-    hipLaunchKernelGGL((mlp_eval_small_kernel<SD, SH, SO, SK>), dim3(W),    \
-                       dim3(THREADS), 0, stream, args);                     \
+    hipLaunchKernelGGL((mlp_eval_small_kernel<SD, SH, SO, SK, 64>),         \
+                       dim3(W), dim3(64), 0, stream, args);                 \
     return true;                                                            \
   }
   SMALL_SHAPE_LIST(TRY_ESHAPE)
