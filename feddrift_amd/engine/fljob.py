@@ -373,8 +373,7 @@ class FLJob:
             self._partial_fused = False
             self._robust_clip(plan)
             return
-        off_t = torch.as_tensor(plan.step_off, dtype=torch.int64, device=dev)
-        len_t = torch.as_tensor(plan.step_len, dtype=torch.int64, device=dev)
+        off_t, len_t = self._upload_steps(plan, dev)
         if hip:
             # per-template constants (rows / model / aggregation weights)
             # are uploaded once and reused every round
@@ -406,6 +405,36 @@ class FLJob:
                 off_t, len_t, self.opt, x_mask=plan.x_mask)
             self._partial_fused = False
             self._robust_clip(plan)
+
+    def _upload_steps(self, plan: TrainPlan, dev: torch.device):
+        """Per-round upload of the [G, E] step windows. On GPU, a pageable
+        torch.as_tensor costs tens of us per array in alloc + staged copy;
+        a per-template pinned double buffer turns both arrays into ONE
+        non-blocking DMA (the buffer slot is event-guarded against reuse
+        while a previous round's copy is still in flight)."""
+        tmpl = plan.template
+        if dev.type != "cuda" or tmpl is None:
+            return (torch.as_tensor(plan.step_off, dtype=torch.int64,
+                                    device=dev),
+                    torch.as_tensor(plan.step_len, dtype=torch.int64,
+                                    device=dev))
+        G, E = plan.step_off.shape
+        st = getattr(tmpl, "_stage", None)
+        if st is None or st["pin"].shape[2:] != (G, E):
+            pin = torch.empty(2, 2, G, E, dtype=torch.int64,
+                              pin_memory=True)
+            tmpl._stage = st = {
+                "pin": pin, "np": pin.numpy(),
+                "buf": torch.empty(2, G, E, dtype=torch.int64, device=dev),
+                "ev": [torch.cuda.Event(), torch.cuda.Event()], "slot": 0}
+        slot = st["slot"]
+        st["slot"] = slot ^ 1
+        st["ev"][slot].synchronize()
+        np.copyto(st["np"][slot][0], plan.step_off)
+        np.copyto(st["np"][slot][1], plan.step_len)
+        st["buf"].copy_(st["pin"][slot], non_blocking=True)
+        st["ev"][slot].record()
+        return st["buf"][0], st["buf"][1]
 
     def _robust_clip(self, plan: TrainPlan) -> None:
         """Optional defense: clip client updates to an L2 ball before they
