@@ -63,10 +63,25 @@ class PlanTemplate:
             return TrainPlan(self.rows, np.zeros((0, epochs), np.int64),
                              np.zeros((0, epochs), np.int64), self.sample_num,
                              template=self)
-        u = rng.random((G, epochs))
-        pick = (u * self.pool_size[:, None]).astype(np.int64)
-        idx = self.pool_start[:, None] + pick
-        return TrainPlan(self.rows, self.pool_off[idx], self.pool_len[idx],
+        # scratch reuse: at thousands of pairs the four fresh [G, E]
+        # allocations cost more than the draw itself.  Generator.random
+        # with `out=` consumes the identical bit stream as random(size),
+        # so cached-buffer draws are bit-equal to the uncached path.
+        sc = getattr(self, "_scratch", None)
+        if sc is None or sc[0].shape != (G, epochs):
+            sc = (np.empty((G, epochs), np.float64),
+                  np.empty((G, epochs), np.int64),
+                  np.empty((G, epochs), np.int64),
+                  np.empty((G, epochs), np.int64))
+            self._scratch = sc
+        u, idx, off_o, len_o = sc
+        rng.random(out=u)
+        np.multiply(u, self.pool_size[:, None], out=u)
+        idx[:] = u                          # float -> int64 truncation
+        idx += self.pool_start[:, None]
+        np.take(self.pool_off, idx, out=off_o)
+        np.take(self.pool_len, idx, out=len_o)
+        return TrainPlan(self.rows, off_o, len_o,
                          self.sample_num, template=self)
 
 
