@@ -37,6 +37,15 @@ class Communicator:
         else:
             self.device = torch.device("cpu")
 
+        # per-rank process naming for ps/top (reference uses setproctitle
+        # in main_fedavg.py:264-266; optional here — not in every image)
+        try:
+            import setproctitle
+            setproctitle.setproctitle(
+                f"feddrift-mi355x:rank{self.rank}/{self.world_size}")
+        except ImportError:
+            pass
+
         self.distributed = self.world_size > 1
         if self.distributed and not dist.is_initialized():
             if backend == "auto":
