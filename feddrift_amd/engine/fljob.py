@@ -130,7 +130,11 @@ class TaskList:
         self.n_tasks += 1
         return self.n_tasks - 1
 
-    CHUNK = 128  # split long windows: more workgroups, short latency chains
+    # split long windows: more workgroups, short latency chains.  With the
+    # single-wave eval kernel each block's fixed cost (weight staging +
+    # reduce + atomics) is amortized over CHUNK samples; env-tunable for
+    # same-box A/B of the block-count/blocking trade-off.
+    CHUNK = int(os.environ.get("FEDDRIFT_EVAL_CHUNK", "128"))
 
     def add_windows(self, task_id: int, row: int,
                     windows: Sequence[Tuple[int, int]]) -> None:
