@@ -28,7 +28,15 @@ def main():
             _phase_breakdown(c, steps)
         else:
             rps = run_scale(comm, c, steps=steps, warmup=max(10, steps // 5))
-            print(f"AB {arm} clients={c} rps={rps:.1f}", flush=True)
+            mem = ""
+            try:
+                import torch
+                if torch.cuda.is_available():
+                    mem = (" hbm_gb="
+                           f"{torch.cuda.max_memory_allocated() / 2**30:.2f}")
+            except Exception:
+                pass
+            print(f"AB {arm} clients={c} rps={rps:.1f}{mem}", flush=True)
 
 
 def _phase_breakdown(n_clients, steps):
