@@ -1,22 +1,29 @@
 """Vmap-batched module training: all (client, model) pairs in ONE batched
 autograd step.
 
-For buffer-free modules (CNN_DropOut — convs, dropout, linears) the
-per-pair training loop of ops/module_engine.py collapses into
-torch.func.vmap(grad(...)) over stacked parameters: every pair's forward+
+The per-pair training loop of ops/module_engine.py collapses into
+torch.func.vmap(grad(...)) over stacked state: every pair's forward+
 backward runs as grouped MIOpen convolutions in a single graph, and the
-Adam(amsgrad, wd) update applies to the whole [G, P] block at once (the
-same update math as ops/mlp_torch.py — exact reference semantics).
-Dropout uses randomness='different' (independent masks per pair, as the
-reference's per-process training would draw).
+Adam(amsgrad, wd) update applies to the whole [G, Pp] parameter block at
+once (the same update math as ops/mlp_torch.py — exact reference
+semantics). Dropout uses randomness='different' (independent masks per
+pair, as the reference's per-process training would draw).
 
-Modules with buffers (BatchNorm ResNets) keep the sequential ModuleEngine:
-running-stat updates are stateful and outside vmap's functional model.
-"""
+Covers BOTH buffer-free modules (CNN_DropOut) and BatchNorm models
+(ResNet / MobileNet / DenseNet): the flat row is the FULL state_dict
+(generic_packer.py), parameters and buffers are exposed to
+functional_call as separate batched pytrees, and train-mode
+F.batch_norm updates each pair's running stats IN PLACE through the
+batched buffer views — per-model stats evolve exactly as the
+sequential engine's eager forwards would (verified by parity tests).
+The optimizer update applies only to the parameter positions of the
+flat row (buffers move only through BN's own momentum updates and
+through FedAvg aggregation, which averages full state_dicts like the
+reference — FedAvgEnsAggregatorSoftCluster.py:174-185)."""
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict, List, Optional
 
 import numpy as np
 import torch
@@ -27,57 +34,81 @@ from torch.func import functional_call, grad, vmap
 from ..models.generic_packer import ModulePacker
 from .mlp_torch import _apply_update
 
+_BN_BUFFERS = ("running_mean", "running_var", "num_batches_tracked")
+
 
 def vmap_compatible(module: nn.Module) -> bool:
-    """True when state_dict == parameters (no buffers)."""
-    return len(list(module.buffers())) == 0
+    """True when the module can run on the vmap engine: either buffer-free
+    (state_dict == parameters) or every buffer is a BatchNorm running
+    statistic (updated in place through batched views under vmap)."""
+    for name, _ in module.named_buffers():
+        if not name.rsplit(".", 1)[-1] in _BN_BUFFERS:
+            return False
+    return True
 
 
 class VmapEngine:
     def __init__(self, template: nn.Module, packer: ModulePacker,
                  device: torch.device):
-        assert vmap_compatible(template), "buffers need ModuleEngine"
+        assert vmap_compatible(template), "non-BN buffers need ModuleEngine"
         self.module = template.to(device)
         self.packer = packer
         self.device = device
-        self.names = [n for n, _ in self.module.named_parameters()]
-        self.shapes = [p.shape for _, p in self.module.named_parameters()]
-        self.numels = [p.numel() for _, p in self.module.named_parameters()]
-        # flat layout must equal state_dict order (packer) — true when no
-        # buffers exist and named_parameters follows state_dict order
-        assert self.names == packer.keys
+        # flat layout follows packer.keys (full state_dict order); split
+        # each key into trainable-parameter vs buffer role
+        pnames = {n for n, _ in self.module.named_parameters()}
+        self.keys = packer.keys
+        self.shapes = [packer.shapes[k] for k in self.keys]
+        self.numels = list(packer.numels)
+        self.is_param = [k in pnames for k in self.keys]
+        self.pnames = [k for k in self.keys if k in pnames]
+        self.bnames = [k for k in self.keys if k not in pnames]
+        # flat positions of the trainable parameters (optimizer targets)
+        idx, i = [], 0
+        for k, n in zip(self.keys, self.numels):
+            if k in pnames:
+                idx.append(torch.arange(i, i + n))
+            i += n
+        self.param_idx = (torch.cat(idx) if idx else
+                          torch.empty(0, dtype=torch.int64)).to(device)
+        self.has_buffers = len(self.bnames) > 0
 
-        def loss_fn(param_list, x, y, mask, inv_n):
-            params = dict(zip(self.names, param_list))
-            logits = functional_call(self.module, params, (x,))
+        def loss_fn(plist, blist, x, y, mask, inv_n):
+            tensors = dict(zip(self.pnames + self.bnames,
+                               list(plist) + list(blist)))
+            logits = functional_call(self.module, tensors, (x,))
             ce = F.cross_entropy(logits, y, reduction="none")
             return (ce * mask).sum() * inv_n
 
         self._grad_fn = vmap(grad(loss_fn),
-                             in_dims=(0, 0, 0, 0, 0),
+                             in_dims=(0, 0, 0, 0, 0, 0),
                              randomness="different")
 
-        def fwd_fn(param_list, x):
-            params = dict(zip(self.names, param_list))
-            return functional_call(self.module, params, (x,))
+        def fwd_fn(plist, blist, x):
+            tensors = dict(zip(self.pnames + self.bnames,
+                               list(plist) + list(blist)))
+            return functional_call(self.module, tensors, (x,))
 
-        self._fwd_fn = vmap(fwd_fn, in_dims=(0, 0))
+        self._fwd_fn = vmap(fwd_fn, in_dims=(0, 0, 0))
 
-    def _param_views(self, flat: torch.Tensor):
-        """[G, P] flat -> list of [G, *shape] views (zero copy)."""
-        out = []
+    def _views(self, flat: torch.Tensor):
+        """[G, P] flat -> (param views, buffer views), zero copy, in
+        state_dict order within each role."""
+        pv, bv = [], []
         i = 0
         G = flat.shape[0]
-        for shape, n in zip(self.shapes, self.numels):
-            out.append(flat[:, i:i + n].reshape(G, *shape))
+        for shape, n, isp in zip(self.shapes, self.numels, self.is_param):
+            v = flat[:, i:i + n].reshape(G, *shape)
+            (pv if isp else bv).append(v)
             i += n
-        return out
+        return pv, bv
 
     def make_opt_state(self, kind: str, n_rows: int, lr: float, wd: float):
-        # full-P state (no buffers, so every entry is trainable)
+        # optimizer state covers trainable parameters only (like the
+        # sequential engine; buffers are not optimizer targets)
         from .mlp_torch import make_opt_state
-        return make_opt_state(kind, n_rows, self.packer.n_params, lr, wd,
-                              self.device)
+        return make_opt_state(kind, n_rows, self.packer.n_train_params,
+                              lr, wd, self.device)
 
     # activation memory bound: vmapped conv activations scale with
     # pairs x batch; chunk the pair dimension (pairs are independent, so
@@ -106,7 +137,12 @@ class VmapEngine:
             return
         self.module.train()
         model_of = rows % n_models
-        work = global_params[model_of].clone()          # [G, P]
+        work = global_params[model_of].clone()          # [G, P_full]
+        pv, bv = self._views(work)
+        # BN's in-place running-stat update must write through CONTIGUOUS
+        # batched tensors (an update through a strided view into `work`
+        # does not land); stage buffers out, copy back after the E steps
+        bufs = [v.contiguous() for v in bv]
         step_off = torch.as_tensor(plan.step_off, dtype=torch.int64,
                                    device=self.device)
         step_len = torch.as_tensor(plan.step_len, dtype=torch.int64,
@@ -133,22 +169,41 @@ class VmapEngine:
                 x = x * x_mask.unsqueeze(1)
             y = y_arena[idx.reshape(-1)].reshape(G, bmax)
             inv_n = 1.0 / ln.clamp(min=1).float()
-            grads = self._grad_fn(self._param_views(work), x, y, mask,
-                                  inv_n)
+            # grads come back for the parameter pytree only; BN updates
+            # its running stats through `bufs` in place during the forward
+            zero = (ln == 0)
+            saved = ([b[zero].clone() for b in bufs]
+                     if (self.has_buffers and bool(zero.any())) else None)
+            grads = self._grad_fn(pv, bufs, x, y, mask, inv_n)
+            if saved is not None:
+                # a zero-length step is SKIPPED by reference semantics:
+                # undo the padded forward's running-stat update for those
+                # pairs (their loss was masked; params stay via `sel`)
+                for b, s in zip(bufs, saved):
+                    b[zero] = s
             gflat = torch.cat([g.reshape(G, -1) for g in grads], dim=1)
+            if self.has_buffers:
+                wp = work[:, self.param_idx]
+            else:
+                wp = work
             sel = ln > 0
             if bool(sel.all()):
                 _apply_update(opt["kind"], lr, opt.get("wd", 0.0), st,
-                              work, gflat)
+                              wp, gflat)
             else:
                 sub = sel.nonzero(as_tuple=True)[0]
-                w_sub = work[sub]
+                w_sub = wp[sub]
                 st_sub = {k: v[sub] for k, v in st.items()}
                 _apply_update(opt["kind"], lr[sub], opt.get("wd", 0.0),
                               st_sub, w_sub, gflat[sub])
-                work[sub] = w_sub
+                wp[sub] = w_sub
                 for k in st:
                     st[k][sub] = st_sub[k]
+            if self.has_buffers:
+                work[:, self.param_idx] = wp
+        if self.has_buffers:
+            for v, b in zip(bv, bufs):       # stats back into the flat rows
+                v.copy_(b)
         replicas[rows] = work
         if opt["kind"] == "adam":
             for k in st:
@@ -189,8 +244,9 @@ class VmapEngine:
                 x = x * (xm.unsqueeze(1) if xm.dim() == 2 else xm)
             y = y_arena[idx.reshape(-1)].reshape(Wc, bmax)
             p = params[tr]
-            logits = self._fwd_fn(self._param_views(p),
-                                  x.reshape(Wc, bmax, -1))
+            pv, bv = self._views(p)
+            bv = [v.contiguous() for v in bv]
+            logits = self._fwd_fn(pv, bv, x.reshape(Wc, bmax, -1))
             logits = logits.reshape(Wc, bmax, -1)
             pred = logits.argmax(-1)
             corr = ((pred == y) & mask).sum(dim=1).double()

@@ -10,7 +10,8 @@ import torch
 
 from feddrift_amd.comm import Communicator
 from feddrift_amd.config import Config
-from feddrift_amd.data.generators import sample_femnist, sample_mnist
+from feddrift_amd.data.generators import (sample_cifar, sample_femnist,
+                                           sample_mnist)
 from feddrift_amd.data.loader import DriftDataset
 from feddrift_amd.engine.fljob import FLJob, TrainPlan
 from feddrift_amd.engine.timeline import clean_state_files
@@ -174,3 +175,80 @@ def test_vmap_engine_eval_matches():
     b = eng_s.eval_tasks_stacked(params, tr, ti, wo, wl, 3, want_mse=True,
                                  x_arena=x, y_arena=y)
     assert torch.allclose(a, b, atol=1e-4), (a - b).abs().max()
+
+def _bn_model():
+    from feddrift_amd.models.resnet import FlatImageModel
+    torch.manual_seed(9)
+    backbone = torch.nn.Sequential(
+        torch.nn.Conv2d(1, 4, 3, padding=1), torch.nn.BatchNorm2d(4),
+        torch.nn.ReLU(), torch.nn.Flatten(), torch.nn.Linear(4 * 6 * 6, 3))
+    return FlatImageModel(backbone, (1, 6, 6))
+
+
+def test_vmap_engine_batchnorm_matches_sequential():
+    """BN models run on the vmap engine too: batched train-mode BN updates
+    each pair's running stats in place and must match the sequential
+    engine's eager per-pair training (params AND buffers)."""
+    from feddrift_amd.ops.module_vmap import VmapEngine, vmap_compatible
+    model = _bn_model()
+    assert vmap_compatible(model)
+    packer = ModulePacker(model)
+    assert packer.n_train_params < packer.n_params   # BN buffers present
+    dev = torch.device("cpu")
+
+    n, G, E, K = 240, 4, 3, 2
+    torch.manual_seed(11)
+    x = torch.rand(n, 36)
+    y = torch.randint(0, 3, (n,))
+    glob = torch.stack([packer.flatten(_bn_model().state_dict())
+                        for _ in range(K)])
+    offs = np.array([[0, 60, 120], [60, 0, 180], [120, 180, 0],
+                     [180, 60, 120]])
+    lens = np.full((G, E), 60)
+    lens[1, 2] = 0   # a skipped step
+    plan = TrainPlan(np.arange(G), offs, lens, np.ones((2, K)))
+
+    eng_v = VmapEngine(_bn_model(), packer, dev)
+    reps_v = torch.zeros(G, packer.n_params)
+    opt_v = eng_v.make_opt_state("adam", G, 0.01, 0.001)
+    eng_v.train(glob.clone(), reps_v, plan, opt_v, x, y, n_models=K)
+
+    eng_s = ModuleEngine(_bn_model(), ModulePacker(model), dev)
+    reps_s = torch.zeros(G, packer.n_params)
+    opt_s = eng_s.make_opt_state("adam", G, 0.01, 0.001)
+    eng_s.train(glob.clone(), reps_s, plan, opt_s, x, y, n_models=K)
+
+    assert torch.allclose(reps_v, reps_s, atol=1e-4), \
+        (reps_v - reps_s).abs().max()
+
+    # eval parity (uses the trained running stats through batched views)
+    tr = torch.tensor([0, 1, 2, 3])
+    ti = torch.tensor([0, 1, 2, 0])
+    wo = torch.tensor([0, 60, 120, 180])
+    wl = torch.tensor([60, 60, 60, 60])
+    a = eng_v.eval_tasks_stacked(reps_v, tr, ti, wo, wl, 3, x_arena=x,
+                                 y_arena=y)
+    b = eng_s.eval_tasks_stacked(reps_s, tr, ti, wo, wl, 3, x_arena=x,
+                                 y_arena=y)
+    assert torch.allclose(a, b, atol=1e-4), (a - b).abs().max()
+
+
+def test_resnet_timeline_uses_vmap(tmp_path):
+    """End-to-end: a ResNet drift timeline now routes through the vmap
+    engine (BN supported) and still learns."""
+    from feddrift_amd.ops.module_vmap import VmapEngine
+    ds = _mini_dataset("cifar", sample_cifar, n=48)
+    comm = Communicator()
+    cfg = Config(model="resnet", dataset="cifar", data_dir="/nonexistent",
+                 client_num_in_total=3, client_num_per_round=3,
+                 batch_size=24, lr=0.01, epochs=2, comm_round=2,
+                 total_train_iteration=1, concept_num=2,
+                 concept_drift_algo="softcluster",
+                 concept_drift_algo_arg="mmacc_06", log_dir=str(tmp_path),
+                 report_client=0)
+    clean_state_files(cfg)
+    logger = MetricLogger(str(tmp_path), enabled=True, to_file=False)
+    job = FLJob(cfg, comm, logger, dataset=ds)
+    assert isinstance(job.mod_engine, VmapEngine)
+    job.run()
+    assert np.isfinite(logger.mean("Test/Acc"))
