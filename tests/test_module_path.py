@@ -233,9 +233,10 @@ def test_vmap_engine_batchnorm_matches_sequential():
     assert torch.allclose(a, b, atol=1e-4), (a - b).abs().max()
 
 
-def test_resnet_timeline_uses_vmap(tmp_path):
-    """End-to-end: a ResNet drift timeline now routes through the vmap
-    engine (BN supported) and still learns."""
+def test_resnet_timeline_uses_vmap(tmp_path, monkeypatch):
+    """End-to-end: a ResNet drift timeline routes through the vmap engine
+    when FEDDRIFT_VMAP_BN=1 (BN supported) and still learns."""
+    monkeypatch.setenv("FEDDRIFT_VMAP_BN", "1")
     from feddrift_amd.ops.module_vmap import VmapEngine
     ds = _mini_dataset("cifar", sample_cifar, n=48)
     comm = Communicator()
