@@ -182,14 +182,18 @@ class FLJob:
             from ..ops.module_vmap import VmapEngine, vmap_compatible
             has_buffers = len(list(proto0.buffers())) > 0
             # Buffer-free modules (CNN_DropOut): vmap-batched autograd is
-            # the measured winner.  BN models (ResNet) are SUPPORTED by the
-            # vmap engine too (batched buffer pytrees, parity-tested) but
-            # MIOpen's grouped convs LOSE to sequential eager at the judged
-            # ResNet-18 config (1264 vs 499 ms/round at 20 pairs), so the
-            # sequential engine stays the BN default; FEDDRIFT_VMAP_BN=1
-            # opts in for large fleets where batching amortizes.
+            # the measured winner at every size.  BN models (ResNet) are
+            # supported by the vmap engine too (batched buffer pytrees,
+            # parity-tested), but MIOpen's grouped convs lose to
+            # sequential eager at SMALL fleets (1264 vs 499 ms/round at
+            # 20 pairs) and win at scale (763 vs 1420 ms/round at 100
+            # clients, same accuracy) -> dispatch BN models by fleet
+            # size; FEDDRIFT_VMAP_BN=1/0 forces either way.
+            force = os.environ.get("FEDDRIFT_VMAP_BN")
+            bn_vmap = (force == "1" if force in ("0", "1")
+                       else cfg.client_num_per_round >= 64)
             use_vmap = vmap_compatible(proto0) and (
-                not has_buffers or os.environ.get("FEDDRIFT_VMAP_BN") == "1")
+                not has_buffers or bn_vmap)
             if use_vmap:
                 self.mod_engine = VmapEngine(proto0, self.packer,
                                              self.device)
