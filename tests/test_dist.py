@@ -119,3 +119,51 @@ def test_world2_matches_world1_aue(tmp_path):
         globals()["_WORKER"] = saved
     assert abs(r1[0] - r2[0]) < 1e-6, (r1, r2)
     assert np.allclose(r1, r2, atol=0.02), (r1, r2)
+
+
+_WORKER_RESNET = r"""
+import json, os, sys
+sys.path.insert(0, {repo!r})
+import numpy as np
+from feddrift_amd.comm import Communicator
+from feddrift_amd.config import Config
+from feddrift_amd.engine.timeline import run_timeline
+
+cfg = Config(model="resnet", dataset="cifar", data_dir={data!r},
+             client_num_in_total=4, client_num_per_round=4,
+             batch_size=48, lr=0.01, epochs=1, comm_round=2,
+             total_train_iteration=2, concept_num=2,
+             concept_drift_algo="softcluster",
+             concept_drift_algo_arg="mmacc_06",
+             change_points="T", dummy_arg=0, log_dir={log!r},
+             report_client=0)
+comm = Communicator()
+out = run_timeline(cfg, comm)
+if comm.is_root:
+    with open(os.path.join({log!r}, "result.json"), "w") as f:
+        json.dump(out["per_iteration_test_acc"], f)
+"""
+
+
+def test_world2_matches_world1_module_path(tmp_path):
+    """The module execution path (ResNet through the sequential engine:
+    client-sharded per-pair training, full-state aggregation incl. BN
+    buffers, eval allreduce) must agree across world sizes too."""
+    from feddrift_amd.data.generators import generate_data
+    d = str(tmp_path / "data")
+    os.makedirs(os.path.join(d, "changepoints"), exist_ok=True)
+    mat = np.zeros((3, 4), dtype=int)
+    mat[2:, :2] = 1
+    np.savetxt(os.path.join(d, "changepoints", "T.cp"), mat, fmt="%u")
+    np.random.seed(1)
+    generate_data("cifar", d, 2, 4, 0, 48, 0.0, 1, "T")
+    global _WORKER
+    saved = _WORKER
+    try:
+        globals()["_WORKER"] = _WORKER_RESNET
+        r1 = _run(1, d, str(tmp_path / "m1"), 29616)
+        r2 = _run(2, d, str(tmp_path / "m2"), 29617)
+    finally:
+        globals()["_WORKER"] = saved
+    assert abs(r1[0] - r2[0]) < 1e-5, (r1, r2)
+    assert np.allclose(r1, r2, atol=0.02), (r1, r2)
