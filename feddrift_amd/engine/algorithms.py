@@ -34,8 +34,17 @@ def make(cfg: Config):
     algo = cfg.concept_drift_algo
     if algo in ("aue", "auepc"):
         return AueAlgo(per_client=(algo == "auepc"))
-    if algo == "driftsurf":
+    if algo in ("driftsurf", "dsurf"):   # reference README spells it dsurf
         return DriftSurfAlgo()
+    if algo in ("win-1", "win-2", "all"):
+        # reference cont_one surface: the window is the DRIFT_ALGO itself
+        cfg.retrain_data = algo
+        return SingleAlgo()
+    if algo == "clusterfl":
+        raise NameError(
+            "clusterfl is obsolete in the reference too — use "
+            "concept_drift_algo=softcluster with "
+            "concept_drift_algo_arg=cfl_{gamma}_{win-1|all}")
     if algo in ("mmacc", "mmgeni", "mmgeniex"):
         return MultiModelAlgo(algo)
     if cfg.is_softcluster:

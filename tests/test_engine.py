@@ -160,3 +160,28 @@ def test_serving_from_checkpoint(sea_dir, tmp_path):
                    np.mean(srv.predict(c, x_easy1) == 1)
                    for c in range(6)]) / 2
     assert acc > 0.8
+
+
+def test_reference_algo_aliases():
+    """The reference README's spellings work directly: dsurf, and the
+    window baselines as DRIFT_ALGO strings (cont_one surface)."""
+    from feddrift_amd.engine.algorithms import (DriftSurfAlgo, SingleAlgo,
+                                                make)
+    from feddrift_amd.config import Config
+
+    def cfg_for(algo):
+        return Config(model="fnn", dataset="sea", data_dir="/x",
+                      client_num_in_total=3, client_num_per_round=3,
+                      batch_size=10, lr=0.01, epochs=1, comm_round=1,
+                      total_train_iteration=1, concept_num=2,
+                      concept_drift_algo=algo, concept_drift_algo_arg="",
+                      report_client=0)
+
+    assert isinstance(make(cfg_for("dsurf")), DriftSurfAlgo)
+    for w in ("win-1", "win-2", "all"):
+        c = cfg_for(w)
+        assert isinstance(make(c), SingleAlgo)
+        assert c.retrain_data == w
+    import pytest
+    with pytest.raises(NameError, match="softcluster"):
+        make(cfg_for("clusterfl"))
