@@ -121,14 +121,37 @@ def sample_femnist(n, concept, rng):
     return _sample_image_like(n, concept, rng, 62, 784, 1703)
 
 
+TEXT_SEQ_LEN = 20
+TEXT_VOCAB = 30
+
+
+def sample_text(n: int, concept: int, rng: np.random.Generator) -> np.ndarray:
+    """Concept drift over TEXT: rows are char-id context windows
+    (TEXT_SEQ_LEN ids) + next-char label, drawn from an order-1 Markov
+    chain; a concept change PERMUTES the chain (a 'language shift'), so
+    the drift algorithms see the same staggered-concept structure the
+    numeric datasets have (data/text_synthetic.py holds the chain
+    machinery; CharLSTM in models/rnn.py is the matching model)."""
+    from .text_synthetic import _client_chain, _sample_chain
+    base = np.random.default_rng(4242)
+    # peaky transitions (low Dirichlet concentration): a learnable
+    # language with a high Bayes next-char accuracy
+    chain = _client_chain(base, TEXT_VOCAB, concentration=0.05)
+    if concept != 0:
+        perm = np.random.default_rng(1000 + concept).permutation(TEXT_VOCAB)
+        chain = chain[perm][:, perm]
+    seqs = _sample_chain(rng, chain, n, TEXT_SEQ_LEN)
+    return seqs.astype(np.float64)
+
+
 _SAMPLERS = {"sea": sample_sea, "sine": sample_sine, "circle": sample_circle,
              "MNIST": sample_mnist, "cifar": sample_cifar,
-             "femnist": sample_femnist}
+             "femnist": sample_femnist, "text": sample_text}
 
 FEATURE_NUM = {"sea": 3, "sine": 2, "circle": 2, "MNIST": 784,
-               "cifar": 3072, "femnist": 784}
+               "cifar": 3072, "femnist": 784, "text": TEXT_SEQ_LEN}
 CLASS_NUM = {"sea": 2, "sine": 2, "circle": 2, "MNIST": 10,
-             "cifar": 10, "femnist": 62}
+             "cifar": 10, "femnist": 62, "text": TEXT_VOCAB}
 
 _SEA_COLS = ["f1", "f2", "f3", "label"]
 

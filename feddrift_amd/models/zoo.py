@@ -123,7 +123,11 @@ def create_model(model_name: str, output_dim: int, feature_dim: int) -> nn.Modul
         model = FlatImageModel(backbones[model_name](), shape)
     elif model_name == "rnn":
         from .rnn import CharLSTM
-        model = CharLSTM(vocab_size=max(output_dim, 90))
+        # vocab = the dataset's class count (synthetic text drift uses a
+        # small alphabet; LEAF shakespeare passes 90)
+        model = CharLSTM(vocab_size=max(output_dim, 2),
+                         embedding_dim=8,
+                         hidden_size=64 if output_dim < 90 else 256)
     else:
         raise NameError(model_name)
     reinitialize(model)

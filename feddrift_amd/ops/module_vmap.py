@@ -40,7 +40,13 @@ _BN_BUFFERS = ("running_mean", "running_var", "num_batches_tracked")
 def vmap_compatible(module: nn.Module) -> bool:
     """True when the module can run on the vmap engine: either buffer-free
     (state_dict == parameters) or every buffer is a BatchNorm running
-    statistic (updated in place through batched views under vmap)."""
+    statistic (updated in place through batched views under vmap).
+    Recurrent modules are excluded: aten::lstm has no vmap batching rule
+    on the inference path (RuntimeError under the eval sweep), so RNNs
+    run on the sequential engine."""
+    for m in module.modules():
+        if isinstance(m, nn.RNNBase):
+            return False
     for name, _ in module.named_buffers():
         if not name.rsplit(".", 1)[-1] in _BN_BUFFERS:
             return False
