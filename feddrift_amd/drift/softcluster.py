@@ -114,6 +114,15 @@ class SoftClusterState:
     def cluster_init(self, hooks: EngineHooks) -> None:
         self.train_data_weights[0] = np.zeros((self.model_num, self.client_num))
         if self.h_cluster == "F":
+            # per-client initial models: one model PER CLIENT, so the model
+            # cap must cover the fleet (the reference crashes with a raw
+            # IndexError here, FedAvgEnsDataLoader.py:630; fail clearly)
+            if self.model_num < self.client_num:
+                raise ValueError(
+                    f"H_*_F (per-client init) needs concept_num >= "
+                    f"client_num: got {self.model_num} models for "
+                    f"{self.client_num} clients — raise CONCEPT_NUM (the "
+                    f"model cap) to at least the client count")
             for c in range(self.client_num):
                 self.train_data_weights[0][c][c] = 1.0
             for c in range(self.client_num):

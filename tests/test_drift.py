@@ -216,3 +216,18 @@ def test_cfl_split_on_divergent_updates():
     assert st.cluster_cfl(h, 1, 2, big, clients)
     w = st.train_data_weights[1]
     assert w[0].sum() == 2 and w[1].sum() == 2
+
+
+def test_per_client_init_requires_enough_models():
+    """H_*_F with fewer models than clients must fail with a clear error
+    (the reference crashes with a raw IndexError at the same spot)."""
+    import pytest
+    from feddrift_amd.drift.softcluster import SoftClusterState
+
+    class _NullHooks:
+        def log_client(self, *a): pass
+        def log_summary(self, *a): pass
+
+    st = SoftClusterState(5, 3, "hierarchical", h_cluster="F")
+    with pytest.raises(ValueError, match="concept_num >= client_num"):
+        st.cluster_init(_NullHooks())
