@@ -43,10 +43,18 @@ class MultiModelAccState:
                 break
         for c in range(self.client_num):
             model_acc = {m: score_fn(m, c) for m in self.models}
-            best_model, best_acc = -1, 0.0
+            # best_acc starts below any real accuracy: when EVERY model
+            # scores exactly 0.0 on this client (possible early on hard
+            # tasks) the reference's `best_acc = 0.0` start leaves
+            # best_model at -1 and crashes on the dict access
+            # (FedAvgEnsDataLoader.py:350-390); pick the first model
+            # instead — identical behavior whenever any score is > 0.
+            best_model, best_acc = -1, -1.0
             for m, a in model_acc.items():
                 if a > best_acc:
                     best_acc, best_model = a, m
+            if best_model == -1:       # no models registered yet
+                best_model = 0
             if self.acc_dict[c] - best_acc > self.delta and \
                     next_free_model != -1:
                 best_model = next_free_model

@@ -349,7 +349,7 @@ class SoftClusterAlgo(AlgoBase):
         flattened deltas (small: active pairs only) to every rank
         (reference cluster_cfl, FedAvgEnsDataLoader.py:1159-1223)."""
         st = self.state
-        K, P = job.n_models, job.spec.n_params
+        K, P = job.n_models, job.n_params   # module path: full state size
         local: Dict[tuple, np.ndarray] = {}
         nW = len(job.owned_workers)
         reps = job.replicas.reshape(nW, K, P)
