@@ -231,3 +231,17 @@ def test_per_client_init_requires_enough_models():
     st = SoftClusterState(5, 3, "hierarchical", h_cluster="F")
     with pytest.raises(ValueError, match="concept_num >= client_num"):
         st.cluster_init(_NullHooks())
+
+
+def test_mmacc_all_zero_scores_pick_first_model():
+    """Degenerate scoring (every model 0.0 on a client) must not crash
+    (the reference's best_acc=0.0 start leaves best_model=-1 and dies)."""
+    from feddrift_amd.drift.mmacc import MultiModelAccState
+    st = MultiModelAccState(2, model_num=2, delta=10.0)   # huge delta:
+    st.run_model_select(lambda m, c: 1.0, 0)              # no drift branch
+    st.set_model(0, np.zeros(3))
+    for c in range(2):
+        st.set_acc(c, 0.0)
+    st.run_model_select(lambda m, c: 0.0, 1)
+    assert st.get_train_model_idx(0) == 0
+    assert st.get_test_model_idx(1) == 0

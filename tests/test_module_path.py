@@ -253,3 +253,22 @@ def test_resnet_timeline_uses_vmap(tmp_path, monkeypatch):
     assert isinstance(job.mod_engine, VmapEngine)
     job.run()
     assert np.isfinite(logger.mean("Test/Acc"))
+
+
+def test_cfl_split_on_module_path(tmp_path):
+    """CFL's weight-update gathering must work on the module path too
+    (full-state flat rows, not the MLP spec)."""
+    ds = _mini_dataset("MNIST", sample_mnist, n=64)
+    comm = Communicator()
+    cfg = Config(model="cnn", dataset="MNIST", data_dir="/nonexistent",
+                 client_num_in_total=3, client_num_per_round=3,
+                 batch_size=32, lr=0.05, epochs=2, comm_round=3,
+                 total_train_iteration=1, concept_num=2,
+                 concept_drift_algo="softcluster",
+                 concept_drift_algo_arg="cfl_0.2_all",
+                 log_dir=str(tmp_path), report_client=0)
+    clean_state_files(cfg)
+    logger = MetricLogger(str(tmp_path), enabled=True, to_file=False)
+    job = FLJob(cfg, comm, logger, dataset=ds)
+    job.run()
+    assert np.isfinite(logger.mean("Test/Acc"))
