@@ -167,3 +167,60 @@ def test_world2_matches_world1_module_path(tmp_path):
         globals()["_WORKER"] = saved
     assert abs(r1[0] - r2[0]) < 1e-5, (r1, r2)
     assert np.allclose(r1, r2, atol=0.02), (r1, r2)
+
+
+_WORKER_CNN = r"""
+import json, os, sys
+sys.path.insert(0, {repo!r})
+import numpy as np
+from feddrift_amd.comm import Communicator
+from feddrift_amd.config import Config
+from feddrift_amd.data.generators import sample_mnist
+from feddrift_amd.data.loader import DriftDataset
+from feddrift_amd.engine.fljob import FLJob
+from feddrift_amd.engine.timeline import clean_state_files
+from feddrift_amd.eval.metrics import MetricLogger
+from feddrift_amd.ops.module_vmap import VmapEngine
+
+ds = DriftDataset(data_dir="/nonexistent", dataset="MNIST", num_client=4)
+rng = np.random.default_rng(0)
+for c in range(4):
+    for t in range(3):
+        arr = sample_mnist(48, 0 if t < 2 else 1, rng)
+        ds.store.put(c, t, arr[:, :-1], arr[:, -1])
+cfg = Config(model="cnn", dataset="MNIST", data_dir="/nonexistent",
+             client_num_in_total=4, client_num_per_round=4,
+             batch_size=24, lr=0.01, epochs=2, comm_round=3,
+             total_train_iteration=1, concept_num=2,
+             concept_drift_algo="softcluster",
+             concept_drift_algo_arg="mmacc_06", log_dir={log!r},
+             report_client=0)
+comm = Communicator()
+if comm.is_root:
+    clean_state_files(cfg)
+comm.barrier()
+logger = MetricLogger({log!r}, enabled=comm.is_root, to_file=False)
+job = FLJob(cfg, comm, logger, dataset=ds)
+assert isinstance(job.mod_engine, VmapEngine), type(job.mod_engine)
+job.run()
+if comm.is_root:
+    acc = logger.mean("Test/Acc")
+    assert np.isfinite(acc), acc
+    with open(os.path.join({log!r}, "result.json"), "w") as f:
+        json.dump([acc], f)
+"""
+
+
+def test_world2_cnn_vmap_engine(tmp_path):
+    """The vmap module engine under client sharding (each rank trains its
+    owned pairs in its own batched autograd step) must run and produce
+    finite metrics at world size 2."""
+    global _WORKER
+    saved = _WORKER
+    try:
+        globals()["_WORKER"] = _WORKER_CNN
+        # _run's data arg is unused by this worker; pass the log dir
+        r2 = _run(2, str(tmp_path), str(tmp_path / "c2"), 29618)
+    finally:
+        globals()["_WORKER"] = saved
+    assert np.isfinite(r2[0])
