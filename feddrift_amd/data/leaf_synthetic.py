@@ -10,7 +10,7 @@ heterogeneity. Sample counts follow a log-normal power law.
 
 from __future__ import annotations
 
-from typing import Dict, List, Tuple
+from typing import Dict, Tuple
 
 import numpy as np
 

@@ -15,7 +15,7 @@ for tests and benchmarks.
 from __future__ import annotations
 
 import csv
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Sequence, Tuple
 
 import numpy as np
 import torch

@@ -12,7 +12,7 @@ per-neighbor message exchange becomes a single [W, W] x [W, P] GEMM
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import List, Optional, Sequence, Tuple
 
 import numpy as np
 import torch

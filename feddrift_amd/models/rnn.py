@@ -13,7 +13,6 @@ flat data layer and cast inside forward).
 
 from __future__ import annotations
 
-import torch
 from torch import nn
 
 

@@ -9,7 +9,6 @@ GPU present.
 from __future__ import annotations
 
 import os
-import shutil
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
 SRC = os.path.join(_DIR, "hip", "feddrift_kernels.hip")

@@ -11,7 +11,7 @@ All tensors live on one device; nothing moves host<->device inside a round.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, Optional
 
 import torch
 import torch.nn.functional as F

@@ -23,9 +23,8 @@ reference — FedAvgEnsAggregatorSoftCluster.py:174-185)."""
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
-import numpy as np
 import torch
 import torch.nn.functional as F
 from torch import nn
