@@ -82,7 +82,11 @@ def test_serve_api(tmp_path):
     run_timeline(cfg)
 
     from feddrift_amd.engine.serve import DriftModelServer
-    from starlette.testclient import TestClient
+    import warnings
+    with warnings.catch_warnings():
+        # third-party: starlette's own testclient deprecation notice
+        warnings.simplefilter("ignore")
+        from starlette.testclient import TestClient
     srv = DriftModelServer(cfg, str(tmp_path))
     app = serve_api.build_app(srv)
     client = TestClient(app)
