@@ -156,8 +156,11 @@ class FLJob:
         self.cfg = cfg
         self.comm = comm
         self.device = comm.device
-        self.logger = logger or MetricLogger(cfg.log_dir,
-                                             enabled=comm.is_root)
+        self.logger = logger or MetricLogger(
+            cfg.log_dir, enabled=comm.is_root,
+            use_wandb=bool(cfg.wandb),
+            run_name=f"FedAvgCont-{cfg.dataset}-{cfg.concept_drift_algo}"
+                     f"-iter{cfg.curr_train_iteration}")
         self.backend = ops.backend_for(self.device, cfg.use_hip_kernels)
 
         # seeds: np partition/batch RNG + torch init seed keyed on dummy_arg

@@ -50,7 +50,10 @@ def run_timeline(cfg: Config, comm: Optional[Communicator] = None,
         icfg = dataclasses.replace(cfg, curr_train_iteration=it)
         icfg.__post_init__()
         logger = (loggers[it] if loggers else
-                  MetricLogger(cfg.log_dir, enabled=comm.is_root))
+                  MetricLogger(cfg.log_dir, enabled=comm.is_root,
+                               use_wandb=bool(cfg.wandb),
+                               run_name=f"FedAvgCont-{cfg.dataset}-"
+                                        f"{cfg.concept_drift_algo}-iter{it}"))
         job = FLJob(icfg, comm, logger)
         job.run()
         s = logger.series("Test/Acc")
