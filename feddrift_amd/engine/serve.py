@@ -69,7 +69,7 @@ class DriftModelServer:
             st = pkl("mm_state.pkl")
             return np.array([min(st.get_test_model_idx(c),
                                  self.n_models - 1) for c in range(C)])
-        if algo == "driftsurf":
+        if algo in ("driftsurf", "dsurf"):
             st = pkl("ds_state.pkl")
             idx = 0
             for i, key in enumerate(st.get_train_keys()):
