@@ -71,3 +71,11 @@ class MetricLogger:
         if self._fh is not None:
             self._fh.close()
             self._fh = None
+        if self._wandb is not None:
+            # one wandb run per iteration, like the reference's
+            # per-iteration runs (main_fedavg.py:280-290)
+            try:
+                self._wandb.finish()
+            except Exception:  # noqa: BLE001
+                pass
+            self._wandb = None
