@@ -48,6 +48,9 @@ class Config:
     change_points: str = "rand"
     time_stretch: int = 1
     reset_models: int = 0
+    # data-GENERATION surface args (consumed by scripts/prepare_data.py /
+    # generate_data at data-prep time, like the reference's main_fedavg
+    # arg list; the engine itself reads the generated CSVs)
     noise_prob: float = 0.0
     dummy_arg: int = 0
     sample_num: int = 500                # samples per (client, iteration)
