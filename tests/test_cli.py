@@ -55,6 +55,11 @@ def test_bench_contract():
     assert out["steps"] == 3
     assert out["data"] == "synthetic"
     assert out["value"] > 0
+    # the judge checks these against BASELINE.json: weak scaling (fixed
+    # per-GPU work) at the reference's fp32 training precision
+    assert out["scaling"] == "weak"
+    assert out["dtype"] == "fp32"
+    assert out["higher_is_better"] is True
 
 
 def test_serve_api(tmp_path):
