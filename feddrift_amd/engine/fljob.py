@@ -578,7 +578,11 @@ class FLJob:
         proto = zoo.create_model(self.cfg.model, self.dataset.class_num,
                                  self.dataset.feature_num)
         for m in range(self.n_models):
-            for layer in proto.children():
+            # recurse modules() (not children()) so wrapped backbones
+            # (FlatImageModel -> ResNet blocks) actually reset; children()
+            # missed every conv/bn inside the wrapper and left all K models
+            # bit-identical on the image paths.
+            for layer in proto.modules():
                 if hasattr(layer, "reset_parameters"):
                     layer.reset_parameters()
             self.global_params[m] = self.packer.flatten(

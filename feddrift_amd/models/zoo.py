@@ -65,9 +65,12 @@ class CNN_DropOut(nn.Module):
         self.softmax = nn.Softmax(dim=1)
 
     def forward(self, x):
+        # reference forward (cv/cnn.py:126-135): NO activation after either
+        # conv — conv1 -> conv2 -> maxpool; the relu member is only used on
+        # linear_1. Quirk preserved exactly.
         x = torch.unsqueeze(x.reshape(-1, 28, 28), 1)
-        x = self.relu(self.conv2d_1(x))
-        x = self.relu(self.conv2d_2(x))
+        x = self.conv2d_1(x)
+        x = self.conv2d_2(x)
         x = self.max_pooling(x)
         x = self.dropout_1(x)
         x = self.flatten(x)
