@@ -184,6 +184,17 @@ class FLJob:
             self.spec = None
             from ..ops.module_vmap import VmapEngine, vmap_compatible
             has_buffers = len(list(proto0.buffers())) > 0
+            # CNN_DropOut on a GPU: the hand-written CDNA4 kernel engine
+            # (ops/cnn_hip.py, ops/hip/cnn_kernels.hip) owns the hot path
+            # end-to-end. Required on GPU devices unless explicitly
+            # disabled (FEDDRIFT_CNN_HIP=0 forces the vmap/MIOpen engine
+            # for A/B) — a missing .so fails loudly rather than silently
+            # falling back (ops/hip_loader.py policy).
+            from ..ops.cnn_hip import CnnHipEngine, is_cnn_dropout
+            use_cnn_hip = (self.device.type == "cuda"
+                           and is_cnn_dropout(proto0)
+                           and os.environ.get("FEDDRIFT_CNN_HIP") != "0"
+                           and cfg.use_hip_kernels != "never")
             # Buffer-free modules (CNN_DropOut): vmap-batched autograd is
             # the measured winner at every size.  BN models (ResNet) are
             # supported by the vmap engine too (batched buffer pytrees,
@@ -197,7 +208,10 @@ class FLJob:
                        else cfg.client_num_per_round >= 64)
             use_vmap = vmap_compatible(proto0) and (
                 not has_buffers or bn_vmap)
-            if use_vmap:
+            if use_cnn_hip:
+                self.mod_engine = CnnHipEngine(proto0, self.packer,
+                                               self.device)
+            elif use_vmap:
                 self.mod_engine = VmapEngine(proto0, self.packer,
                                              self.device)
             else:
@@ -329,6 +343,13 @@ class FLJob:
                 self.spec, self.global_params, weights, self.arena.x,
                 self.arena.y, idx[1], idx[2], idx[3], tl.n_tasks,
                 mode=mode, masks=masks)
+        if hasattr(self.mod_engine, "vote_multi"):
+            # batched kernel path (CnnHipEngine): one dump sweep per active
+            # model, vote accumulation in batched torch ops — replaces the
+            # per-task Python loop below
+            return self.mod_engine.vote_multi(
+                self.global_params, weights, self.arena.x, self.arena.y,
+                idx[1], idx[2], idx[3], tl.n_tasks, mode=mode, masks=masks)
         out = torch.zeros(2, tl.n_tasks, dtype=torch.float64,
                           device=self.device)
         tid = idx[1].cpu().numpy()

@@ -12,6 +12,7 @@ import os
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
 SRC = os.path.join(_DIR, "hip", "feddrift_kernels.hip")
+SRC_CNN = os.path.join(_DIR, "hip", "cnn_kernels.hip")
 BUILD_DIR = os.path.join(_DIR, "hip", "_build")
 MODULE_NAME = "feddrift_hip"
 
@@ -22,7 +23,7 @@ def build(verbose: bool = False):
     from torch.utils.cpp_extension import load
     mod = load(
         name=MODULE_NAME,
-        sources=[SRC],
+        sources=[SRC, SRC_CNN],
         build_directory=BUILD_DIR,
         extra_cflags=["-O3"],
         extra_cuda_cflags=["-O3"],

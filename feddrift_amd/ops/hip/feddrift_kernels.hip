@@ -927,9 +927,12 @@ void apply_aggregate_hip(torch::Tensor global_params, torch::Tensor partial,
   TORCH_CHECK(hipGetLastError() == hipSuccess, "apply_aggregate launch");
 }
 
+void register_cnn(pybind11::module_& mod);  // cnn_kernels.hip
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("train_fused", &train_fused_hip, "fused batched MLP local training");
   mod.def("eval_tasks", &eval_tasks_hip, "batched MLP accuracy/loss sweep");
   mod.def("apply_aggregate", &apply_aggregate_hip,
           "masked weighted-average model update");
+  register_cnn(mod);
 }

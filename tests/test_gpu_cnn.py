@@ -1,0 +1,200 @@
+"""GPU numerics tests for the hand-written CDNA4 CNN kernels
+(ops/hip/cnn_kernels.hip via ops/cnn_hip.CnnHipEngine).
+
+Every test compares the HIP pipeline against the torch fp32 engines
+(VmapEngine / ModuleEngine) that are themselves parity-tested against
+eager autograd on CPU (tests/test_module_path.py). Dropout is disabled
+for exact comparisons (the kernel's hash-based masks are statistically,
+not bitwise, equivalent to torch RNG draws); a separate test checks the
+mask statistics.
+"""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():          # pragma: no cover
+    pytest.skip("GPU only", allow_module_level=True)
+
+from feddrift_amd.engine.fljob import TrainPlan
+from feddrift_amd.models import zoo
+from feddrift_amd.models.generic_packer import ModulePacker
+from feddrift_amd.ops.cnn_hip import CnnHipEngine
+from feddrift_amd.ops.module_engine import ModuleEngine
+from feddrift_amd.ops.module_vmap import VmapEngine
+
+DEV = torch.device("cuda:0")
+
+
+def make_setup(seed=0, n_models=3, n_workers=2, E=3, batch=6, O=10):
+    torch.manual_seed(seed)
+    rng = np.random.default_rng(seed)
+    proto = zoo.CNN_DropOut(only_digits=(O == 10))
+    packer = ModulePacker(proto)
+    P = packer.n_params
+    K = n_models
+    gp = torch.randn(K, P, device=DEV) * 0.05
+    N = 400
+    x = torch.randn(N, 784, device=DEV)
+    y = torch.from_numpy(rng.integers(0, O, N)).to(DEV)
+    G = n_workers * K
+    rows = np.arange(G, dtype=np.int64)
+    step_off = rng.integers(0, N - batch, (G, E)).astype(np.int64)
+    step_len = rng.integers(1, batch + 1, (G, E)).astype(np.int64)
+    step_len[1, 1] = 0      # a skipped step (reference skips n==0)
+    plan = TrainPlan(rows, step_off, step_len,
+                     np.ones((n_workers, K), dtype=np.float32))
+    return proto, packer, gp, x, y, plan, K, G, P
+
+
+def fresh_engines(proto, packer, dropout=False):
+    import copy
+    p1 = copy.deepcopy(proto)
+    p2 = copy.deepcopy(proto)
+    hipE = CnnHipEngine(p1, packer, DEV)
+    if not dropout:
+        hipE.dropout_override = (0.0, 0.0)
+        p2.dropout_1.p = 0.0
+        p2.dropout_2.p = 0.0
+    vmapE = VmapEngine(p2, packer, DEV)
+    return hipE, vmapE
+
+
+@pytest.mark.parametrize("opt_kind", ["adam", "sgd"])
+def test_cnn_train_parity(opt_kind):
+    proto, packer, gp, x, y, plan, K, G, P = make_setup()
+    hipE, vmapE = fresh_engines(proto, packer)
+    res = {}
+    for name, eng in (("hip", hipE), ("vmap", vmapE)):
+        reps = torch.zeros(G, P, device=DEV)
+        opt = eng.make_opt_state(opt_kind, G, 0.03, 1e-3)
+        eng.train(gp.clone(), reps, plan, opt, x, y, K)
+        torch.cuda.synchronize()
+        res[name] = (reps.clone(), {k: v.clone() for k, v in opt.items()
+                                    if isinstance(v, torch.Tensor)})
+    r_h, o_h = res["hip"]
+    r_v, o_v = res["vmap"]
+    err = (r_h - r_v).abs().max().item()
+    assert err < 1e-4, f"param mismatch {err}"
+    if opt_kind == "adam":
+        assert torch.equal(o_h["t"], o_v["t"])
+        for k in ("m", "v", "vmax"):
+            e = (o_h[k] - o_v[k]).abs().max().item()
+            assert e < 1e-4, f"{k} mismatch {e}"
+
+
+def test_cnn_train_with_mask():
+    proto, packer, gp, x, y, plan, K, G, P = make_setup(seed=3)
+    rng = np.random.default_rng(7)
+    xm = torch.from_numpy(
+        (rng.random((G, 784)) > 0.3).astype(np.float32)).to(DEV)
+    hipE, vmapE = fresh_engines(proto, packer)
+    res = {}
+    for name, eng in (("hip", hipE), ("vmap", vmapE)):
+        reps = torch.zeros(G, P, device=DEV)
+        opt = eng.make_opt_state("adam", G, 0.03, 1e-3)
+        eng.train(gp.clone(), reps, plan, opt, x, y, K, x_mask=xm)
+        torch.cuda.synchronize()
+        res[name] = reps.clone()
+    err = (res["hip"] - res["vmap"]).abs().max().item()
+    assert err < 1e-4, err
+
+
+def test_cnn_train_chunked_pairs():
+    proto, packer, gp, x, y, plan, K, G, P = make_setup(seed=5)
+    hipE, _ = fresh_engines(proto, packer)
+    reps_a = torch.zeros(G, P, device=DEV)
+    opt_a = hipE.make_opt_state("adam", G, 0.03, 1e-3)
+    hipE.train(gp.clone(), reps_a, plan, opt_a, x, y, K)
+    # force chunking (pairs are independent -> bit-identical results)
+    hipE2, _ = fresh_engines(proto, packer)
+    hipE2.WS_BUDGET = 1  # one pair per chunk
+    reps_b = torch.zeros(G, P, device=DEV)
+    opt_b = hipE2.make_opt_state("adam", G, 0.03, 1e-3)
+    hipE2.train(gp.clone(), reps_b, plan, opt_b, x, y, K)
+    torch.cuda.synchronize()
+    assert torch.equal(reps_a, reps_b)
+
+
+def _mk_tasks(rng, n_tasks, n_win, N, K, max_len=40):
+    task_row, task_id, off, ln = [], [], [], []
+    for w in range(n_win):
+        task_row.append(rng.integers(0, K))
+        task_id.append(rng.integers(0, n_tasks))
+        l = int(rng.integers(1, max_len))
+        off.append(int(rng.integers(0, N - l)))
+        ln.append(l)
+    t = lambda a: torch.as_tensor(a, dtype=torch.int64, device=DEV)
+    return t(task_row), t(task_id), t(off), t(ln)
+
+
+def test_cnn_eval_parity():
+    proto, packer, gp, x, y, plan, K, G, P = make_setup(seed=11)
+    hipE, vmapE = fresh_engines(proto, packer)
+    rng = np.random.default_rng(2)
+    tr, ti, off, ln = _mk_tasks(rng, n_tasks=5, n_win=12, N=x.shape[0], K=K)
+    a = hipE.eval_tasks_stacked(gp, tr, ti, off, ln, 5, want_mse=True,
+                                x_arena=x, y_arena=y)
+    b = vmapE.eval_tasks_stacked(gp, tr, ti, off, ln, 5, want_mse=True,
+                                 x_arena=x, y_arena=y)
+    torch.cuda.synchronize()
+    assert torch.equal(a[0], b[0]), "correct counts differ"
+    assert torch.equal(a[1], b[1]), "totals differ"
+    assert (a[2] - b[2]).abs().max().item() < 1e-3, "loss"
+    assert (a[3] - b[3]).abs().max().item() < 1e-3, "mse"
+    # chunked sweep identical
+    hipE.EVAL_SLOT_BUDGET = 8
+    c = hipE.eval_tasks_stacked(gp, tr, ti, off, ln, 5, want_mse=True,
+                                x_arena=x, y_arena=y)
+    assert torch.allclose(a, c)
+
+
+def test_cnn_confusion_parity():
+    proto, packer, gp, x, y, plan, K, G, P = make_setup(seed=13)
+    import copy
+    hipE, _ = fresh_engines(proto, packer)
+    modE = ModuleEngine(copy.deepcopy(proto), packer, DEV)
+    rng = np.random.default_rng(4)
+    tr, ti, off, ln = _mk_tasks(rng, n_tasks=4, n_win=8, N=x.shape[0], K=K)
+    a = hipE.confusion_tasks(gp, x, y, tr, ti, off, ln, 4, 10)
+    b = modE.confusion_tasks(gp, x, y, tr, ti, off, ln, 4, 10)
+    torch.cuda.synchronize()
+    assert torch.equal(a, b)
+
+
+@pytest.mark.parametrize("mode", ["hard", "soft"])
+def test_cnn_vote_parity(mode):
+    proto, packer, gp, x, y, plan, K, G, P = make_setup(seed=17)
+    import copy
+    hipE, _ = fresh_engines(proto, packer)
+    modE = ModuleEngine(copy.deepcopy(proto), packer, DEV)
+    weights = torch.tensor([0.5, 0.0, 1.5], device=DEV)
+    windows = [(0, 30), (60, 25)]
+    c_h, n_h = hipE.ens_vote_eval(gp, weights, x, y, windows, mode=mode)
+    c_m, n_m = modE.ens_vote_eval(gp, weights, x, y, windows, mode=mode)
+    assert n_h == n_m
+    assert abs(c_h - c_m) <= 1e-6, (c_h, c_m)
+
+
+def test_cnn_dropout_statistics():
+    """Mask hash: keep-rate and scaling are statistically right — train
+    with dropout on, check the update isn't degenerate and differs from
+    the dropout-off run."""
+    proto, packer, gp, x, y, plan, K, G, P = make_setup(seed=19)
+    hipE, _ = fresh_engines(proto, packer, dropout=True)   # p=.25/.5 live
+    reps = torch.zeros(G, P, device=DEV)
+    opt = hipE.make_opt_state("adam", G, 0.03, 1e-3)
+    hipE.train(gp.clone(), reps, plan, opt, x, y, K)
+    hipE2, _ = fresh_engines(proto, packer)                 # dropout off
+    reps2 = torch.zeros(G, P, device=DEV)
+    opt2 = hipE2.make_opt_state("adam", G, 0.03, 1e-3)
+    hipE2.train(gp.clone(), reps2, plan, opt2, x, y, K)
+    torch.cuda.synchronize()
+    assert torch.isfinite(reps).all()
+    d = (reps - reps2).abs().max().item()
+    assert d > 1e-6, "dropout had no effect"
+    # params should still be in the same ballpark (scaled masks)
+    assert (reps - gp[torch.as_tensor(plan.rows, device=DEV) % K]
+            ).abs().max().item() < 1.0
