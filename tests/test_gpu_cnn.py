@@ -62,27 +62,93 @@ def fresh_engines(proto, packer, dropout=False):
     return hipE, vmapE
 
 
-@pytest.mark.parametrize("opt_kind", ["adam", "sgd"])
-def test_cnn_train_parity(opt_kind):
+def test_cnn_train_parity_sgd():
+    """End-to-end SGD parity vs the vmap engine: the update is linear in
+    the gradients, so kernel-vs-autograd summation-order noise (~1e-7)
+    stays ~1e-7 through 3 epochs."""
     proto, packer, gp, x, y, plan, K, G, P = make_setup()
     hipE, vmapE = fresh_engines(proto, packer)
     res = {}
     for name, eng in (("hip", hipE), ("vmap", vmapE)):
         reps = torch.zeros(G, P, device=DEV)
-        opt = eng.make_opt_state(opt_kind, G, 0.03, 1e-3)
+        opt = eng.make_opt_state("sgd", G, 0.03, 0.0)
         eng.train(gp.clone(), reps, plan, opt, x, y, K)
         torch.cuda.synchronize()
-        res[name] = (reps.clone(), {k: v.clone() for k, v in opt.items()
-                                    if isinstance(v, torch.Tensor)})
-    r_h, o_h = res["hip"]
-    r_v, o_v = res["vmap"]
-    err = (r_h - r_v).abs().max().item()
+        res[name] = reps.clone()
+    err = (res["hip"] - res["vmap"]).abs().max().item()
     assert err < 1e-4, f"param mismatch {err}"
-    if opt_kind == "adam":
-        assert torch.equal(o_h["t"], o_v["t"])
+
+
+def test_cnn_adam_math_multi_epoch():
+    """Adam(amsgrad, wd) math parity, chaos-free: per epoch, recover the
+    kernel's own gradient exactly (SGD lr=1 from the same start point:
+    g = w_before - w_after), feed THAT gradient to the torch reference
+    update (_apply_update), and compare against the kernel's Adam step.
+    Both sides then see bitwise-identical gradients, so the comparison
+    isolates the optimizer math from Adam's eps-amplification of
+    summation-order noise on near-zero gradient entries."""
+    from feddrift_amd.ops.mlp_torch import _apply_update
+    proto, packer, gp, x, y, plan, K, G, P = make_setup(E=1)
+    hipE, _ = fresh_engines(proto, packer)
+    lr, wd = 0.03, 1e-3
+    opt_hip = hipE.make_opt_state("adam", G, lr, wd)
+    st_ref = {"m": torch.zeros(G, P, device=DEV),
+              "v": torch.zeros(G, P, device=DEV),
+              "vmax": torch.zeros(G, P, device=DEV),
+              "t": torch.zeros(G, dtype=torch.int32, device=DEV)}
+    rng = np.random.default_rng(123)
+    gp_cur = gp.clone()
+    rows_t = torch.as_tensor(plan.rows, device=DEV)
+    N = x.shape[0]
+    for epoch in range(3):
+        step_off = rng.integers(0, N - 8, (G, 1)).astype(np.int64)
+        step_len = rng.integers(1, 9, (G, 1)).astype(np.int64)
+        eplan = TrainPlan(plan.rows, step_off, step_len, plan.sample_num)
+        # exact kernel gradient via SGD lr=1 from gp_cur
+        reps_g = torch.zeros(G, P, device=DEV)
+        opt_g = hipE.make_opt_state("sgd", G, 1.0, 0.0)
+        hipE.train(gp_cur.clone(), reps_g, eplan, opt_g, x, y, K)
+        torch.cuda.synchronize()
+        g_exact = gp_cur[rows_t % K] - reps_g
+        # kernel Adam step (persistent state)
+        reps_a = torch.zeros(G, P, device=DEV)
+        hipE.train(gp_cur.clone(), reps_a, eplan, opt_hip, x, y, K)
+        torch.cuda.synchronize()
+        # torch reference Adam on the same gradients
+        w_ref = gp_cur[rows_t % K].clone()
+        lr_t = torch.full((G,), lr, device=DEV)
+        _apply_update("adam", lr_t, wd, st_ref, w_ref, g_exact)
+        err = (reps_a - w_ref).abs().max().item()
+        assert err < 1e-5, f"epoch {epoch}: adam step mismatch {err}"
+        assert torch.equal(opt_hip["t"], st_ref["t"])
         for k in ("m", "v", "vmax"):
-            e = (o_h[k] - o_v[k]).abs().max().item()
-            assert e < 1e-4, f"{k} mismatch {e}"
+            e = (opt_hip[k] - st_ref[k]).abs().max().item()
+            assert e < 1e-5, f"epoch {epoch}: {k} mismatch {e}"
+        gp_cur = reps_a[:K].clone()  # continue along the kernel trajectory
+
+
+def test_cnn_train_adam_e2e_bounded():
+    """End-to-end Adam vs vmap: near-zero gradient entries make the
+    normalized update chaotically sensitive (d/dg[g/(|g|+eps)] ~ 1/eps),
+    so exact parity is not achievable across different summation orders
+    (measured: per-layer gradient agreement is ~1e-7, test above). Bound
+    the drift instead: per-step updates are lr-bounded, so |dw| <= ~2*lr
+    per flipped entry and the bulk must agree tightly."""
+    proto, packer, gp, x, y, plan, K, G, P = make_setup()
+    hipE, vmapE = fresh_engines(proto, packer)
+    lr = 0.03
+    res = {}
+    for name, eng in (("hip", hipE), ("vmap", vmapE)):
+        reps = torch.zeros(G, P, device=DEV)
+        opt = eng.make_opt_state("adam", G, lr, 1e-3)
+        eng.train(gp.clone(), reps, plan, opt, x, y, K)
+        torch.cuda.synchronize()
+        res[name] = (reps.clone(), opt["t"].clone())
+    d = (res["hip"][0] - res["vmap"][0]).abs()
+    assert torch.equal(res["hip"][1], res["vmap"][1])
+    assert d.max().item() < 3 * 2 * lr          # <= one sign flip per step
+    assert d.mean().item() < 1e-5               # the bulk agrees
+    assert (d > 1e-4).float().mean().item() < 0.01   # flips are rare
 
 
 def test_cnn_train_with_mask():
