@@ -43,7 +43,8 @@ def make_setup(seed=0, n_models=3, n_workers=2, E=3, batch=6, O=10):
     rows = np.arange(G, dtype=np.int64)
     step_off = rng.integers(0, N - batch, (G, E)).astype(np.int64)
     step_len = rng.integers(1, batch + 1, (G, E)).astype(np.int64)
-    step_len[1, 1] = 0      # a skipped step (reference skips n==0)
+    if E > 1:
+        step_len[1, 1] = 0  # a skipped step (reference skips n==0)
     plan = TrainPlan(rows, step_off, step_len,
                      np.ones((n_workers, K), dtype=np.float32))
     return proto, packer, gp, x, y, plan, K, G, P
@@ -160,7 +161,9 @@ def test_cnn_train_with_mask():
     res = {}
     for name, eng in (("hip", hipE), ("vmap", vmapE)):
         reps = torch.zeros(G, P, device=DEV)
-        opt = eng.make_opt_state("adam", G, 0.03, 1e-3)
+        # SGD: linear in the gradients, so parity is tight (the Adam
+        # eps-amplification caveat of the tests above applies here too)
+        opt = eng.make_opt_state("sgd", G, 0.03, 0.0)
         eng.train(gp.clone(), reps, plan, opt, x, y, K, x_mask=xm)
         torch.cuda.synchronize()
         res[name] = reps.clone()
