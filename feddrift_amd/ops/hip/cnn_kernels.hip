@@ -540,8 +540,10 @@ void cnn_opt_step(CnnArgs a) {
       continue;
     }
     const int tnew = a.t[row] + 1;  // tick kernel commits after
-    const float bc1 = 1.f - __powf(b1, (float)tnew);
-    const float bc2 = 1.f - __powf(b2, (float)tnew);
+    // precise powf: __powf's fast-math error lands ~1e-5 off the torch
+    // reference through the bias corrections (measured on the box)
+    const float bc1 = 1.f - powf(b1, (float)tnew);
+    const float bc2 = 1.f - powf(b2, (float)tnew);
     const long long gp = row * a.P + p;
     const float gr = gr0 + a.wd * wv;
     const float mn = b1 * a.m[gp] + (1.f - b1) * gr;
