@@ -76,13 +76,14 @@ def main():
     p.add_argument("--full", action="store_true")
     p.add_argument("--skip", default="")
     a = p.parse_args()
+    skip = set(a.skip.split(","))
     comm = Communicator()
     results = {}
     tmp = "/tmp/cfg_runs"
     os.makedirs(tmp, exist_ok=True)
 
     # -- config 3: MNIST label-swap CNN, FedDrift-Eager -----------------
-    if "mnist" not in a.skip:
+    if "mnist" not in skip:
         iters = 6 if not a.full else 10
         ds = build_ds("MNIST", sample_mnist, 10, iters, 200,
                       lambda c, t: (c % 4) if t >= 3 else 0)
@@ -97,7 +98,7 @@ def main():
         run_cfg("mnist_cnn_feddrift_eager", cfg, ds, comm, results)
 
     # -- config 4: CIFAR-shaped ResNet-18, IFCA --------------------------
-    if "cifar" not in a.skip:
+    if "cifar" not in skip:
         n_cl = 20 if not a.full else 100
         iters = 3
         ds = build_ds("cifar", sample_cifar, n_cl, iters, 128,
@@ -114,7 +115,7 @@ def main():
         run_cfg("cifar_resnet18_ifca", cfg, ds, comm, results)
 
     # -- config 5: FEMNIST-scale AUE ensemble, 3400 clients --------------
-    if "femnist" not in a.skip:
+    if "femnist" not in skip:
         n_cl = 3400 if not a.full else 3400
         iters = 3
         ds = build_ds("femnist", sample_femnist, n_cl, iters, 100,
@@ -130,7 +131,7 @@ def main():
         run_cfg("femnist3400_aue_ensemble", cfg, ds, comm, results)
 
     # -- config 5b: FEMNIST CNN ensemble (vmap-batched, 400 clients) -----
-    if "femnistcnn" not in a.skip:
+    if "femnistcnn" not in skip:
         n_cl = 400
         iters = 2
         ds = build_ds("femnist", sample_femnist, n_cl, iters, 100,
