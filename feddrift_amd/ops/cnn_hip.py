@@ -95,6 +95,7 @@ class CnnHipEngine:
             "da2": f(G * B * NF),
             "dz2c": f(G * B * Z2N),
             "dx1": f(G * B * X1N),
+            "c1part": f(G * B * 320),
             "grad": f(G, self.P),
         }
         self._ws = ws
@@ -155,6 +156,7 @@ class CnnHipEngine:
                 step_off, step_len, e, xm,
                 ws["x1"], ws["a2"], ws["pidx"], ws["z1"], ws["a1"],
                 ws["dz2"], ws["dz1"], ws["da2"], ws["dz2c"], ws["dx1"],
+                ws["c1part"],
                 opt["m"] if adam else None,
                 opt["v"] if adam else None,
                 opt["vmax"] if adam else None,
@@ -170,7 +172,11 @@ class CnnHipEngine:
 
     def _eval_sweep(self, params, task_row, task_id, win_off, win_len,
                     n_tasks, mode, want_mse=False, x_mask=None,
-                    out_accum=None, dump_sink=None):
+                    dump_sink=None):
+        """Chunked batched sweep. Windows are sorted by model row (stable)
+        so the fc1 MFMA GEMM can group slots per weight row; accuracy/
+        confusion accumulation is order-invariant, and the DUMP sink
+        receives the chunk's (sorted) window arrays plus per-slot ids."""
         dev = self.device
         W = task_row.numel()
         if mode == EV_ACC:
@@ -183,11 +189,16 @@ class CnnHipEngine:
             total_out = None
         if W == 0:
             return total_out
-        lens = win_len
-        # chunk windows by slot budget
-        csum = torch.cumsum(lens, 0)
+        order = torch.argsort(task_row, stable=True)
+        task_row = task_row[order]
+        task_id = task_id[order]
+        win_off = win_off[order]
+        win_len = win_len[order]
+        if x_mask is not None and x_mask.dim() == 2:
+            x_mask = x_mask[order]
+        csum = torch.cumsum(win_len, 0)
         start = 0
-        a2e = None
+        a2e = z1e = None
         while start < W:
             base = csum[start - 1] if start > 0 else csum.new_zeros(())
             end_idx = int(torch.searchsorted(
@@ -196,12 +207,21 @@ class CnnHipEngine:
             tr = task_row[start:end].contiguous()
             ti = task_id[start:end].contiguous()
             wo = win_off[start:end].contiguous()
-            wl = lens[start:end].contiguous()
+            wl = win_len[start:end].contiguous()
             slot = torch.cumsum(wl, 0) - wl
             slots = int(wl.sum())
             max_len = int(wl.max()) if wl.numel() else 0
             if a2e is None or a2e.shape[0] < slots:
                 a2e = torch.empty(max(slots, 1), NF, device=dev)
+                z1e = torch.empty(max(slots, 1), NH, device=dev)
+            # per-slot metadata
+            srow = tr.repeat_interleave(wl)
+            stid = ti.repeat_interleave(wl)
+            within = (torch.arange(slots, device=dev)
+                      - slot.repeat_interleave(wl))
+            sy = self._y_arena[wo.repeat_interleave(wl) + within]
+            # fc1 GEMM blocks: runs of equal row, tiled by 64 slots
+            blk_row, blk_s0, blk_len = self._fc1_blocks(tr, wl, slot)
             xm = None
             if x_mask is not None:
                 xm = (x_mask[start:end].contiguous()
@@ -211,14 +231,40 @@ class CnnHipEngine:
                 outp = torch.empty(slots, self.O, device=dev)
             out = self.mod.cnn_eval(
                 params.contiguous(), tr, ti, wo, wl, slot,
-                self._x_arena, self._y_arena, a2e, xm, n_tasks, self.O,
-                mode, want_mse, max_len, outp)
+                self._x_arena, self._y_arena, a2e, z1e,
+                blk_row, blk_s0, blk_len, srow, stid, sy.contiguous(),
+                xm, n_tasks, self.O, mode, want_mse, max_len, slots, outp)
             if mode == EV_DUMP:
-                dump_sink(start, end, wo, wl, slot, outp)
+                dump_sink(stid, sy, outp)
             else:
                 total_out += out
             start = end
         return total_out
+
+    @staticmethod
+    def _fc1_blocks(tr: torch.Tensor, wl: torch.Tensor, slot: torch.Tensor):
+        """Tile the (row-sorted) slot space into <=64-slot blocks that
+        never cross a model-row boundary."""
+        dev = tr.device
+        trc = tr.cpu().numpy()
+        wlc = wl.cpu().numpy()
+        slc = slot.cpu().numpy()
+        rows, s0s, lens = [], [], []
+        i = 0
+        Wn = len(trc)
+        while i < Wn:
+            j = i
+            while j < Wn and trc[j] == trc[i]:
+                j += 1
+            run_s0 = int(slc[i])
+            run_end = int(slc[j - 1] + wlc[j - 1])
+            for s in range(run_s0, run_end, 64):
+                rows.append(int(trc[i]))
+                s0s.append(s)
+                lens.append(min(64, run_end - s))
+            i = j
+        t = lambda a: torch.as_tensor(a, dtype=torch.int64, device=dev)
+        return t(rows), t(s0s), t(lens)
 
     @torch.no_grad()
     def eval_tasks_stacked(self, params: torch.Tensor, task_row, task_id,
@@ -265,37 +311,38 @@ class CnnHipEngine:
             return out
         slots_total = int(win_len.sum())
         votes = torch.zeros(slots_total, self.O, device=dev)
-        slot_all = torch.cumsum(win_len, 0) - win_len
-        # per-slot task ids + sample y for the final argmax-vs-y reduction
-        ar = torch.arange(slots_total, device=dev)
-        w_idx = torch.searchsorted(torch.cumsum(win_len, 0), ar, right=True)
-        sample_idx = win_off[w_idx] + (ar - slot_all[w_idx])
-        y = y_arena[sample_idx]
-        tids = task_id[w_idx]
         wv = weights.to(dev)
+        # per-model sweeps: all rows equal per sweep, so the row-sort in
+        # _eval_sweep is the identity and slot order is consistent across
+        # models; votes accumulate per slot in that order
+        slot_meta = {}
 
         for m in active:
             rowv = torch.full_like(task_id, m)
             xm = masks[m] if masks is not None else None
+            written = [0]
 
-            def sink(start, end, wo, wl, slot, outp, _m=m):
-                s0 = int(slot_all[start])
-                probs = outp
+            def sink(stid, sy, probs, _m=m, _w=written):
+                s0 = _w[0]
+                ns = probs.shape[0]
                 if mode == "hard":
-                    onehot = torch.zeros_like(probs)
-                    onehot.scatter_(1, probs.argmax(1, keepdim=True), 1.0)
-                    contrib = onehot
+                    contrib = torch.zeros_like(probs)
+                    contrib.scatter_(1, probs.argmax(1, keepdim=True), 1.0)
                 else:
                     contrib = probs
                 if per_task:
-                    wslice = wv[tids[s0:s0 + probs.shape[0]], _m]
+                    wslice = wv[stid, _m]
                 else:
-                    wslice = wv[_m].expand(probs.shape[0])
-                votes[s0:s0 + probs.shape[0]] += \
-                    wslice.unsqueeze(1).float() * contrib
+                    wslice = wv[_m].expand(ns)
+                votes[s0:s0 + ns] += wslice.unsqueeze(1).float() * contrib
+                if _m == active[0]:
+                    slot_meta[s0] = (stid, sy)
+                _w[0] = s0 + ns
 
             self._eval_sweep(params, rowv, task_id, win_off, win_len,
                              n_tasks, EV_DUMP, x_mask=xm, dump_sink=sink)
+        tids = torch.cat([slot_meta[k][0] for k in sorted(slot_meta)])
+        y = torch.cat([slot_meta[k][1] for k in sorted(slot_meta)])
         pred = votes.argmax(1)
         corr = (pred == y).double()
         out[0].scatter_add_(0, tids, corr)

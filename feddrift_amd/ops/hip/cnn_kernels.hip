@@ -77,6 +77,7 @@ struct CnnArgs {
   float* __restrict__ da2;           // [G, B, NF]
   float* __restrict__ dz2c;          // [G, B, Z2N]
   float* __restrict__ dx1;           // [G, B, X1N]
+  float* __restrict__ c1part;        // [G, B, 320] conv1-wgrad partials
   // optimizer state (indexed by rows[g])
   float* __restrict__ m;
   float* __restrict__ v;
@@ -475,49 +476,60 @@ void cnn_conv2_dgrad(CnnArgs a) {
   }
 }
 
-// conv1 wgrad: block per g; threads own the 288 + 32 grad entries
+// conv1 wgrad, stage 1: per-(g, b) partials into the dz2 scratch region
+// (reused: dz2 is [G, B, 64] and conv1 has 288+32=320 grad entries, so
+// partials use their own ws buffer c1part [G, B, 320]). Deterministic
+// two-stage reduce (no atomics).
 extern "C" __global__ __launch_bounds__(WG)
-void cnn_conv1_wgrad(CnnArgs a) {
-  const int g = blockIdx.x;
-  const int n = step_n(a, g);
-  if (n == 0) return;
+void cnn_conv1_wgrad_part(CnnArgs a) {
+  const int g = blockIdx.x / a.B;
+  const int b = blockIdx.x - g * a.B;
+  if (b >= step_n(a, g)) return;
   const int tid = threadIdx.x;
-  float wacc[2] = {0.f, 0.f};
-  float bacc = 0.f;
-  for (int b = 0; b < n; ++b) {
-    const float* xs = a.x + (step_o(a, g) + b) * D_IN;
-    const float* xm = a.x_mask ? a.x_mask + (long long)g * D_IN : nullptr;
-    const float* dx = a.dx1 + ((long long)g * a.B + b) * X1N;
-    for (int t = 0; t < 2; ++t) {
-      const int tap = tid + t * WG;
-      if (tap < 288) {
-        const int c = tap / 9;
-        const int k = tap - c * 9;
-        const int ky = k / 3, kx = k - (k / 3) * 3;
-        const float* dc = dx + c * S1 * S1;
-        float s = 0.f;
-        for (int y = 0; y < S1; ++y)
-          for (int x = 0; x < S1; ++x) {
-            const int xi = (y + ky) * IN_W + x + kx;
-            float xv = xs[xi];
-            if (xm) xv *= xm[xi];
-            s = fmaf(xv, dc[y * S1 + x], s);
-          }
-        wacc[t] += s;
-      } else if (tap < 288 + C1) {
-        const int c = tap - 288;
-        const float* dc = dx + c * S1 * S1;
-        float s = 0.f;
-        for (int p = 0; p < S1 * S1; ++p) s += dc[p];
-        bacc += s;
+  __shared__ __attribute__((aligned(16))) float xin[D_IN];
+  const float* xs = a.x + (step_o(a, g) + b) * D_IN;
+  const float* xm = a.x_mask ? a.x_mask + (long long)g * D_IN : nullptr;
+  for (int d = tid; d < D_IN; d += WG)
+    xin[d] = xm ? xs[d] * xm[d] : xs[d];
+  __syncthreads();
+  const float* dx = a.dx1 + ((long long)g * a.B + b) * X1N;
+  float* out = a.c1part + ((long long)g * a.B + b) * 320;
+  for (int tap = tid; tap < 320; tap += WG) {
+    float s = 0.f;
+    if (tap < 288) {
+      const int c = tap / 9;
+      const int k = tap - c * 9;
+      const int ky = k / 3, kx = k - (k / 3) * 3;
+      const float* dc = dx + c * S1 * S1;
+      for (int y = 0; y < S1; ++y) {
+        const float* xr = xin + (y + ky) * IN_W + kx;
+        const float* dr = dc + y * S1;
+        for (int x = 0; x < S1; ++x) s = fmaf(xr[x], dr[x], s);
       }
+    } else {
+      const float* dc = dx + (tap - 288) * S1 * S1;
+      for (int p = 0; p < S1 * S1; ++p) s += dc[p];
     }
+    out[tap] = s;
   }
-  float* gr = a.grad + (long long)g * a.P;
-  for (int t = 0; t < 2; ++t) {
-    const int tap = tid + t * WG;
-    if (tap < 288) gr[OFF_W1C + tap] = wacc[t];
-    else if (tap < 288 + C1) gr[OFF_B1C + tap - 288] = bacc;
+}
+
+// conv1 wgrad, stage 2: reduce partials over the batch
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_conv1_wgrad_reduce(CnnArgs a) {
+  const long long total = (long long)a.G * 320;
+  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
+       q += (long long)gridDim.x * WG) {
+    const int g = (int)(q / 320);
+    const int tap = (int)(q - (long long)g * 320);
+    const int n = step_n(a, g);
+    if (n == 0) continue;
+    const float* part = a.c1part + (long long)g * a.B * 320 + tap;
+    float s = 0.f;
+    for (int b = 0; b < n; ++b) s += part[(long long)b * 320];
+    float* gr = a.grad + (long long)g * a.P;
+    if (tap < 288) gr[OFF_W1C + tap] = s;
+    else gr[OFF_B1C + tap - 288] = s;
   }
 }
 
@@ -583,6 +595,16 @@ struct CnnEvalArgs {
   const float* __restrict__ x_mask;     // [W, 784] or [784] or null
   int xm_per_task;
   float* __restrict__ a2e;              // [slots, NF]
+  float* __restrict__ z1e;              // [slots, NH] (fc1 output)
+  // fc1 GEMM block metadata (slots grouped by model row, <=64 per block)
+  const int64_t* __restrict__ blk_row;
+  const int64_t* __restrict__ blk_s0;
+  const int64_t* __restrict__ blk_len;
+  // per-slot metadata for the head kernel
+  const int64_t* __restrict__ srow;     // model row per slot
+  const int64_t* __restrict__ stid;     // task id per slot
+  const int64_t* __restrict__ sy;       // label per slot
+  long long n_slots;
   // mode EV_ACC
   double* __restrict__ correct;         // [T]
   double* __restrict__ total;
@@ -651,85 +673,124 @@ void cnn_eval_conv(CnnEvalArgs a) {
   }
 }
 
-// per-window block (128 threads): fc1 + relu + fc2 + softmax + tail
-extern "C" __global__ __launch_bounds__(NH)
-void cnn_eval_fc(CnnEvalArgs a) {
-  const int w = blockIdx.x;
-  const int n = (int)a.len[w];
-  if (n == 0) return;
+// fc1 eval as an MFMA tile GEMM: z1e[slot, h] = relu(a2e[slot, :] @
+// W1f[row(slot)]^T + b). Slots are grouped by model row (host sorts the
+// windows), each block owns up to 64 consecutive slots of one row:
+// 4 waves x 16 rows x 128 cols, f32-input MFMA (v_mfma_f32_16x16x4_f32 —
+// exact f32, the gfx950 f32 matrix path), BK=32 LDS staging with +1
+// padding against bank conflicts.
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define EVAL_BK 32
+
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_eval_fc1_mfma(CnnEvalArgs a) {
+  const int blk = blockIdx.x;
+  const long long row = a.blk_row[blk];
+  const long long s0 = a.blk_s0[blk];
+  const int mlen = (int)a.blk_len[blk];
   const int tid = threadIdx.x;
-  const float* wp = a.params + a.task_row[w] * (long long)a.P;
-  const long long tsk = a.task_id[w];
-  __shared__ __attribute__((aligned(16))) float sa2[NF];
-  __shared__ float sa1[NH];
-  __shared__ float sz2[64];
-  __shared__ float sred[2];
-  float c_acc = 0.f, l_acc = 0.f, e_acc = 0.f;
-  for (int i = 0; i < n; ++i) {
-    const float* src = a.a2e + (a.slot[w] + i) * (long long)NF;
-    for (int j = tid; j < NF; j += NH) sa2[j] = src[j];
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15;        // fragment row/col index
+  const int lk = l >> 4;        // fragment k index (0..3)
+
+  __shared__ __attribute__((aligned(16))) float sA[64][EVAL_BK + 1];
+  __shared__ __attribute__((aligned(16))) float sB[EVAL_BK][NH + 1];
+
+  f32x4 acc[8];
+#pragma unroll
+  for (int t = 0; t < 8; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+
+  const float* wp = a.params + row * (long long)a.P + OFF_W1F;
+  const int r8 = tid >> 5;       // 8 stager rows per pass
+  const int kk = tid & 31;
+  for (int k0 = 0; k0 < NF; k0 += EVAL_BK) {
+    for (int rr = r8; rr < 64; rr += 8)
+      sA[rr][kk] = (rr < mlen)
+          ? a.a2e[(s0 + rr) * (long long)NF + k0 + kk] : 0.f;
+    for (int hh = r8; hh < NH; hh += 8)
+      sB[kk][hh] = wp[(long long)hh * NF + k0 + kk];
     __syncthreads();
-    // fc1: thread h computes z1[h]
-    {
-      const float* wr = wp + OFF_W1F + (long long)tid * NF;
-      float z = wp[OFF_B1F + tid];
-      for (int j = 0; j < NF; ++j) z = fmaf(sa2[j], wr[j], z);
-      sa1[tid] = z > 0.f ? z : 0.f;
-    }
-    __syncthreads();
-    // fc2: thread o < O
-    if (tid < a.O) {
-      const float* wo = wp + OFF_W2F + (long long)tid * NH;
-      float z = wp[OFF_W2F + (long long)a.O * NH + tid];
-      for (int h = 0; h < NH; ++h) z = fmaf(sa1[h], wo[h], z);
-      sz2[tid] = z;
-    }
-    __syncthreads();
-    if (tid == 0) {
-      const int yi = (int)a.y[a.off[w] + i];
-      float zmax = -1e30f;
-      for (int o = 0; o < a.O; ++o) zmax = fmaxf(zmax, sz2[o]);
-      float zsum = 0.f;
-      float s[64];
-      for (int o = 0; o < a.O; ++o) {
-        s[o] = __expf(sz2[o] - zmax);
-        zsum += s[o];
-      }
-      int best = 0;
-      float bv = -1e30f;
-      for (int o = 0; o < a.O; ++o) {
-        s[o] /= zsum;  // the model OUTPUT (in-graph softmax)
-        if (s[o] > bv) { bv = s[o]; best = o; }
-      }
-      if (a.mode == EV_DUMP) {
-        for (int o = 0; o < a.O; ++o)
-          a.outp[(a.slot[w] + i) * (long long)a.O + o] = s[o];
-      } else if (a.mode == EV_CONF) {
-        atomicAdd(&a.conf[(tsk * a.O + yi) * a.O + best], 1.0);
-      } else {
-        // CE / mse on the softmax OUTPUT (double-softmax quirk)
-        float smax = -1e30f;
-        for (int o = 0; o < a.O; ++o) smax = fmaxf(smax, s[o]);
-        float ssum = 0.f;
-        for (int o = 0; o < a.O; ++o) ssum += __expf(s[o] - smax);
-        const float lse = logf(ssum) + smax;
-        c_acc += (best == yi) ? 1.f : 0.f;
-        l_acc += lse - s[yi];
-        if (a.mse) {
-          const float pt = __expf(s[yi] - lse);
-          e_acc += (1.f - pt) * (1.f - pt);
-        }
+#pragma unroll
+    for (int ks = 0; ks < EVAL_BK / 4; ++ks) {
+      const float av = sA[wv * 16 + li][ks * 4 + lk];
+#pragma unroll
+      for (int ct = 0; ct < 8; ++ct) {
+        const float bv = sB[ks * 4 + lk][ct * 16 + li];
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc[ct],
+                                                       0, 0, 0);
       }
     }
     __syncthreads();
   }
-  if (a.mode == EV_ACC && tid == 0) {
-    atomicAdd(&a.correct[tsk], (double)c_acc);
-    atomicAdd(&a.total[tsk], (double)n);
-    atomicAdd(&a.loss[tsk], (double)l_acc);
-    if (a.mse) atomicAdd(&a.mse[tsk], (double)e_acc);
+  // epilogue: bias + relu -> z1e[slot, h]
+  const float* bias = a.params + row * (long long)a.P + OFF_B1F;
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int mrow = wv * 16 + lk * 4 + r;
+      if (mrow < mlen) {
+        const int h = ct * 16 + li;
+        const float z = acc[ct][r] + bias[h];
+        a.z1e[(s0 + mrow) * (long long)NH + h] = z > 0.f ? z : 0.f;
+      }
+    }
   }
-  (void)sred;
+}
+
+// head: per-slot fc2 + in-graph softmax + mode tail (thread per slot)
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_eval_head(CnnEvalArgs a) {
+  for (long long slot = (long long)blockIdx.x * WG + threadIdx.x;
+       slot < a.n_slots; slot += (long long)gridDim.x * WG) {
+    const long long row = a.srow[slot];
+    const float* wp = a.params + row * (long long)a.P;
+    const float* z1 = a.z1e + slot * NH;
+    float s[64];
+    float zmax = -1e30f;
+    for (int o = 0; o < a.O; ++o) {
+      const float* wo = wp + OFF_W2F + (long long)o * NH;
+      float z = wp[OFF_W2F + (long long)a.O * NH + o];
+      for (int h = 0; h < NH; ++h) z = fmaf(z1[h], wo[h], z);
+      s[o] = z;
+      zmax = fmaxf(zmax, z);
+    }
+    float zsum = 0.f;
+    for (int o = 0; o < a.O; ++o) {
+      s[o] = __expf(s[o] - zmax);
+      zsum += s[o];
+    }
+    int best = 0;
+    float bv = -1e30f;
+    for (int o = 0; o < a.O; ++o) {
+      s[o] /= zsum;  // the model OUTPUT (in-graph softmax)
+      if (s[o] > bv) { bv = s[o]; best = o; }
+    }
+    const long long tsk = a.stid[slot];
+    const int yi = (int)a.sy[slot];
+    if (a.mode == EV_DUMP) {
+      for (int o = 0; o < a.O; ++o)
+        a.outp[slot * (long long)a.O + o] = s[o];
+    } else if (a.mode == EV_CONF) {
+      atomicAdd(&a.conf[(tsk * a.O + yi) * a.O + best], 1.0);
+    } else {
+      // CE / mse on the softmax OUTPUT (double-softmax quirk)
+      float smax = -1e30f;
+      for (int o = 0; o < a.O; ++o) smax = fmaxf(smax, s[o]);
+      float ssum = 0.f;
+      for (int o = 0; o < a.O; ++o) ssum += __expf(s[o] - smax);
+      const float lse = logf(ssum) + smax;
+      atomicAdd(&a.correct[tsk], (double)((best == yi) ? 1.f : 0.f));
+      atomicAdd(&a.total[tsk], 1.0);
+      atomicAdd(&a.loss[tsk], (double)(lse - s[yi]));
+      if (a.mse) {
+        const float pt = __expf(s[yi] - lse);
+        atomicAdd(&a.mse[tsk], (double)((1.f - pt) * (1.f - pt)));
+      }
+    }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -750,7 +811,7 @@ void cnn_train_epoch_impl(
     torch::Tensor ws_x1, torch::Tensor ws_a2, torch::Tensor ws_pidx,
     torch::Tensor ws_z1, torch::Tensor ws_a1, torch::Tensor ws_dz2,
     torch::Tensor ws_dz1, torch::Tensor ws_da2, torch::Tensor ws_dz2c,
-    torch::Tensor ws_dx1,
+    torch::Tensor ws_dx1, torch::Tensor ws_c1part,
     c10::optional<torch::Tensor> m, c10::optional<torch::Tensor> v,
     c10::optional<torch::Tensor> vmax, c10::optional<torch::Tensor> t,
     torch::Tensor lr, double wd, double p1, double p2,
@@ -776,6 +837,7 @@ void cnn_train_epoch_impl(
   a.da2 = ws_da2.data_ptr<float>();
   a.dz2c = ws_dz2c.data_ptr<float>();
   a.dx1 = ws_dx1.data_ptr<float>();
+  a.c1part = ws_c1part.data_ptr<float>();
   const bool adam = m.has_value();
   a.m = adam ? m->data_ptr<float>() : nullptr;
   a.v = adam ? v->data_ptr<float>() : nullptr;
@@ -810,7 +872,9 @@ void cnn_train_epoch_impl(
   L(cnn_pool_bwd, GB * NF);
   hipLaunchKernelGGL(cnn_conv2_wgrad, dim3(G * C2), dim3(WG), 0, s, a);
   L(cnn_conv2_dgrad, GB * X1N);
-  hipLaunchKernelGGL(cnn_conv1_wgrad, dim3(G), dim3(WG), 0, s, a);
+  hipLaunchKernelGGL(cnn_conv1_wgrad_part, dim3(G * (int)B), dim3(WG), 0,
+                     s, a);
+  L(cnn_conv1_wgrad_reduce, (long long)G * 320);
   L(cnn_opt_step, (long long)G * a.P);
   hipLaunchKernelGGL(cnn_opt_tick, dim3((G + WG - 1) / WG), dim3(WG), 0, s,
                      a);
@@ -821,9 +885,11 @@ void cnn_train_epoch_impl(
 torch::Tensor cnn_eval(
     torch::Tensor params, torch::Tensor task_row, torch::Tensor task_id,
     torch::Tensor off, torch::Tensor len, torch::Tensor slot,
-    torch::Tensor x, torch::Tensor y, torch::Tensor a2e,
+    torch::Tensor x, torch::Tensor y, torch::Tensor a2e, torch::Tensor z1e,
+    torch::Tensor blk_row, torch::Tensor blk_s0, torch::Tensor blk_len,
+    torch::Tensor srow, torch::Tensor stid, torch::Tensor sy,
     c10::optional<torch::Tensor> x_mask, int64_t n_tasks, int64_t O,
-    int64_t mode, bool want_mse, int64_t max_len,
+    int64_t mode, bool want_mse, int64_t max_len, int64_t n_slots,
     c10::optional<torch::Tensor> outp) {
   const int W = task_row.size(0);
   auto optd = torch::TensorOptions().dtype(torch::kFloat64)
@@ -835,7 +901,7 @@ torch::Tensor cnn_eval(
     out = torch::zeros({n_tasks, O, O}, optd);
   else
     out = torch::zeros({0}, optd);
-  if (W == 0) return out;
+  if (W == 0 || n_slots == 0) return out;
 
   CnnEvalArgs a;
   a.params = params.data_ptr<float>();
@@ -849,6 +915,14 @@ torch::Tensor cnn_eval(
   a.x_mask = x_mask.has_value() ? x_mask->data_ptr<float>() : nullptr;
   a.xm_per_task = x_mask.has_value() && x_mask->dim() == 2 ? 1 : 0;
   a.a2e = a2e.data_ptr<float>();
+  a.z1e = z1e.data_ptr<float>();
+  a.blk_row = blk_row.data_ptr<int64_t>();
+  a.blk_s0 = blk_s0.data_ptr<int64_t>();
+  a.blk_len = blk_len.data_ptr<int64_t>();
+  a.srow = srow.data_ptr<int64_t>();
+  a.stid = stid.data_ptr<int64_t>();
+  a.sy = sy.data_ptr<int64_t>();
+  a.n_slots = (long long)n_slots;
   double* base = (mode == EV_ACC) ? out.data_ptr<double>() : nullptr;
   a.correct = base;
   a.total = base ? base + n_tasks : nullptr;
@@ -863,7 +937,10 @@ torch::Tensor cnn_eval(
   auto s = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(cnn_eval_conv, dim3(W, (int)max_len), dim3(WG),
                      X1N * sizeof(float), s, a);
-  hipLaunchKernelGGL(cnn_eval_fc, dim3(W), dim3(NH), 0, s, a);
+  hipLaunchKernelGGL(cnn_eval_fc1_mfma, dim3((int)blk_row.size(0)),
+                     dim3(WG), 0, s, a);
+  hipLaunchKernelGGL(cnn_eval_head, dim3(grid_for(n_slots)), dim3(WG), 0,
+                     s, a);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "cnn_eval launch");
   return out;
 }
