@@ -93,9 +93,13 @@ class CnnHipEngine:
             "dz2": f(G * B * 64),
             "dz1": f(G * B * NH),
             "da2": f(G * B * NF),
-            "dz2c": f(G * B * Z2N),
+            "zz2": f(G * B * Z2N),
             "dx1": f(G * B * X1N),
             "c1part": f(G * B * 320),
+            "wtf": f(G * 9 * 2048),
+            "wtd": f(G * 9 * 2048),
+            "z1part": f(G * 4 * B * NH),
+            "w2part": f(G * 9 * 8 * 2048),
             "grad": f(G, self.P),
         }
         self._ws = ws
@@ -155,8 +159,9 @@ class CnnHipEngine:
                 work, ws["grad"], rows, x_arena, y_arena,
                 step_off, step_len, e, xm,
                 ws["x1"], ws["a2"], ws["pidx"], ws["z1"], ws["a1"],
-                ws["dz2"], ws["dz1"], ws["da2"], ws["dz2c"], ws["dx1"],
-                ws["c1part"],
+                ws["dz2"], ws["dz1"], ws["da2"], ws["zz2"], ws["dx1"],
+                ws["c1part"], ws["wtf"], ws["wtd"], ws["z1part"],
+                ws["w2part"],
                 opt["m"] if adam else None,
                 opt["v"] if adam else None,
                 opt["vmax"] if adam else None,

@@ -75,9 +75,14 @@ struct CnnArgs {
   float* __restrict__ dz2;           // [G, B, O]   dL/d(fc2 out)
   float* __restrict__ dz1;           // [G, B, NH]
   float* __restrict__ da2;           // [G, B, NF]
-  float* __restrict__ dz2c;          // [G, B, Z2N]
-  float* __restrict__ dx1;           // [G, B, X1N]
+  float* __restrict__ zz2;           // [G, B, Z2N] z2 (fwd) / dz2 (bwd),
+                                     //   channels-last
+  float* __restrict__ dx1;           // [G, B, X1N] channels-last
   float* __restrict__ c1part;        // [G, B, 320] conv1-wgrad partials
+  float* __restrict__ wtf;           // [G, 9, 32, 64] conv2 W (fwd)
+  float* __restrict__ wtd;           // [G, 9, 64, 32] conv2 W (dgrad)
+  float* __restrict__ z1part;        // [G, FC1_KS, B, NH]
+  float* __restrict__ w2part;        // [G, 9, W2_KS, 32, 64]
   // optimizer state (indexed by rows[g])
   float* __restrict__ m;
   float* __restrict__ v;
@@ -93,6 +98,9 @@ struct CnnArgs {
 
 #define OPT_SGD 0
 #define OPT_ADAM 1
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+#define EVAL_BK 32
 
 __device__ __forceinline__ float hash_u01(unsigned long long s,
                                           unsigned long long id) {
@@ -128,7 +136,9 @@ __device__ __forceinline__ long long step_o(const CnnArgs& a, int g) {
 // forward
 // ---------------------------------------------------------------------------
 
-// conv1: one thread per (g, b, c1, y, x) output element, grid-stride
+// conv1: one thread per (g, b, y, x, c1) output element, grid-stride.
+// x1 is CHANNELS-LAST [G, B, 26*26, 32] so the conv2 MFMA stagers read
+// contiguous 32-float ci slices per (ky, kx) tap.
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_conv1_fwd(CnnArgs a) {
   const long long total = (long long)a.G * a.B * X1N;
@@ -139,8 +149,8 @@ void cnn_conv1_fwd(CnnArgs a) {
     const int b = (int)(r / X1N);
     if (b >= step_n(a, g)) continue;
     const int e = (int)(r - (long long)b * X1N);
-    const int c = e / (S1 * S1);
-    const int p = e - c * S1 * S1;
+    const int p = e / C1;                  // pixel (channels-last)
+    const int c = e - p * C1;
     const int oy = p / S1, ox = p - (p / S1) * S1;
     const float* w = a.work + (long long)g * a.P;
     const float* xs = a.x + (step_o(a, g) + b) * D_IN;
@@ -159,9 +169,85 @@ void cnn_conv1_fwd(CnnArgs a) {
   }
 }
 
-// conv2 + bias + maxpool + dropout1: thread per (g, b, c2, py, px)
+// reshape conv2 weights for the MFMA stagers (per pair, per epoch):
+// wtf[g][kyx][ci][co] (fwd B-operand) and wtd[g][kyx][co][ci] (dgrad)
 extern "C" __global__ __launch_bounds__(WG)
-void cnn_conv2_pool_fwd(CnnArgs a) {
+void cnn_w2_reshape(CnnArgs a) {
+  const long long total = (long long)a.G * C2 * C1 * 9;
+  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
+       q += (long long)gridDim.x * WG) {
+    const int g = (int)(q / (C2 * C1 * 9));
+    const int r = (int)(q - (long long)g * C2 * C1 * 9);
+    const int co = r / (C1 * 9);
+    const int t = r - co * C1 * 9;
+    const int ci = t / 9;
+    const int kyx = t - ci * 9;
+    const float wv = a.work[(long long)g * a.P + OFF_W2C + r];
+    a.wtf[((long long)g * 9 + kyx) * 2048 + ci * C2 + co] = wv;
+    a.wtd[((long long)g * 9 + kyx) * 2048 + co * C1 + ci] = wv;
+  }
+}
+
+// conv2 forward as implicit GEMM on f32 MFMA: per (g, b, ptile) block,
+// 64 output pixels x 64 channels, K = 9 taps x 32 ci. A tiles (x1
+// slices) and B tiles (reshaped weights) stage through LDS; z2 lands
+// CHANNELS-LAST in the zz2 buffer for the pool kernel.
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_conv2_fwd_mfma(CnnArgs a) {
+  const int pt = blockIdx.x % 9;           // 576 / 64
+  const int gb = blockIdx.x / 9;
+  const int g = gb / a.B;
+  const int b = gb - g * a.B;
+  if (b >= step_n(a, g)) return;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sA[64][C1 + 1];
+  __shared__ __attribute__((aligned(16))) float sB[C1][C2 + 1];
+  f32x4 acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+  const float* x1 = a.x1 + ((long long)g * a.B + b) * X1N;
+  const int r8 = tid >> 5, kk = tid & 31;
+  for (int kyx = 0; kyx < 9; ++kyx) {
+    const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+    for (int rr = r8; rr < 64; rr += 8) {
+      const int p = pt * 64 + rr;
+      const int oy = p / S2 + ky, ox = p - (p / S2) * S2 + kx;
+      sA[rr][kk] = x1[(oy * S1 + ox) * C1 + kk];
+    }
+    const float* wt = a.wtf + ((long long)g * 9 + kyx) * 2048;
+    for (int cc = r8; cc < C2; cc += 8) sB[kk][cc] = wt[kk * C2 + cc];
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < C1 / 4; ++ks) {
+      const float av = sA[wv * 16 + li][ks * 4 + lk];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct)
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            av, sB[ks * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  const float* bias = a.work + (long long)g * a.P + OFF_B2C;
+  float* z2 = a.zz2 + ((long long)g * a.B + b) * Z2N;
+#pragma unroll
+  for (int ct = 0; ct < 4; ++ct) {
+    const int co = ct * 16 + li;
+    const float bb = bias[co];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int p = pt * 64 + wv * 16 + lk * 4 + r;
+      z2[(long long)p * C2 + co] = acc[ct][r] + bb;
+    }
+  }
+}
+
+// maxpool + dropout1 over the channels-last z2: thread per (g, b, e)
+// where e = c*144 + py*12 + px is the torch-flatten a2 layout
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_pool_fwd(CnnArgs a) {
   const long long total = (long long)a.G * a.B * NF;
   for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
        q += (long long)gridDim.x * WG) {
@@ -173,25 +259,15 @@ void cnn_conv2_pool_fwd(CnnArgs a) {
     const int c = e / (SP * SP);
     const int p = e - c * SP * SP;
     const int py = p / SP, px = p - (p / SP) * SP;
-    const float* w = a.work + (long long)g * a.P;
-    const float* x1 = a.x1 + ((long long)g * a.B + b) * X1N;
+    const float* z2 = a.zz2 + ((long long)g * a.B + b) * Z2N;
     float best = -1e30f;
     int arg = 0;
 #pragma unroll
     for (int dy = 0; dy < 2; ++dy)
 #pragma unroll
       for (int dx = 0; dx < 2; ++dx) {
-        const int oy = 2 * py + dy, ox = 2 * px + dx;
-        float z = w[OFF_B2C + c];
-        for (int ci = 0; ci < C1; ++ci) {
-          const float* xc = x1 + ci * S1 * S1 + oy * S1 + ox;
-          const float* wc = w + OFF_W2C + (c * C1 + ci) * 9;
-#pragma unroll
-          for (int ky = 0; ky < 3; ++ky)
-#pragma unroll
-            for (int kx = 0; kx < 3; ++kx)
-              z = fmaf(xc[ky * S1 + kx], wc[ky * 3 + kx], z);
-        }
+        const float z = z2[(long long)((2 * py + dy) * S2 + 2 * px + dx)
+                           * C2 + c];
         if (z > best) { best = z; arg = dy * 2 + dx; }
       }
     a.pidx[q] = (unsigned char)arg;
@@ -199,9 +275,67 @@ void cnn_conv2_pool_fwd(CnnArgs a) {
   }
 }
 
-// fc1 + relu + dropout2: thread per (g, b, h)
+// fc1 forward as MFMA GEMM with K split over blocks (fills the chip at
+// small fleets): block (g, mtile, ks) computes partial z1 for 64 batch
+// rows x 128 h over K-range [ks*NF/KS, ...). Partials land in z1part.
+#define FC1_KS 4
+
 extern "C" __global__ __launch_bounds__(WG)
-void cnn_fc1_fwd(CnnArgs a) {
+void cnn_fc1_fwd_mfma(CnnArgs a) {
+  const int mtiles = (a.B + 63) / 64;
+  const int ks = blockIdx.x % FC1_KS;
+  const int rest = blockIdx.x / FC1_KS;
+  const int mt = rest % mtiles;
+  const int g = rest / mtiles;
+  const int n = step_n(a, g);
+  if (mt * 64 >= n) return;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sA[64][EVAL_BK + 1];
+  __shared__ __attribute__((aligned(16))) float sB[EVAL_BK][NH + 1];
+  f32x4 acc[8];
+#pragma unroll
+  for (int t = 0; t < 8; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+  const float* wp = a.work + (long long)g * a.P + OFF_W1F;
+  const float* a2 = a.a2 + ((long long)g * a.B + (long long)mt * 64) * NF;
+  const int mlen = min(64, n - mt * 64);
+  const int r8 = tid >> 5, kk = tid & 31;
+  const int k_lo = ks * (NF / FC1_KS), k_hi = (ks + 1) * (NF / FC1_KS);
+  for (int k0 = k_lo; k0 < k_hi; k0 += EVAL_BK) {
+    for (int rr = r8; rr < 64; rr += 8)
+      sA[rr][kk] = (rr < mlen) ? a2[(long long)rr * NF + k0 + kk] : 0.f;
+    for (int hh = r8; hh < NH; hh += 8)
+      sB[kk][hh] = wp[(long long)hh * NF + k0 + kk];
+    __syncthreads();
+#pragma unroll
+    for (int kq = 0; kq < EVAL_BK / 4; ++kq) {
+      const float av = sA[wv * 16 + li][kq * 4 + lk];
+#pragma unroll
+      for (int ct = 0; ct < 8; ++ct)
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            av, sB[kq * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int mrow = wv * 16 + lk * 4 + r;
+      if (mrow < mlen) {
+        const long long b = (long long)mt * 64 + mrow;
+        a.z1part[(((long long)g * FC1_KS + ks) * a.B + b) * NH
+                 + ct * 16 + li] = acc[ct][r];
+      }
+    }
+  }
+}
+
+// combine fc1 K-split partials + bias, apply relu + dropout2
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_fc1_act(CnnArgs a) {
   const long long total = (long long)a.G * a.B * NH;
   for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
        q += (long long)gridDim.x * WG) {
@@ -210,11 +344,10 @@ void cnn_fc1_fwd(CnnArgs a) {
     const int b = (int)(r / NH);
     if (b >= step_n(a, g)) continue;
     const int h = (int)(r - (long long)b * NH);
-    const float* w = a.work + (long long)g * a.P + OFF_W1F +
-                     (long long)h * NF;
-    const float* xin = a.a2 + ((long long)g * a.B + b) * NF;
     float z = a.work[(long long)g * a.P + OFF_B1F + h];
-    for (int j = 0; j < NF; ++j) z = fmaf(xin[j], w[j], z);
+#pragma unroll
+    for (int ks = 0; ks < FC1_KS; ++ks)
+      z += a.z1part[(((long long)g * FC1_KS + ks) * a.B + b) * NH + h];
     a.z1[q] = z;
     const float rl = z > 0.f ? z : 0.f;
     a.a1[q] = rl * drop_scale(a, 1, g, b, h);
@@ -315,52 +448,99 @@ void cnn_fc2_dgrad(CnnArgs a) {
   }
 }
 
-// fc1 wgrad: thread per (g, h, j), exclusive writes; bias at j == 0
+// fc1 wgrad: block per (g, jtile of 128 columns). dz1 reads come from
+// global (L2-hot, ~50 KB per pair); a2 column tiles stage through LDS in
+// batch chunks; exclusive-owner writes, no atomics.
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_fc1_wgrad(CnnArgs a) {
-  const long long total = (long long)a.G * NH * NF;
+  const int jt = blockIdx.x % (NF / 128);
+  const int g = blockIdx.x / (NF / 128);
+  const int n = step_n(a, g);
+  if (n == 0) return;
+  const int tid = threadIdx.x;
+  __shared__ __attribute__((aligned(16))) float sa2[96][129];
+  // thread owns (h0.., j): j fixed per thread, 64 h's
+  const int j = tid & 127;
+  const int h0 = (tid >> 7) * 64;       // 0 or 64
+  float acc[64];
+#pragma unroll
+  for (int t = 0; t < 64; ++t) acc[t] = 0.f;
+  for (int b0 = 0; b0 < n; b0 += 96) {
+    const int bc = min(96, n - b0);
+    for (int q = tid; q < bc * 128; q += WG) {
+      const int bb = q >> 7;
+      sa2[bb][q & 127] =
+          a.a2[((long long)g * a.B + b0 + bb) * NF + jt * 128 + (q & 127)];
+    }
+    __syncthreads();
+    const float* d1 = a.dz1 + ((long long)g * a.B + b0) * NH;
+    for (int bb = 0; bb < bc; ++bb) {
+      const float av = sa2[bb][j];
+      const float* dr = d1 + (long long)bb * NH + h0;
+#pragma unroll
+      for (int t = 0; t < 64; ++t) acc[t] = fmaf(dr[t], av, acc[t]);
+    }
+    __syncthreads();
+  }
+  float* gr = a.grad + (long long)g * a.P + OFF_W1F;
+#pragma unroll
+  for (int t = 0; t < 64; ++t)
+    gr[(long long)(h0 + t) * NF + jt * 128 + j] = acc[t];
+}
+
+// fc1 bias grad: thread per (g, h)
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_fc1_bias_grad(CnnArgs a) {
+  const long long total = (long long)a.G * NH;
   for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
        q += (long long)gridDim.x * WG) {
-    const int g = (int)(q / ((long long)NH * NF));
-    const long long r = q - (long long)g * NH * NF;
-    const int h = (int)(r / NF);
-    const int j = (int)(r - (long long)h * NF);
+    const int g = (int)(q / NH);
+    const int h = (int)(q - (long long)g * NH);
     const int n = step_n(a, g);
     if (n == 0) continue;
-    float acc = 0.f, accb = 0.f;
-    for (int b = 0; b < n; ++b) {
-      const float d = a.dz1[((long long)g * a.B + b) * NH + h];
-      acc = fmaf(d, a.a2[((long long)g * a.B + b) * NF + j], acc);
-      if (j == 0) accb += d;
-    }
-    float* gr = a.grad + (long long)g * a.P;
-    gr[OFF_W1F + (long long)h * NF + j] = acc;
-    if (j == 0) gr[OFF_B1F + h] = accb;
+    float s = 0.f;
+    for (int b = 0; b < n; ++b)
+      s += a.dz1[((long long)g * a.B + b) * NH + h];
+    a.grad[(long long)g * a.P + OFF_B1F + h] = s;
   }
 }
 
-// fc1 dgrad: da2[b, j] = sum_h dz1[b, h] * W1f[h, j]; thread per (g, b, j)
+// fc1 dgrad: block per (g, jtile of 256); dz1 tile staged in LDS, W1f
+// columns streamed once per 16-sample tile (B/16 re-reads, L2-resident)
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_fc1_dgrad(CnnArgs a) {
-  const long long total = (long long)a.G * a.B * NF;
-  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
-       q += (long long)gridDim.x * WG) {
-    const int g = (int)(q / ((long long)a.B * NF));
-    const long long r = q - (long long)g * a.B * NF;
-    const int b = (int)(r / NF);
-    if (b >= step_n(a, g)) continue;
-    const int j = (int)(r - (long long)b * NF);
-    const float* w = a.work + (long long)g * a.P + OFF_W1F;
-    const float* d1 = a.dz1 + ((long long)g * a.B + b) * NH;
-    float acc = 0.f;
-    for (int h = 0; h < NH; ++h)
-      acc = fmaf(d1[h], w[(long long)h * NF + j], acc);
-    a.da2[q] = acc;
+  const int jt = blockIdx.x % (NF / 256);
+  const int g = blockIdx.x / (NF / 256);
+  const int n = step_n(a, g);
+  if (n == 0) return;
+  const int tid = threadIdx.x;
+  const int j = jt * 256 + tid;
+  __shared__ __attribute__((aligned(16))) float sd1[16][NH];
+  const float* w = a.work + (long long)g * a.P + OFF_W1F;
+  for (int b0 = 0; b0 < n; b0 += 16) {
+    const int bc = min(16, n - b0);
+    for (int q = tid; q < bc * NH; q += WG)
+      sd1[q >> 7][q & 127] =
+          a.dz1[((long long)g * a.B + b0 + (q >> 7)) * NH + (q & 127)];
+    __syncthreads();
+    float acc[16];
+#pragma unroll
+    for (int t = 0; t < 16; ++t) acc[t] = 0.f;
+    for (int h = 0; h < NH; ++h) {
+      const float wv = w[(long long)h * NF + j];
+#pragma unroll
+      for (int t = 0; t < 16; ++t)
+        acc[t] = fmaf(sd1[t][h], wv, acc[t]);
+    }
+    for (int t = 0; t < bc; ++t)
+      a.da2[((long long)g * a.B + b0 + t) * NF + j] = acc[t];
+    __syncthreads();
   }
 }
 
 // pool backward: route da2 (through dropout1) to the argmax position;
-// the other 3 positions of the 2x2 cell are written zero (exclusive owner)
+// dz2 lands CHANNELS-LAST in zz2 (reused as dz buffer) for the conv2
+// MFMA backward stagers; the other 3 cell positions are written zero
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_pool_bwd(CnnArgs a) {
   const long long total = (long long)a.G * a.B * NF;
@@ -376,103 +556,187 @@ void cnn_pool_bwd(CnnArgs a) {
     const int py = p / SP, px = p - (p / SP) * SP;
     const float dv = a.da2[q] * drop_scale(a, 0, g, b, e);
     const int arg = a.pidx[q];
-    float* dzc = a.dz2c + ((long long)g * a.B + b) * Z2N + c * S2 * S2;
+    float* dzc = a.zz2 + ((long long)g * a.B + b) * Z2N;
 #pragma unroll
     for (int dy = 0; dy < 2; ++dy)
 #pragma unroll
       for (int dx = 0; dx < 2; ++dx)
-        dzc[(2 * py + dy) * S2 + 2 * px + dx] =
+        dzc[(long long)((2 * py + dy) * S2 + 2 * px + dx) * C2 + c] =
             (dy * 2 + dx == arg) ? dv : 0.f;
   }
 }
 
-// conv2 wgrad: block per (g, c2); threads own the 288 (ci, ky, kx) taps
-// and loop (b, pixels); bias via cooperative reduce
+// conv2 wgrad as MFMA GEMM: dW[(kyx, ci)][co] = sum_m A[m, ci] dz[m, co]
+// with m = (b, pixel). Blocks (g, kyx, msplit): 4 waves share staged
+// A/dz tiles, each owning 2 of the 8 (ci-tile, co-tile) positions;
+// per-msplit partials land in w2part and reduce deterministically.
+#define W2_KS 8
+
 extern "C" __global__ __launch_bounds__(WG)
-void cnn_conv2_wgrad(CnnArgs a) {
+void cnn_conv2_wgrad_mfma(CnnArgs a) {
+  const int ms = blockIdx.x % W2_KS;
+  const int rest = blockIdx.x / W2_KS;
+  const int kyx = rest % 9;
+  const int g = rest / 9;
+  const int n = step_n(a, g);
+  float* part = a.w2part +
+      (((long long)g * 9 + kyx) * W2_KS + ms) * 2048;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+  __shared__ __attribute__((aligned(16))) float sA[64][C1 + 1];
+  __shared__ __attribute__((aligned(16))) float sD[64][C2 + 1];
+  f32x4 acc[2];
+  acc[0] = f32x4{0.f, 0.f, 0.f, 0.f};
+  acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+  const int r8 = tid >> 5, kk = tid & 31;
+  const long long mtot = (long long)n * (S2 * S2);
+  const int mtiles = (int)((mtot + 63) / 64);
+  for (int mt = ms; mt < mtiles; mt += W2_KS) {
+    // stage 64 m-rows: x1 tap slice + dz slice (channels-last)
+    for (int rr = r8; rr < 64; rr += 8) {
+      const long long m = (long long)mt * 64 + rr;
+      const int b = (int)(m / (S2 * S2));
+      const int p = (int)(m - (long long)b * (S2 * S2));
+      const bool ok = m < mtot;
+      const int oy = p / S2 + ky, ox = p - (p / S2) * S2 + kx;
+      const long long gb = (long long)g * a.B + b;
+      sA[rr][kk] = ok ? a.x1[(gb * 676 + oy * S1 + ox) * C1 + kk] : 0.f;
+      sD[rr][kk] = ok ? a.zz2[(gb * 576 + p) * C2 + kk] : 0.f;
+      sD[rr][kk + 32] =
+          ok ? a.zz2[(gb * 576 + p) * C2 + kk + 32] : 0.f;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int km = 0; km < 16; ++km) {
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const int tile = wv * 2 + t;
+        const int rt = tile >> 2, ct = tile & 3;
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            sA[km * 4 + lk][rt * 16 + li],
+            sD[km * 4 + lk][ct * 16 + li], acc[t], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    const int tile = wv * 2 + t;
+    const int rt = tile >> 2, ct = tile & 3;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int ci = rt * 16 + lk * 4 + r;
+      part[ci * C2 + ct * 16 + li] = acc[t][r];
+    }
+  }
+}
+
+// reduce the conv2 wgrad msplit partials into the (co, ci, ky, kx) grad
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_conv2_wgrad_reduce(CnnArgs a) {
+  const long long total = (long long)a.G * 9 * 2048;
+  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
+       q += (long long)gridDim.x * WG) {
+    const int g = (int)(q / (9 * 2048));
+    const int r = (int)(q - (long long)g * 9 * 2048);
+    const int kyx = r / 2048;
+    const int t = r - kyx * 2048;
+    const int ci = t / C2, co = t - (t / C2) * C2;
+    if (step_n(a, g) == 0) continue;
+    const float* part = a.w2part + ((long long)g * 9 + kyx) * W2_KS * 2048
+                        + t;
+    float s = 0.f;
+#pragma unroll
+    for (int ms = 0; ms < W2_KS; ++ms) s += part[(long long)ms * 2048];
+    a.grad[(long long)g * a.P + OFF_W2C + (co * C1 + ci) * 9 + kyx] = s;
+  }
+}
+
+// conv2 bias grad: block per (g, co), strided sum over (b, pixel)
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_conv2_bias_grad(CnnArgs a) {
   const int g = blockIdx.x / C2;
-  const int c = blockIdx.x - g * C2;
+  const int co = blockIdx.x - g * C2;
   const int n = step_n(a, g);
   if (n == 0) return;
   const int tid = threadIdx.x;
-  __shared__ float sdz[S2 * S2];
   __shared__ float sred[4];
-  float acc[2] = {0.f, 0.f};  // up to 2 taps per thread (288 <= 2*256)
-  float bacc = 0.f;
-  for (int b = 0; b < n; ++b) {
-    const float* dzc =
-        a.dz2c + ((long long)g * a.B + b) * Z2N + c * S2 * S2;
-    for (int p = tid; p < S2 * S2; p += WG) sdz[p] = dzc[p];
-    __syncthreads();
-    const float* x1 = a.x1 + ((long long)g * a.B + b) * X1N;
-    for (int t = 0; t < 2; ++t) {
-      const int tap = tid + t * WG;
-      if (tap < 288) {
-        const int ci = tap / 9;
-        const int k = tap - ci * 9;
-        const int ky = k / 3, kx = k - (k / 3) * 3;
-        const float* xc = x1 + ci * S1 * S1 + ky * S1 + kx;
-        float s = 0.f;
-        for (int oy = 0; oy < S2; ++oy) {
-          const float* xr = xc + oy * S1;
-          const float* dr = sdz + oy * S2;
-          for (int ox = 0; ox < S2; ++ox) s = fmaf(xr[ox], dr[ox], s);
-        }
-        acc[t] += s;
-      }
-    }
-    // bias: cooperative sum of sdz
-    float bs = 0.f;
-    for (int p = tid; p < S2 * S2; p += WG) bs += sdz[p];
-    bacc += bs;
-    __syncthreads();
+  float s = 0.f;
+  const long long mtot = (long long)n * (S2 * S2);
+  const float* dz = a.zz2 + (long long)g * a.B * Z2N;
+  for (long long m = tid; m < mtot; m += WG) {
+    const int b = (int)(m / (S2 * S2));
+    const int p = (int)(m - (long long)b * (S2 * S2));
+    s += dz[((long long)b * 576 + p) * C2 + co];
   }
-  float* gr = a.grad + (long long)g * a.P;
-  for (int t = 0; t < 2; ++t) {
-    const int tap = tid + t * WG;
-    if (tap < 288) gr[OFF_W2C + c * 288 + tap] = acc[t];
-  }
-  // block-reduce bacc
-  for (int off = 32; off > 0; off >>= 1) bacc += __shfl_down(bacc, off, 64);
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
   const int lane = tid & 63, wave = tid >> 6;
-  if (lane == 0) sred[wave] = bacc;
+  if (lane == 0) sred[wave] = s;
   __syncthreads();
   if (tid == 0)
-    gr[OFF_B2C + c] = sred[0] + sred[1] + sred[2] + sred[3];
+    a.grad[(long long)g * a.P + OFF_B2C + co] =
+        sred[0] + sred[1] + sred[2] + sred[3];
 }
 
-// conv2 dgrad: dx1[b, ci, y, x] = sum over valid (c2, ky, kx); thread per
-// (g, b, ci, y, x)
+// conv2 dgrad as MFMA GEMM: dx1[m=(b,y,x)][ci] over K = (kyx, co) with
+// zero-padded dz staging at the borders. Blocks (g, b, mtile of 64).
 extern "C" __global__ __launch_bounds__(WG)
-void cnn_conv2_dgrad(CnnArgs a) {
-  const long long total = (long long)a.G * a.B * X1N;
-  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
-       q += (long long)gridDim.x * WG) {
-    const int g = (int)(q / ((long long)a.B * X1N));
-    const long long r = q - (long long)g * a.B * X1N;
-    const int b = (int)(r / X1N);
-    if (b >= step_n(a, g)) continue;
-    const int e = (int)(r - (long long)b * X1N);
-    const int ci = e / (S1 * S1);
-    const int p = e - ci * S1 * S1;
-    const int y = p / S1, x = p - (p / S1) * S1;
-    const float* w = a.work + (long long)g * a.P + OFF_W2C;
-    const float* dzc = a.dz2c + ((long long)g * a.B + b) * Z2N;
-    float acc = 0.f;
-#pragma unroll
-    for (int ky = 0; ky < 3; ++ky) {
-      const int oy = y - ky;
-      if (oy < 0 || oy >= S2) continue;
-#pragma unroll
-      for (int kx = 0; kx < 3; ++kx) {
-        const int ox = x - kx;
-        if (ox < 0 || ox >= S2) continue;
-        for (int co = 0; co < C2; ++co)
-          acc = fmaf(dzc[co * S2 * S2 + oy * S2 + ox],
-                     w[(co * C1 + ci) * 9 + ky * 3 + kx], acc);
-      }
+void cnn_conv2_dgrad_mfma(CnnArgs a) {
+  const int mtiles = 11;                  // ceil(676 / 64)
+  const int mt = blockIdx.x % mtiles;
+  const int gb = blockIdx.x / mtiles;
+  const int g = gb / a.B;
+  const int b = gb - g * a.B;
+  if (b >= step_n(a, g)) return;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sA[64][C2 + 1];
+  __shared__ __attribute__((aligned(16))) float sB[C2][C1 + 1];
+  f32x4 acc[2];
+  acc[0] = f32x4{0.f, 0.f, 0.f, 0.f};
+  acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+  const float* dz = a.zz2 + ((long long)g * a.B + b) * Z2N;
+  const int r8 = tid >> 5, kk = tid & 31;
+  for (int kyx = 0; kyx < 9; ++kyx) {
+    const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+    for (int rr = r8; rr < 64; rr += 8) {
+      const int pix = mt * 64 + rr;
+      const int y = pix / S1, x = pix - (pix / S1) * S1;
+      const int oy = y - ky, ox = x - kx;
+      const bool ok = pix < 676 && oy >= 0 && oy < S2 && ox >= 0
+                      && ox < S2;
+      const long long src = ((long long)oy * S2 + ox) * C2;
+      sA[rr][kk] = ok ? dz[src + kk] : 0.f;
+      sA[rr][kk + 32] = ok ? dz[src + kk + 32] : 0.f;
     }
-    a.dx1[q] = acc;
+    const float* wt = a.wtd + ((long long)g * 9 + kyx) * 2048;
+    for (int cc = r8; cc < C2; cc += 8)
+      sB[cc][kk] = wt[cc * C1 + kk];
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < C2 / 4; ++ks) {
+      const float av = sA[wv * 16 + li][ks * 4 + lk];
+#pragma unroll
+      for (int ct = 0; ct < 2; ++ct)
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            av, sB[ks * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  float* dx1 = a.dx1 + ((long long)g * a.B + b) * X1N;
+#pragma unroll
+  for (int ct = 0; ct < 2; ++ct) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int pix = mt * 64 + wv * 16 + lk * 4 + r;
+      if (pix < 676)
+        dx1[(long long)pix * C1 + ct * 16 + li] = acc[ct][r];
+    }
   }
 }
 
@@ -492,7 +756,7 @@ void cnn_conv1_wgrad_part(CnnArgs a) {
   for (int d = tid; d < D_IN; d += WG)
     xin[d] = xm ? xs[d] * xm[d] : xs[d];
   __syncthreads();
-  const float* dx = a.dx1 + ((long long)g * a.B + b) * X1N;
+  const float* dx = a.dx1 + ((long long)g * a.B + b) * X1N;  // chans-last
   float* out = a.c1part + ((long long)g * a.B + b) * 320;
   for (int tap = tid; tap < 320; tap += WG) {
     float s = 0.f;
@@ -500,15 +764,15 @@ void cnn_conv1_wgrad_part(CnnArgs a) {
       const int c = tap / 9;
       const int k = tap - c * 9;
       const int ky = k / 3, kx = k - (k / 3) * 3;
-      const float* dc = dx + c * S1 * S1;
       for (int y = 0; y < S1; ++y) {
         const float* xr = xin + (y + ky) * IN_W + kx;
-        const float* dr = dc + y * S1;
-        for (int x = 0; x < S1; ++x) s = fmaf(xr[x], dr[x], s);
+        const float* dr = dx + (long long)y * S1 * C1 + c;
+        for (int x = 0; x < S1; ++x)
+          s = fmaf(xr[x], dr[(long long)x * C1], s);
       }
     } else {
-      const float* dc = dx + (tap - 288) * S1 * S1;
-      for (int p = 0; p < S1 * S1; ++p) s += dc[p];
+      const int c = tap - 288;
+      for (int p = 0; p < S1 * S1; ++p) s += dx[(long long)p * C1 + c];
     }
     out[tap] = s;
   }
@@ -679,10 +943,6 @@ void cnn_eval_conv(CnnEvalArgs a) {
 // 4 waves x 16 rows x 128 cols, f32-input MFMA (v_mfma_f32_16x16x4_f32 —
 // exact f32, the gfx950 f32 matrix path), BK=32 LDS staging with +1
 // padding against bank conflicts.
-typedef float f32x4 __attribute__((ext_vector_type(4)));
-
-#define EVAL_BK 32
-
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_eval_fc1_mfma(CnnEvalArgs a) {
   const int blk = blockIdx.x;
@@ -810,8 +1070,9 @@ void cnn_train_epoch_impl(
     c10::optional<torch::Tensor> x_mask,
     torch::Tensor ws_x1, torch::Tensor ws_a2, torch::Tensor ws_pidx,
     torch::Tensor ws_z1, torch::Tensor ws_a1, torch::Tensor ws_dz2,
-    torch::Tensor ws_dz1, torch::Tensor ws_da2, torch::Tensor ws_dz2c,
-    torch::Tensor ws_dx1, torch::Tensor ws_c1part,
+    torch::Tensor ws_dz1, torch::Tensor ws_da2, torch::Tensor ws_zz2,
+    torch::Tensor ws_dx1, torch::Tensor ws_c1part, torch::Tensor ws_wtf,
+    torch::Tensor ws_wtd, torch::Tensor ws_z1part, torch::Tensor ws_w2part,
     c10::optional<torch::Tensor> m, c10::optional<torch::Tensor> v,
     c10::optional<torch::Tensor> vmax, c10::optional<torch::Tensor> t,
     torch::Tensor lr, double wd, double p1, double p2,
@@ -835,9 +1096,13 @@ void cnn_train_epoch_impl(
   a.dz2 = ws_dz2.data_ptr<float>();
   a.dz1 = ws_dz1.data_ptr<float>();
   a.da2 = ws_da2.data_ptr<float>();
-  a.dz2c = ws_dz2c.data_ptr<float>();
+  a.zz2 = ws_zz2.data_ptr<float>();
   a.dx1 = ws_dx1.data_ptr<float>();
   a.c1part = ws_c1part.data_ptr<float>();
+  a.wtf = ws_wtf.data_ptr<float>();
+  a.wtd = ws_wtd.data_ptr<float>();
+  a.z1part = ws_z1part.data_ptr<float>();
+  a.w2part = ws_w2part.data_ptr<float>();
   const bool adam = m.has_value();
   a.m = adam ? m->data_ptr<float>() : nullptr;
   a.v = adam ? v->data_ptr<float>() : nullptr;
@@ -861,20 +1126,36 @@ void cnn_train_epoch_impl(
   const long long GB = (long long)G * B;
 #define L(kern, total) \
   hipLaunchKernelGGL(kern, dim3(grid_for(total)), dim3(WG), 0, s, a)
+  const int mtiles = (a.B + 63) / 64;
+  // forward
+  L(cnn_w2_reshape, (long long)G * C2 * C1 * 9);
   L(cnn_conv1_fwd, GB * X1N);
-  L(cnn_conv2_pool_fwd, GB * NF);
-  L(cnn_fc1_fwd, GB * NH);
+  hipLaunchKernelGGL(cnn_conv2_fwd_mfma, dim3((int)GB * 9), dim3(WG), 0,
+                     s, a);
+  L(cnn_pool_fwd, GB * NF);
+  hipLaunchKernelGGL(cnn_fc1_fwd_mfma, dim3(G * mtiles * FC1_KS), dim3(WG),
+                     0, s, a);
+  L(cnn_fc1_act, GB * NH);
   L(cnn_head_fwd, GB);
+  // backward
   L(cnn_fc2_wgrad, (long long)G * O * NH);
   L(cnn_fc2_dgrad, GB * NH);
-  L(cnn_fc1_wgrad, (long long)G * NH * NF);
-  L(cnn_fc1_dgrad, GB * NF);
+  hipLaunchKernelGGL(cnn_fc1_wgrad, dim3(G * (NF / 128)), dim3(WG), 0, s,
+                     a);
+  L(cnn_fc1_bias_grad, (long long)G * NH);
+  hipLaunchKernelGGL(cnn_fc1_dgrad, dim3(G * (NF / 256)), dim3(WG), 0, s,
+                     a);
   L(cnn_pool_bwd, GB * NF);
-  hipLaunchKernelGGL(cnn_conv2_wgrad, dim3(G * C2), dim3(WG), 0, s, a);
-  L(cnn_conv2_dgrad, GB * X1N);
+  hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * 9 * W2_KS), dim3(WG),
+                     0, s, a);
+  L(cnn_conv2_wgrad_reduce, (long long)G * 9 * 2048);
+  hipLaunchKernelGGL(cnn_conv2_bias_grad, dim3(G * C2), dim3(WG), 0, s, a);
+  hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB * 11), dim3(WG), 0,
+                     s, a);
   hipLaunchKernelGGL(cnn_conv1_wgrad_part, dim3(G * (int)B), dim3(WG), 0,
                      s, a);
   L(cnn_conv1_wgrad_reduce, (long long)G * 320);
+  // optimizer
   L(cnn_opt_step, (long long)G * a.P);
   hipLaunchKernelGGL(cnn_opt_tick, dim3((G + WG - 1) / WG), dim3(WG), 0, s,
                      a);
