@@ -98,7 +98,7 @@ class CnnHipEngine:
             "c1part": f(G * B * 320),
             "wtf": f(G * 9 * 2048),
             "wtd": f(G * 9 * 2048),
-            "z1part": f(G * 4 * B * NH),
+            "z1part": f(G * 8 * B * NH),
             "w2part": f(G * 9 * 8 * 2048),
             "grad": f(G, self.P),
         }

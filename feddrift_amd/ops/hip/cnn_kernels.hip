@@ -185,6 +185,9 @@ void cnn_w2_reshape(CnnArgs a) {
     const float wv = a.work[(long long)g * a.P + OFF_W2C + r];
     a.wtf[((long long)g * 9 + kyx) * 2048 + ci * C2 + co] = wv;
     a.wtd[((long long)g * 9 + kyx) * 2048 + co * C1 + ci] = wv;
+    // zero the conv2 bias-grad slice (pool_bwd accumulates into it)
+    if (t == 0 && co < C2)
+      a.grad[(long long)g * a.P + OFF_B2C + co] = 0.f;
   }
 }
 
@@ -244,41 +247,53 @@ void cnn_conv2_fwd_mfma(CnnArgs a) {
   }
 }
 
-// maxpool + dropout1 over the channels-last z2: thread per (g, b, e)
-// where e = c*144 + py*12 + px is the torch-flatten a2 layout
+// maxpool + dropout1 over the channels-last z2, block per (g, b):
+// z2 rows stage through LDS (coalesced channels-last reads), the pooled
+// a2/pidx tiles assemble in LDS and store with one coalesced pass in the
+// torch-flatten [c][py][px] layout
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_pool_fwd(CnnArgs a) {
-  const long long total = (long long)a.G * a.B * NF;
-  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
-       q += (long long)gridDim.x * WG) {
-    const int g = (int)(q / ((long long)a.B * NF));
-    const long long r = q - (long long)g * a.B * NF;
-    const int b = (int)(r / NF);
-    if (b >= step_n(a, g)) continue;
-    const int e = (int)(r - (long long)b * NF);
-    const int c = e / (SP * SP);
-    const int p = e - c * SP * SP;
-    const int py = p / SP, px = p - (p / SP) * SP;
-    const float* z2 = a.zz2 + ((long long)g * a.B + b) * Z2N;
-    float best = -1e30f;
-    int arg = 0;
+  const int g = blockIdx.x / a.B;
+  const int b = blockIdx.x - g * a.B;
+  if (b >= step_n(a, g)) return;
+  const int tid = threadIdx.x;
+  __shared__ __attribute__((aligned(16))) float srow[2 * S2 * C2];
+  __shared__ __attribute__((aligned(16))) float sa2[NF];
+  __shared__ unsigned char spidx[NF];
+  const float* z2 = a.zz2 + ((long long)g * a.B + b) * Z2N;
+  for (int py = 0; py < SP; ++py) {
+    for (int q = tid; q < 2 * S2 * C2; q += WG)
+      srow[q] = z2[(long long)(2 * py) * S2 * C2 + q];
+    __syncthreads();
+    for (int q = tid; q < C2 * SP; q += WG) {
+      const int c = q / SP;
+      const int px = q - c * SP;
+      float best = -1e30f;
+      int arg = 0;
 #pragma unroll
-    for (int dy = 0; dy < 2; ++dy)
+      for (int dy = 0; dy < 2; ++dy)
 #pragma unroll
-      for (int dx = 0; dx < 2; ++dx) {
-        const float z = z2[(long long)((2 * py + dy) * S2 + 2 * px + dx)
-                           * C2 + c];
-        if (z > best) { best = z; arg = dy * 2 + dx; }
-      }
-    a.pidx[q] = (unsigned char)arg;
-    a.a2[q] = best * drop_scale(a, 0, g, b, e);
+        for (int dx = 0; dx < 2; ++dx) {
+          const float z = srow[(dy * S2 + 2 * px + dx) * C2 + c];
+          if (z > best) { best = z; arg = dy * 2 + dx; }
+        }
+      sa2[c * (SP * SP) + py * SP + px] = best;
+      spidx[c * (SP * SP) + py * SP + px] = (unsigned char)arg;
+    }
+    __syncthreads();
+  }
+  float* a2o = a.a2 + ((long long)g * a.B + b) * NF;
+  unsigned char* po = a.pidx + ((long long)g * a.B + b) * NF;
+  for (int e = tid; e < NF; e += WG) {
+    a2o[e] = sa2[e] * drop_scale(a, 0, g, b, e);
+    po[e] = spidx[e];
   }
 }
 
 // fc1 forward as MFMA GEMM with K split over blocks (fills the chip at
 // small fleets): block (g, mtile, ks) computes partial z1 for 64 batch
 // rows x 128 h over K-range [ks*NF/KS, ...). Partials land in z1part.
-#define FC1_KS 4
+#define FC1_KS 8
 
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_fc1_fwd_mfma(CnnArgs a) {
@@ -538,32 +553,45 @@ void cnn_fc1_dgrad(CnnArgs a) {
   }
 }
 
-// pool backward: route da2 (through dropout1) to the argmax position;
-// dz2 lands CHANNELS-LAST in zz2 (reused as dz buffer) for the conv2
-// MFMA backward stagers; the other 3 cell positions are written zero
+// pool backward, block per (g, b): da2 reads coalesced (flatten layout,
+// dropout reapplied), routed dz2 writes coalesced channels-last via an
+// LDS-staged transpose; the conv2 bias grad (sum of routed dz) folds in
+// as one atomicAdd per (pair, channel) per sample block
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_pool_bwd(CnnArgs a) {
-  const long long total = (long long)a.G * a.B * NF;
-  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
-       q += (long long)gridDim.x * WG) {
-    const int g = (int)(q / ((long long)a.B * NF));
-    const long long r = q - (long long)g * a.B * NF;
-    const int b = (int)(r / NF);
-    if (b >= step_n(a, g)) continue;
-    const int e = (int)(r - (long long)b * NF);
-    const int c = e / (SP * SP);
-    const int p = e - c * SP * SP;
-    const int py = p / SP, px = p - (p / SP) * SP;
-    const float dv = a.da2[q] * drop_scale(a, 0, g, b, e);
-    const int arg = a.pidx[q];
-    float* dzc = a.zz2 + ((long long)g * a.B + b) * Z2N;
-#pragma unroll
-    for (int dy = 0; dy < 2; ++dy)
-#pragma unroll
-      for (int dx = 0; dx < 2; ++dx)
-        dzc[(long long)((2 * py + dy) * S2 + 2 * px + dx) * C2 + c] =
-            (dy * 2 + dx == arg) ? dv : 0.f;
+  const int g = blockIdx.x / a.B;
+  const int b = blockIdx.x - g * a.B;
+  if (b >= step_n(a, g)) return;
+  const int tid = threadIdx.x;
+  __shared__ __attribute__((aligned(16))) float sda[NF];
+  __shared__ float sbias[C2];
+  const float* da2 = a.da2 + ((long long)g * a.B + b) * NF;
+  for (int e = tid; e < NF; e += WG)
+    sda[e] = da2[e] * drop_scale(a, 0, g, b, e);
+  if (tid < C2) sbias[tid] = 0.f;
+  __syncthreads();
+  const unsigned char* pidx = a.pidx + ((long long)g * a.B + b) * NF;
+  float* dzc = a.zz2 + ((long long)g * a.B + b) * Z2N;
+  float bacc = 0.f;
+  const int c = tid & 63;            // fixed per thread (WG % 64 == 0)
+  for (int q = tid; q < Z2N; q += WG) {
+    const int p2 = q >> 6;           // z2 pixel
+    const int oy = p2 / S2, ox = p2 - (p2 / S2) * S2;
+    const int e = c * (SP * SP) + (oy >> 1) * SP + (ox >> 1);
+    const int arg = (oy & 1) * 2 + (ox & 1);
+    const float v = ((int)pidx[e] == arg) ? sda[e] : 0.f;
+    dzc[q] = v;
+    bacc += v;
   }
+  // per-channel reduce (4 threads share each c) + one atomic per c
+  for (int off = 192; off > 0; off -= 64) {
+    __syncthreads();
+    if (tid >= off && tid < off + 64) sbias[c] += bacc;
+  }
+  __syncthreads();
+  if (tid < C2)
+    atomicAdd(&a.grad[(long long)g * a.P + OFF_B2C + tid],
+              sbias[tid] + bacc);
 }
 
 // conv2 wgrad as MFMA GEMM: dW[(kyx, ci)][co] = sum_m A[m, ci] dz[m, co]
@@ -653,32 +681,6 @@ void cnn_conv2_wgrad_reduce(CnnArgs a) {
     for (int ms = 0; ms < W2_KS; ++ms) s += part[(long long)ms * 2048];
     a.grad[(long long)g * a.P + OFF_W2C + (co * C1 + ci) * 9 + kyx] = s;
   }
-}
-
-// conv2 bias grad: block per (g, co), strided sum over (b, pixel)
-extern "C" __global__ __launch_bounds__(WG)
-void cnn_conv2_bias_grad(CnnArgs a) {
-  const int g = blockIdx.x / C2;
-  const int co = blockIdx.x - g * C2;
-  const int n = step_n(a, g);
-  if (n == 0) return;
-  const int tid = threadIdx.x;
-  __shared__ float sred[4];
-  float s = 0.f;
-  const long long mtot = (long long)n * (S2 * S2);
-  const float* dz = a.zz2 + (long long)g * a.B * Z2N;
-  for (long long m = tid; m < mtot; m += WG) {
-    const int b = (int)(m / (S2 * S2));
-    const int p = (int)(m - (long long)b * (S2 * S2));
-    s += dz[((long long)b * 576 + p) * C2 + co];
-  }
-  for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
-  const int lane = tid & 63, wave = tid >> 6;
-  if (lane == 0) sred[wave] = s;
-  __syncthreads();
-  if (tid == 0)
-    a.grad[(long long)g * a.P + OFF_B2C + co] =
-        sred[0] + sred[1] + sred[2] + sred[3];
 }
 
 // conv2 dgrad as MFMA GEMM: dx1[m=(b,y,x)][ci] over K = (kyx, co) with
@@ -1132,7 +1134,7 @@ void cnn_train_epoch_impl(
   L(cnn_conv1_fwd, GB * X1N);
   hipLaunchKernelGGL(cnn_conv2_fwd_mfma, dim3((int)GB * 9), dim3(WG), 0,
                      s, a);
-  L(cnn_pool_fwd, GB * NF);
+  hipLaunchKernelGGL(cnn_pool_fwd, dim3((int)GB), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_fc1_fwd_mfma, dim3(G * mtiles * FC1_KS), dim3(WG),
                      0, s, a);
   L(cnn_fc1_act, GB * NH);
@@ -1145,11 +1147,10 @@ void cnn_train_epoch_impl(
   L(cnn_fc1_bias_grad, (long long)G * NH);
   hipLaunchKernelGGL(cnn_fc1_dgrad, dim3(G * (NF / 256)), dim3(WG), 0, s,
                      a);
-  L(cnn_pool_bwd, GB * NF);
+  hipLaunchKernelGGL(cnn_pool_bwd, dim3((int)GB), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * 9 * W2_KS), dim3(WG),
                      0, s, a);
   L(cnn_conv2_wgrad_reduce, (long long)G * 9 * 2048);
-  hipLaunchKernelGGL(cnn_conv2_bias_grad, dim3(G * C2), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB * 11), dim3(WG), 0,
                      s, a);
   hipLaunchKernelGGL(cnn_conv1_wgrad_part, dim3(G * (int)B), dim3(WG), 0,
