@@ -100,6 +100,7 @@ class CnnHipEngine:
             "wtd": f(G * 9 * 2048),
             "z1part": f(G * 8 * B * NH),
             "w2part": f(G * 9 * 8 * 2048),
+            "b2part": f(G * B * 64),
             "grad": f(G, self.P),
         }
         self._ws = ws
@@ -161,7 +162,7 @@ class CnnHipEngine:
                 ws["x1"], ws["a2"], ws["pidx"], ws["z1"], ws["a1"],
                 ws["dz2"], ws["dz1"], ws["da2"], ws["zz2"], ws["dx1"],
                 ws["c1part"], ws["wtf"], ws["wtd"], ws["z1part"],
-                ws["w2part"],
+                ws["w2part"], ws["b2part"],
                 opt["m"] if adam else None,
                 opt["v"] if adam else None,
                 opt["vmax"] if adam else None,

@@ -83,6 +83,7 @@ struct CnnArgs {
   float* __restrict__ wtd;           // [G, 9, 64, 32] conv2 W (dgrad)
   float* __restrict__ z1part;        // [G, FC1_KS, B, NH]
   float* __restrict__ w2part;        // [G, 9, W2_KS, 32, 64]
+  float* __restrict__ b2part;        // [G, B, 64] conv2 bias partials
   // optimizer state (indexed by rows[g])
   float* __restrict__ m;
   float* __restrict__ v;
@@ -185,9 +186,6 @@ void cnn_w2_reshape(CnnArgs a) {
     const float wv = a.work[(long long)g * a.P + OFF_W2C + r];
     a.wtf[((long long)g * 9 + kyx) * 2048 + ci * C2 + co] = wv;
     a.wtd[((long long)g * 9 + kyx) * 2048 + co * C1 + ci] = wv;
-    // zero the conv2 bias-grad slice (pool_bwd accumulates into it)
-    if (t == 0 && co < C2)
-      a.grad[(long long)g * a.P + OFF_B2C + co] = 0.f;
   }
 }
 
@@ -555,8 +553,8 @@ void cnn_fc1_dgrad(CnnArgs a) {
 
 // pool backward, block per (g, b): da2 reads coalesced (flatten layout,
 // dropout reapplied), routed dz2 writes coalesced channels-last via an
-// LDS-staged transpose; the conv2 bias grad (sum of routed dz) folds in
-// as one atomicAdd per (pair, channel) per sample block
+// LDS-staged transpose; the conv2 bias grad (sum of routed dz) emits a
+// per-sample partial (b2part) that the wgrad reduce sums deterministically
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_pool_bwd(CnnArgs a) {
   const int g = blockIdx.x / a.B;
@@ -583,15 +581,16 @@ void cnn_pool_bwd(CnnArgs a) {
     dzc[q] = v;
     bacc += v;
   }
-  // per-channel reduce (4 threads share each c) + one atomic per c
+  // per-channel reduce (4 threads share each c), then a per-sample
+  // partial into the dz2 scratch ([G, B, 64], free after fc2 backward);
+  // cnn_conv2_wgrad_reduce sums it deterministically
   for (int off = 192; off > 0; off -= 64) {
     __syncthreads();
     if (tid >= off && tid < off + 64) sbias[c] += bacc;
   }
   __syncthreads();
   if (tid < C2)
-    atomicAdd(&a.grad[(long long)g * a.P + OFF_B2C + tid],
-              sbias[tid] + bacc);
+    a.b2part[((long long)g * a.B + b) * C2 + tid] = sbias[tid] + bacc;
 }
 
 // conv2 wgrad as MFMA GEMM: dW[(kyx, ci)][co] = sum_m A[m, ci] dz[m, co]
@@ -662,18 +661,30 @@ void cnn_conv2_wgrad_mfma(CnnArgs a) {
   }
 }
 
-// reduce the conv2 wgrad msplit partials into the (co, ci, ky, kx) grad
+// reduce the conv2 wgrad msplit partials into the (co, ci, ky, kx) grad;
+// the tail entries (one per (g, co)) reduce the bias partials from
+// pool_bwd over the batch — all deterministic
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_conv2_wgrad_reduce(CnnArgs a) {
-  const long long total = (long long)a.G * 9 * 2048;
+  const long long per_g = 9 * 2048 + C2;
+  const long long total = (long long)a.G * per_g;
   for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
        q += (long long)gridDim.x * WG) {
-    const int g = (int)(q / (9 * 2048));
-    const int r = (int)(q - (long long)g * 9 * 2048);
+    const int g = (int)(q / per_g);
+    const int r = (int)(q - (long long)g * per_g);
+    const int n = step_n(a, g);
+    if (n == 0) continue;
+    if (r >= 9 * 2048) {
+      const int co = r - 9 * 2048;
+      float s = 0.f;
+      for (int b = 0; b < n; ++b)
+        s += a.b2part[((long long)g * a.B + b) * C2 + co];
+      a.grad[(long long)g * a.P + OFF_B2C + co] = s;
+      continue;
+    }
     const int kyx = r / 2048;
     const int t = r - kyx * 2048;
     const int ci = t / C2, co = t - (t / C2) * C2;
-    if (step_n(a, g) == 0) continue;
     const float* part = a.w2part + ((long long)g * 9 + kyx) * W2_KS * 2048
                         + t;
     float s = 0.f;
@@ -1075,6 +1086,7 @@ void cnn_train_epoch_impl(
     torch::Tensor ws_dz1, torch::Tensor ws_da2, torch::Tensor ws_zz2,
     torch::Tensor ws_dx1, torch::Tensor ws_c1part, torch::Tensor ws_wtf,
     torch::Tensor ws_wtd, torch::Tensor ws_z1part, torch::Tensor ws_w2part,
+    torch::Tensor ws_b2part,
     c10::optional<torch::Tensor> m, c10::optional<torch::Tensor> v,
     c10::optional<torch::Tensor> vmax, c10::optional<torch::Tensor> t,
     torch::Tensor lr, double wd, double p1, double p2,
@@ -1105,6 +1117,7 @@ void cnn_train_epoch_impl(
   a.wtd = ws_wtd.data_ptr<float>();
   a.z1part = ws_z1part.data_ptr<float>();
   a.w2part = ws_w2part.data_ptr<float>();
+  a.b2part = ws_b2part.data_ptr<float>();
   const bool adam = m.has_value();
   a.m = adam ? m->data_ptr<float>() : nullptr;
   a.v = adam ? v->data_ptr<float>() : nullptr;
@@ -1150,7 +1163,7 @@ void cnn_train_epoch_impl(
   hipLaunchKernelGGL(cnn_pool_bwd, dim3((int)GB), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * 9 * W2_KS), dim3(WG),
                      0, s, a);
-  L(cnn_conv2_wgrad_reduce, (long long)G * 9 * 2048);
+  L(cnn_conv2_wgrad_reduce, (long long)G * (9 * 2048 + C2));
   hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB * 11), dim3(WG), 0,
                      s, a);
   hipLaunchKernelGGL(cnn_conv1_wgrad_part, dim3(G * (int)B), dim3(WG), 0,
