@@ -345,23 +345,44 @@ class SoftClusterAlgo(AlgoBase):
 
     def _cfl_round(self, job: FLJob, round_idx: int, plan: TrainPlan,
                    client_idx: np.ndarray) -> bool:
-        """CFL split check needs each client's raw weight update; gather the
-        flattened deltas (small: active pairs only) to every rank
-        (reference cluster_cfl, FedAvgEnsDataLoader.py:1159-1223)."""
+        """CFL split check needs each client's raw weight update
+        (reference cluster_cfl, FedAvgEnsDataLoader.py:1159-1223).
+
+        Redesigned for scale: the active (worker, model) set is agreed via
+        one small all_reduce, then the flat deltas travel as ONE dense
+        device tensor all_reduce over RCCL/xGMI (each rank owns disjoint
+        rows, sum = gather) — no pickled-object gathers; at ResNet-18 x
+        100 clients the old all_gather_object serialized O(pairs x 11M)
+        floats through the host per split check. The collective is
+        chunked to bound peak traffic per call."""
         st = self.state
         K, P = job.n_models, job.n_params   # module path: full state size
-        local: Dict[tuple, np.ndarray] = {}
+        W = job.n_workers
         nW = len(job.owned_workers)
         reps = job.replicas.reshape(nW, K, P)
+        act = torch.zeros(W, K, device=job.device)
         for wi, w in enumerate(job.owned_workers):
             for m in range(K):
                 if plan.sample_num[wi, m] > 0:
-                    delta = (reps[wi, m] - job.global_params[m]).cpu().numpy()
-                    local[(int(client_idx[w]), m)] = delta
-        gathered = job.comm.all_gather_object(local)
-        updates: Dict[tuple, np.ndarray] = {}
-        for d in gathered:
-            updates.update(d)
+                    act[w, m] = 1.0
+        job.comm.all_reduce_(act)
+        act_np = act.cpu().numpy() > 0
+        pairs = [(w, m) for w in range(W) for m in range(K)
+                 if act_np[w, m]]
+        idx_of = {p: i for i, p in enumerate(pairs)}
+        deltas = torch.zeros(len(pairs), P, device=job.device)
+        for wi, w in enumerate(job.owned_workers):
+            for m in range(K):
+                if plan.sample_num[wi, m] > 0:
+                    deltas[idx_of[(w, m)]] = \
+                        reps[wi, m] - job.global_params[m]
+        rows_per_chunk = max(1, (256 << 20) // max(4 * P, 1))
+        for i0 in range(0, len(pairs), rows_per_chunk):
+            job.comm.all_reduce_(deltas[i0:i0 + rows_per_chunk])
+        dall = deltas.cpu().numpy()
+        updates: Dict[tuple, np.ndarray] = {
+            (int(client_idx[w]), m): dall[i]
+            for i, (w, m) in enumerate(pairs)}
         models_in_use = [m for m in range(K)
                          if np.any(st.get_weights()[job.curr_iter][m] > 0)]
         clients_by_model = {

@@ -773,9 +773,17 @@ class FLJob:
     def save_state_pickle(self, name: str, state) -> None:
         if self.cfg.bench_mode:
             return
+        # barrier BEFORE the write: every rank must be done reading the
+        # previous state file before root truncates it (at world size 4
+        # the iteration-start load on slow ranks raced root's re-save and
+        # read a partial pickle); write-to-temp + atomic rename so a
+        # reader never sees a half-written file; barrier after publishes.
+        self.comm.barrier()
         if self.comm.is_root:
-            with open(self.ckpt_path(name), "wb") as f:
+            tmp = self.ckpt_path(name) + ".tmp"
+            with open(tmp, "wb") as f:
                 pickle.dump(state, f)
+            os.replace(tmp, self.ckpt_path(name))
         self.comm.barrier()
 
     def load_state_pickle(self, name: str):
