@@ -64,6 +64,8 @@ class Config:
     bench_mode: int = 0                  # 1: synthetic steady-state (bench.py)
 
     # robust aggregation (fedavg_robust / fedml_core robustness equivalent)
+    secure_agg: int = 0                  # 1: pairwise-mask uploads
+                                         # (turboaggregate equivalent)
     robust_norm_bound: float = 0.0       # >0: clip client updates to this L2
     robust_noise: float = 0.0            # >0: Gaussian noise stddev on avg
     # FedOpt server optimizer ('avg' = plain FedAvg replacement)

@@ -276,3 +276,47 @@ def test_world4_cfl_tensor_gather(tmp_path):
         globals()["_WORKER"] = saved
     assert abs(r1[0] - r4[0]) < 1e-6, (r1, r4)
     assert np.allclose(r1, r4, atol=0.02), (r1, r4)
+
+
+_WORKER_SECURE = r"""
+import json, os, sys
+sys.path.insert(0, {repo!r})
+import numpy as np
+from feddrift_amd.comm import Communicator
+from feddrift_amd.config import Config
+from feddrift_amd.engine.timeline import run_timeline
+
+cfg = Config(model="fnn", dataset="sea", data_dir={data!r},
+             client_num_in_total=6, client_num_per_round=6,
+             batch_size=300, lr=0.01, epochs=5, comm_round=6,
+             total_train_iteration=3, concept_num=2,
+             concept_drift_algo="softcluster",
+             concept_drift_algo_arg="H_A_C_1_10_0",
+             change_points="T", dummy_arg=0, log_dir={log!r},
+             report_client=0, secure_agg=int(os.environ["FD_SECURE"]))
+comm = Communicator()
+out = run_timeline(cfg, comm)
+if comm.is_root:
+    with open(os.path.join({log!r}, "result.json"), "w") as f:
+        json.dump(out["per_iteration_test_acc"], f)
+"""
+
+
+def test_world2_secure_agg_matches_plain(tmp_path):
+    """Secure aggregation across ranks: every rank's all_reduce input is
+    masked (pairwise turboaggregate-style), yet the 2-rank masked run
+    reproduces the 2-rank plain run (masks cancel in the collective)."""
+    global _WORKER
+    data = _write_data(tmp_path)
+    saved = _WORKER
+    try:
+        globals()["_WORKER"] = _WORKER_SECURE
+        os.environ["FD_SECURE"] = "0"
+        r_plain = _run(2, data, str(tmp_path / "sp"), 29623)
+        os.environ["FD_SECURE"] = "1"
+        r_sec = _run(2, data, str(tmp_path / "ss"), 29624)
+    finally:
+        globals()["_WORKER"] = saved
+        os.environ.pop("FD_SECURE", None)
+    assert abs(r_plain[0] - r_sec[0]) < 5e-3, (r_plain, r_sec)
+    assert np.allclose(r_plain, r_sec, atol=0.02), (r_plain, r_sec)
