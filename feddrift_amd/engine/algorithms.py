@@ -578,23 +578,21 @@ class DriftSurfAlgo(AlgoBase):
 
     def _score_newest(self, job: FLJob, view) -> None:
         """run_ds_algo at data-load time (DriftSurf_data_loader:269-314):
-        score the pickled models on the newest global batch (CPU eval; runs
-        once per iteration, data is host-side at this point)."""
+        score the pickled models on the newest global batch. One batched
+        device sweep for ALL candidate models, sharded by client +
+        allreduced (job.score_models_on_segments) — replaces the
+        reference's per-(model, client) CPU eager forwards."""
+        keys = [k for k in ("pred", "stab", "reac")
+                if self.state.models.get(k) is not None]
+        flats = [np.asarray(self.state.models[k]) for k in keys]
+        correct, total = job.score_models_on_segments(flats, view.train)
+        accs = {}
+        for i, k in enumerate(keys):
+            t = total[i].sum()
+            accs[k] = float(correct[i].sum() / t) if t else 0.0
 
-        def score(key: str) -> float:
-            flat = self.state.models.get(key)
-            if flat is None:
-                return 0.0
-            correct = total = 0.0
-            for c, seg in view.train.items():
-                if seg.n == 0:
-                    continue
-                logits = job.forward_flat(np.asarray(flat), seg.x)
-                correct += float((logits.argmax(-1) == seg.y).sum())
-                total += seg.n
-            return correct / total if total else 0.0
-
-        self.state.run_ds_algo(score, job.curr_iter)
+        self.state.run_ds_algo(lambda key: accs.get(key, 0.0),
+                               job.curr_iter)
 
     def load_checkpoint(self, job: FLJob) -> None:
         # handled via ds_state (FedAvgEnsAggregatorDriftSurf.py:45-64)
@@ -680,24 +678,23 @@ class MultiModelAlgo(AlgoBase):
 
     def _run_select(self, job: FLJob, newest_view) -> None:
         """run_model_select with per-(model, client) scoring on the newest
-        local batch (reference FedAvgEnsDataLoader.py:350-390)."""
-        cache: Dict[tuple, float] = {}
+        local batch (reference FedAvgEnsDataLoader.py:350-390). All
+        (model, client) accuracies come from ONE batched device sweep
+        (job.score_models_on_segments), replacing the reference's fresh
+        per-call CPU eager forwards."""
+        present = [m for m in range(job.cfg.concept_num)
+                   if self.state.models.get(m) is not None]
+        flats = [np.asarray(self.state.models[m]) for m in present]
+        correct, total = job.score_models_on_segments(flats,
+                                                      newest_view.train)
+        acc = {}
+        for i, m in enumerate(present):
+            for c in range(job.cfg.client_num_in_total):
+                if total[i, c] > 0:
+                    acc[(m, c)] = float(correct[i, c] / total[i, c])
 
-        def score(m: int, c: int) -> float:
-            if (m, c) in cache:
-                return cache[(m, c)]
-            flat = self.state.models.get(m)
-            if flat is None or c not in newest_view.train:
-                return 0.0
-            seg = newest_view.train[c]
-            if seg.n == 0:
-                return 0.0
-            logits = job.forward_flat(np.asarray(flat), seg.x)
-            acc = float((logits.argmax(-1) == seg.y).sum()) / seg.n
-            cache[(m, c)] = acc
-            return acc
-
-        self.state.run_model_select(score, job.curr_iter)
+        self.state.run_model_select(lambda m, c: acc.get((m, c), 0.0),
+                                    job.curr_iter)
 
     def load_checkpoint(self, job: FLJob) -> None:
         cfg = job.cfg

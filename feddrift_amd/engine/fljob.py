@@ -380,24 +380,64 @@ class FLJob:
             task_row, task_id, win_off, win_len, n_tasks, n_classes,
             x_mask=x_mask)
 
-    def forward_flat(self, flat: np.ndarray, x: np.ndarray) -> np.ndarray:
-        """Model output for a flat parameter vector on host data (used by
-        the DriftSurf / MultiModelAcc data-load-time scoring, which runs
-        before the device arena exists). CPU eval, once per iteration."""
-        xt = torch.from_numpy(np.ascontiguousarray(x, dtype=np.float32))
-        ft = torch.from_numpy(np.ascontiguousarray(flat, dtype=np.float32))
+    def score_models_on_segments(self, flats, segs) -> tuple:
+        """Batched accuracy of M flat parameter vectors on per-client
+        host segments, used by the DriftSurf / MultiModelAcc data-load-
+        time scoring (reference runs these as per-(model, client) CPU
+        eager forwards at data-load time, DriftSurf_data_loader:269-314,
+        FedAvgEnsDataLoader.py:350-390 — at FEMNIST scale that is
+        minutes of host time per iteration). Here: one throwaway device
+        arena + ONE batched eval sweep, sharded by client + allreduced.
+
+        Returns ([M, C] correct, [M, C] total) numpy arrays."""
+        M = len(flats)
+        C = self.cfg.client_num_in_total
+        dev = self.device
+        if M == 0:
+            return (np.zeros((0, C)), np.zeros((0, C)))
+        params = torch.as_tensor(
+            np.ascontiguousarray(np.stack(flats), dtype=np.float32),
+            device=dev)
+        xs, ys, wins = [], [], {}
+        off = 0
+        for c, seg in segs.items():
+            if not self.comm.owns_client(c) or seg.n == 0:
+                continue
+            xs.append(np.asarray(seg.x, dtype=np.float32))
+            ys.append(np.asarray(seg.y, dtype=np.int64))
+            wins[c] = (off, seg.n)
+            off += seg.n
+        tl = TaskList()
+        ids = {}
+        for m in range(M):
+            for c in range(C):
+                tid = tl.new_task()
+                ids[(m, c)] = tid
+                if c in wins:
+                    tl.add_windows(tid, m, [wins[c]])
+        if xs:
+            x_t = torch.as_tensor(np.concatenate(xs, 0), device=dev)
+            y_t = torch.as_tensor(np.concatenate(ys, 0), device=dev)
+        else:
+            x_t = torch.zeros(1, self.dataset.feature_num, device=dev)
+            y_t = torch.zeros(1, dtype=torch.int64, device=dev)
+        idx = self.eval_tensors(tl)
         if self.is_module_path:
-            proto = zoo.create_model(self.cfg.model, self.dataset.class_num,
-                                     self.dataset.feature_num)
-            self.packer.load_into(proto, ft)
-            proto.eval()
-            with torch.no_grad():
-                return proto(xt).numpy()
-        from ..ops import mlp_torch
-        with torch.no_grad():
-            return mlp_torch.forward_logits(
-                self.spec, ft.reshape(1, -1), xt.unsqueeze(0)) \
-                .squeeze(0).numpy()
+            res = self.mod_engine.eval_tasks_stacked(
+                params, idx[0], idx[1], idx[2], idx[3], tl.n_tasks,
+                x_arena=x_t, y_arena=y_t)
+        else:
+            res = self.backend.eval_tasks_stacked(
+                self.spec, params, x_t, y_t, idx[0], idx[1], idx[2],
+                idx[3], tl.n_tasks)
+        self.comm.all_reduce_(res)
+        cv = res.cpu().numpy()
+        correct = np.zeros((M, C))
+        total = np.zeros((M, C))
+        for (m, c), tid in ids.items():
+            correct[m, c] = cv[0][tid]
+            total[m, c] = cv[1][tid]
+        return correct, total
 
     def train(self, plan: TrainPlan) -> None:
         """All local training of this round in ONE fused launch. On the HIP
