@@ -121,6 +121,12 @@ def sample_femnist(n, concept, rng):
     return _sample_image_like(n, concept, rng, 62, 784, 1703)
 
 
+def sample_fmow(n, concept, rng):
+    # synthetic twin of the FMoW real-drift features (real path:
+    # data/real.py FmowIndexStore over the reference partition layout)
+    return _sample_image_like(n, concept, rng, 62, 1024, 2027)
+
+
 TEXT_SEQ_LEN = 20
 TEXT_VOCAB = 30
 
@@ -146,12 +152,15 @@ def sample_text(n: int, concept: int, rng: np.random.Generator) -> np.ndarray:
 
 _SAMPLERS = {"sea": sample_sea, "sine": sample_sine, "circle": sample_circle,
              "MNIST": sample_mnist, "cifar": sample_cifar,
-             "femnist": sample_femnist, "text": sample_text}
+             "femnist": sample_femnist, "text": sample_text,
+             "fmow": sample_fmow}
 
 FEATURE_NUM = {"sea": 3, "sine": 2, "circle": 2, "MNIST": 784,
-               "cifar": 3072, "femnist": 784, "text": TEXT_SEQ_LEN}
+               "cifar": 3072, "femnist": 784, "text": TEXT_SEQ_LEN,
+               "fmow": 1024}
 CLASS_NUM = {"sea": 2, "sine": 2, "circle": 2, "MNIST": 10,
-             "cifar": 10, "femnist": 62, "text": TEXT_VOCAB}
+             "cifar": 10, "femnist": 62, "text": TEXT_VOCAB,
+             "fmow": 62}
 
 _SEA_COLS = ["f1", "f2", "f3", "label"]
 
@@ -204,6 +213,14 @@ def generate_data(dataset: str, data_dir: str, train_iteration: int,
     change_point = load_change_points(data_dir, change_point_str)
 
     sampler = _SAMPLERS[dataset]
+    # real-pixel source: when the LEAF json layout is present under the
+    # dataset dir (a user brought the reference's data/MNIST/train+test
+    # along), draw from it with the reference MNIST_Data semantics
+    # instead of the synthetic prototypes (data/real.py)
+    if dataset in ("MNIST", "femnist"):
+        from .real import LeafSampleSource, leaf_layout_present
+        if leaf_layout_present(ds_dir):
+            sampler = LeafSampleSource(ds_dir).generate_sample
     rng = np.random.default_rng(np.random.randint(0, 2**31))
     n_classes = CLASS_NUM[dataset]
 

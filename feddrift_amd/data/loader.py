@@ -196,9 +196,23 @@ class ClientData:
 
 
 class DriftDataset:
-    """Top-level handle: raw store + metadata."""
+    """Top-level handle: raw store + metadata. For fmow, the reference's
+    lazy partition layout (partitions/{P}/client_{c}_iter_{t}.csv index
+    files + a feature backing store) is consumed directly when present
+    on disk (data/real.py:FmowIndexStore); otherwise the synthetic twin
+    generates the same shape."""
 
-    def __init__(self, data_dir: str, dataset: str, num_client: int):
+    def __init__(self, data_dir: str, dataset: str, num_client: int,
+                 partition: str = "A"):
+        if dataset == "fmow":
+            from .real import FmowIndexStore
+            if FmowIndexStore.layout_present(data_dir, partition):
+                self.store = FmowIndexStore(data_dir, partition,
+                                            num_client)
+                self.dataset = "fmow"
+                self.feature_num = int(self.store.features.shape[1])
+                self.class_num = int(self.store.labels.max()) + 1
+                return
         self.store = RawStore(data_dir, dataset, num_client)
         self.dataset = self.store.dataset
         self.feature_num = FEATURE_NUM[self.dataset]

@@ -170,8 +170,10 @@ class FLJob:
         zoo.set_torch_seed(cfg.dummy_arg)
 
         self.curr_iter = cfg.curr_train_iteration
-        self.dataset = dataset or DriftDataset(cfg.data_dir, cfg.dataset,
-                                               cfg.client_num_in_total)
+        self.dataset = dataset or DriftDataset(
+            cfg.data_dir, cfg.dataset, cfg.client_num_in_total,
+            partition=cfg.change_points)  # fmow: partition letter = CP name
+                                          # (fmow/data_loader.py:65)
         # two model paths: the MLP family runs on the fused HIP kernels /
         # batched torch ops; convolutional models (cnn / resnet) run as one
         # live torch module per rank over flat state rows (MIOpen convs)
