@@ -20,11 +20,16 @@ def load():
     global _mod
     if _mod is not None:
         return _mod
-    so = os.path.join(BUILD_DIR, MODULE_NAME + ".so")
+    # FEDDRIFT_HIP_SO: explicit .so override (the sanitizer lane points
+    # this at the ASAN build of the same extension)
+    so = os.environ.get("FEDDRIFT_HIP_SO") or \
+        os.path.join(BUILD_DIR, MODULE_NAME + ".so")
     if not os.path.exists(so):
         raise FileNotFoundError(
             f"{so} not found — run `python -m feddrift_amd.ops.build` first")
-    spec = importlib.util.spec_from_file_location(MODULE_NAME, so)
+    # module name must match the .so's PyInit_<name> symbol
+    mod_name = os.path.splitext(os.path.basename(so))[0]
+    spec = importlib.util.spec_from_file_location(mod_name, so)
     _mod = importlib.util.module_from_spec(spec)
     spec.loader.exec_module(_mod)
     return _mod
