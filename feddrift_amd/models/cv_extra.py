@@ -173,3 +173,86 @@ class DenseNet(nn.Module):
 
 def densenet121(num_classes: int = 1000):
     return DenseNet((6, 12, 24, 16), 32, 64, num_classes)
+
+
+class CifarBottleneck(nn.Module):
+    """Bottleneck block (1x1 -> 3x3 -> 1x1, expansion 4) — the reference
+    GKT server model is built from these
+    (resnet56_gkt/resnet_server.py:70-110)."""
+    expansion = 4
+
+    def __init__(self, inplanes, planes, stride=1, downsample=None):
+        super().__init__()
+        self.conv1 = nn.Conv2d(inplanes, planes, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = nn.Conv2d(planes, planes, 3, stride, 1, bias=False)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.conv3 = nn.Conv2d(planes, planes * 4, 1, bias=False)
+        self.bn3 = nn.BatchNorm2d(planes * 4)
+        self.downsample = downsample
+
+    def forward(self, x):
+        identity = x
+        out = F.relu(self.bn1(self.conv1(x)))
+        out = F.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        return F.relu(out + identity)
+
+
+class GKTClientNet(nn.Module):
+    """FedGKT edge model at the reference's split
+    (resnet56_gkt/resnet_client.py:112-216, resnet5_56): 3x3 stem to 16
+    channels (the EXTRACTED FEATURES that ship to the server), one local
+    16-channel stage, and a local classifier head for the client-side
+    CE + distillation loss."""
+
+    def __init__(self, num_classes: int = 10, n_blocks: int = 1):
+        super().__init__()
+        self.conv1 = nn.Conv2d(3, 16, 3, 1, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(16)
+        self.layer1 = nn.Sequential(*[
+            CifarBasicBlock(16, 16) for _ in range(n_blocks)])
+        self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
+        self.fc = nn.Linear(16, num_classes)
+
+    def forward(self, x):
+        feats = F.relu(self.bn1(self.conv1(x)))   # B x 16 x H x W
+        h = self.layer1(feats)
+        logits = self.fc(self.avgpool(h).flatten(1))
+        return logits, feats
+
+
+class GKTServerNet(nn.Module):
+    """FedGKT server model consuming the 16-channel client features
+    (resnet56_gkt/resnet_server.py:185-208, resnet56_server =
+    Bottleneck [6,6,6]): three bottleneck stages 16 -> 32 -> 64 with
+    stride-2 reductions, then the classifier."""
+
+    def __init__(self, num_classes: int = 10, n_blocks: int = 6):
+        super().__init__()
+        self.inplanes = 16
+
+        def stage(planes, blocks, stride):
+            down = None
+            if stride != 1 or self.inplanes != planes * 4:
+                down = nn.Sequential(
+                    nn.Conv2d(self.inplanes, planes * 4, 1, stride,
+                              bias=False),
+                    nn.BatchNorm2d(planes * 4))
+            layers = [CifarBottleneck(self.inplanes, planes, stride, down)]
+            self.inplanes = planes * 4
+            layers += [CifarBottleneck(self.inplanes, planes)
+                       for _ in range(1, blocks)]
+            return nn.Sequential(*layers)
+
+        self.layer1 = stage(16, n_blocks, 1)
+        self.layer2 = stage(32, n_blocks, 2)
+        self.layer3 = stage(64, n_blocks, 2)
+        self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
+        self.fc = nn.Linear(64 * 4, num_classes)
+
+    def forward(self, feats):
+        x = self.layer3(self.layer2(self.layer1(feats)))
+        return self.fc(self.avgpool(x).flatten(1))

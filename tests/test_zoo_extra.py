@@ -71,3 +71,78 @@ def test_fednas_search_converges():
     assert nas.evaluate(x, y) > 0.8
     geno = nas.genotype()
     assert len(geno) == 2 and all(0 <= g < 3 for g in geno)
+
+
+def test_darts_network_full_space():
+    """The full 8-op DARTS search supernet (models/darts.py): 14 mixed
+    edges x 8 primitives per cell, reduction cells at 1/3 and 2/3,
+    genotype derivation picks 2 input edges per node."""
+    from feddrift_amd.models.darts import PRIMITIVES, DartsNetwork
+    torch.manual_seed(0)
+    assert len(PRIMITIVES) == 8
+    net = DartsNetwork(c=4, num_classes=10, layers=3, in_ch=3)
+    assert net.alphas_normal.shape == (14, 8)
+    x = torch.randn(2, 3, 16, 16)
+    out = net(x)
+    assert out.shape == (2, 10)
+    out.sum().backward()
+    assert net.alphas_normal.grad is not None
+    gene_n, gene_r = net.genotype()
+    assert len(gene_n) == 8 and len(gene_r) == 8   # 2 edges x 4 nodes
+    for name, j in gene_n:
+        assert name in PRIMITIVES and name != "none"
+        assert 0 <= j < 5
+    # reduction structure: exactly one cell at layers//3 and 2*layers//3
+    reductions = [c.reduction for c in net.cells]
+    assert sum(reductions) == 2
+
+
+def test_fednas_darts_round_moves_alphas():
+    from feddrift_amd.engine.fednas import FedNASDarts
+    torch.manual_seed(2)
+    nas = FedNASDarts(n_clients=2, in_ch=3, n_classes=4, c=4, layers=3,
+                      w_lr=0.05, a_lr=0.01)
+    x = torch.randn(80, 3, 12, 12)
+    y = torch.randint(0, 4, (80,))
+    train = {0: (x[:20], y[:20]), 1: (x[20:40], y[20:40])}
+    val = {0: (x[40:60], y[40:60]), 1: (x[60:], y[60:])}
+    a0 = nas.global_model.alphas_normal.detach().clone()
+    w0 = nas.global_model.stem[0].weight.detach().clone()
+    nas.round(train, val, epochs=2)
+    assert (nas.global_model.alphas_normal - a0).abs().max() > 0
+    assert (nas.global_model.stem[0].weight - w0).abs().max() > 0
+    acc = nas.evaluate(x, y)
+    assert 0.0 <= acc <= 1.0
+    gene_n, gene_r = nas.genotype()
+    assert len(gene_n) == 8 and len(gene_r) == 8
+
+
+def test_fedgkt_resnet_split():
+    """The reference-scale GKT split: client stem features (16ch) feed
+    the bottleneck server; bidirectional KD trains both sides."""
+    from feddrift_amd.engine.fedgkt import FedGKTResNet
+    from feddrift_amd.models.cv_extra import GKTClientNet, GKTServerNet
+    torch.manual_seed(3)
+    cl = GKTClientNet(10)
+    sv = GKTServerNet(10, n_blocks=2)
+    x = torch.randn(4, 3, 16, 16)
+    logits, feats = cl(x)
+    assert logits.shape == (4, 10) and feats.shape == (4, 16, 16, 16)
+    assert sv(feats).shape == (4, 10)
+
+    # tiny learnable task: class = dominant color channel
+    n = 60
+    y = torch.randint(0, 3, (n,))
+    xs = torch.randn(n, 3, 8, 8) * 0.1
+    for i in range(n):
+        xs[i, y[i]] += 1.0
+    gkt = FedGKTResNet(n_clients=2, n_classes=3, client_blocks=1,
+                       server_blocks=1, lr=0.05)
+    data = {0: (xs[:30], y[:30]), 1: (xs[30:], y[30:])}
+    first = None
+    for r in range(8):
+        gkt.round(data, epochs=2)
+        acc = gkt.evaluate(0, xs[:30], y[:30])
+        if first is None:
+            first = acc
+    assert acc > max(0.5, first - 0.1), (first, acc)
