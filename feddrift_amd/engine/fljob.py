@@ -340,8 +340,7 @@ class FLJob:
                        masks: Optional[torch.Tensor] = None) -> torch.Tensor:
         """Batched many-task ensemble vote ([2, n_tasks] correct/total)."""
         if not self.is_module_path:
-            from ..ops import mlp_torch
-            return mlp_torch.ens_vote_multi(
+            return self.backend.ens_vote_multi(
                 self.spec, self.global_params, weights, self.arena.x,
                 self.arena.y, idx[1], idx[2], idx[3], tl.n_tasks,
                 mode=mode, masks=masks)

@@ -801,6 +801,146 @@ static bool launch_small_eval(const EvalArgs& args, int W,
 }
 
 // ---------------------------------------------------------------------------
+// ensemble-vote + confusion kernels (AUE/AUE-PC/KUE server paths):
+// the reference runs _infer_ens / _confusion_matrix as per-client torch
+// loops on the server (FedAvgEnsAggregatorAue.py:256-283,
+// FedAvgEnsAggregatorKue.py:234-303); here ONE launch covers every
+// (task, window) with per-task ensemble weights and per-model masks.
+// ---------------------------------------------------------------------------
+
+#define VOTE_HARD 0
+#define VOTE_SOFT 1
+
+struct VoteArgs {
+  const float* __restrict__ params;     // [M, P]
+  const float* __restrict__ weights;    // [W, M]
+  const int64_t* __restrict__ task_id;  // [W]
+  const int64_t* __restrict__ off;      // [W]
+  const int64_t* __restrict__ len;      // [W]
+  const float* __restrict__ x;
+  const int64_t* __restrict__ y;
+  const float* __restrict__ masks;      // [M, D] or null
+  double* __restrict__ correct;         // [T]
+  double* __restrict__ total;           // [T]
+  double* __restrict__ conf;            // [T, O, O] (confusion kernel)
+  const int64_t* __restrict__ task_row; // [W] (confusion kernel)
+  int M, D, H, O, P, kind, mode;
+  int mask_bcast;                       // confusion: masks is [D], not [W, D]
+};
+
+__device__ __forceinline__ void mlp_logits_for(
+    const float* __restrict__ w, const float* __restrict__ xrow,
+    const float* __restrict__ msk, int D, int H, int O, int kind,
+    float* logits) {
+  if (kind == KIND_FNN) {
+    const int HD = H * D;
+    const int OH = O * H;
+    for (int o = 0; o < O; ++o) logits[o] = w[HD + H + OH + o];
+    for (int h = 0; h < H; ++h) {
+      float z = w[HD + h];
+      for (int d = 0; d < D; ++d) {
+        float xd = xrow[d];
+        if (msk) xd *= msk[d];
+        z += w[h * D + d] * xd;
+      }
+      if (z > 0.f)
+        for (int o = 0; o < O; ++o) logits[o] += w[HD + H + o * H + h] * z;
+    }
+  } else {
+    const int OD = O * D;
+    for (int o = 0; o < O; ++o) {
+      float z = w[OD + o];
+      for (int d = 0; d < D; ++d) {
+        float xd = xrow[d];
+        if (msk) xd *= msk[d];
+        z += w[o * D + d] * xd;
+      }
+      logits[o] = 1.f / (1.f + __expf(-z));
+    }
+  }
+}
+
+extern "C" __global__ __launch_bounds__(THREADS)
+void mlp_vote_kernel(VoteArgs a) {
+  const int wdx = blockIdx.x;
+  const int n = (int)a.len[wdx];
+  const long long tsk = a.task_id[wdx];
+  const long long off = a.off[wdx];
+  const int tid = threadIdx.x;
+  extern __shared__ __attribute__((aligned(16))) float lds[];
+  float* wt = lds;            // [M] this task's ensemble weights
+  float* red = wt + a.M;      // [8]
+  for (int m = tid; m < a.M; m += THREADS)
+    wt[m] = a.weights[(long long)wdx * a.M + m];
+  __syncthreads();
+  float c_acc = 0.f;
+  for (int i = tid; i < n; i += THREADS) {
+    const float* xrow = a.x + (off + i) * a.D;
+    const int yi = (int)a.y[off + i];
+    float votes[64];
+    for (int o = 0; o < a.O; ++o) votes[o] = 0.f;
+    for (int m = 0; m < a.M; ++m) {
+      const float wm = wt[m];
+      if (wm == 0.f) continue;
+      const float* msk = a.masks ? a.masks + (long long)m * a.D : nullptr;
+      float logits[64];
+      mlp_logits_for(a.params + (long long)m * a.P, xrow, msk,
+                     a.D, a.H, a.O, a.kind, logits);
+      if (a.mode == VOTE_HARD) {
+        int best = 0;
+        float bv = logits[0];
+        for (int o = 1; o < a.O; ++o)
+          if (logits[o] > bv) { bv = logits[o]; best = o; }
+        votes[best] += wm;
+      } else {
+        float zmax = -1e30f;
+        for (int o = 0; o < a.O; ++o) zmax = fmaxf(zmax, logits[o]);
+        float zsum = 0.f;
+        for (int o = 0; o < a.O; ++o) {
+          logits[o] = __expf(logits[o] - zmax);
+          zsum += logits[o];
+        }
+        for (int o = 0; o < a.O; ++o) votes[o] += wm * logits[o] / zsum;
+      }
+    }
+    int best = 0;
+    float bv = votes[0];
+    for (int o = 1; o < a.O; ++o)
+      if (votes[o] > bv) { bv = votes[o]; best = o; }
+    c_acc += (best == yi) ? 1.f : 0.f;
+  }
+  const float cs = block_reduce_sum(c_acc, red);
+  if (tid == 0) {
+    atomicAdd(&a.correct[tsk], (double)cs);
+    atomicAdd(&a.total[tsk], (double)n);
+  }
+}
+
+extern "C" __global__ __launch_bounds__(THREADS)
+void mlp_confusion_kernel(VoteArgs a) {
+  const int wdx = blockIdx.x;
+  const int n = (int)a.len[wdx];
+  const long long tsk = a.task_id[wdx];
+  const long long off = a.off[wdx];
+  const long long row = a.task_row[wdx];
+  const int tid = threadIdx.x;
+  const float* w = a.params + row * a.P;
+  const float* msk = a.masks
+      ? a.masks + (a.mask_bcast ? 0 : (long long)wdx * a.D) : nullptr;
+  for (int i = tid; i < n; i += THREADS) {
+    const float* xrow = a.x + (off + i) * a.D;
+    const int yi = (int)a.y[off + i];
+    float logits[64];
+    mlp_logits_for(w, xrow, msk, a.D, a.H, a.O, a.kind, logits);
+    int best = 0;
+    float bv = logits[0];
+    for (int o = 1; o < a.O; ++o)
+      if (logits[o] > bv) { bv = logits[o]; best = o; }
+    atomicAdd(&a.conf[(tsk * a.O + yi) * a.O + best], 1.0);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
 
@@ -929,10 +1069,82 @@ void apply_aggregate_hip(torch::Tensor global_params, torch::Tensor partial,
 
 void register_cnn(pybind11::module_& mod);  // cnn_kernels.hip
 
+torch::Tensor ens_vote_multi_hip(
+    torch::Tensor params, torch::Tensor weights, torch::Tensor x,
+    torch::Tensor y, torch::Tensor task_id, torch::Tensor off,
+    torch::Tensor len, int64_t n_tasks, int64_t D, int64_t H, int64_t O,
+    int64_t kind, int64_t mode, c10::optional<torch::Tensor> masks) {
+  const int W = task_id.size(0);
+  auto optd = torch::TensorOptions().dtype(torch::kFloat64)
+                  .device(params.device());
+  auto out = torch::zeros({2, n_tasks}, optd);
+  if (W == 0) return out;
+  VoteArgs a;
+  a.params = params.data_ptr<float>();
+  a.weights = weights.data_ptr<float>();
+  a.task_id = task_id.data_ptr<int64_t>();
+  a.off = off.data_ptr<int64_t>();
+  a.len = len.data_ptr<int64_t>();
+  a.x = x.data_ptr<float>();
+  a.y = y.data_ptr<int64_t>();
+  a.masks = masks.has_value() ? masks->data_ptr<float>() : nullptr;
+  double* base = out.data_ptr<double>();
+  a.correct = base;
+  a.total = base + n_tasks;
+  a.conf = nullptr;
+  a.task_row = nullptr;
+  a.M = (int)params.size(0);
+  a.D = (int)D; a.H = (int)H; a.O = (int)O; a.P = (int)params.size(1);
+  a.kind = (int)kind; a.mode = (int)mode; a.mask_bcast = 0;
+  const size_t lds = (size_t)(a.M + 8) * sizeof(float);
+  hipLaunchKernelGGL(mlp_vote_kernel, dim3(W), dim3(THREADS), lds,
+                     c10::hip::getCurrentHIPStream(), a);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "mlp_vote_kernel launch");
+  return out;
+}
+
+torch::Tensor confusion_tasks_hip(
+    torch::Tensor params, torch::Tensor x, torch::Tensor y,
+    torch::Tensor task_row, torch::Tensor task_id, torch::Tensor off,
+    torch::Tensor len, int64_t n_tasks, int64_t n_classes, int64_t D,
+    int64_t H, int64_t O, int64_t kind,
+    c10::optional<torch::Tensor> masks) {
+  const int W = task_row.size(0);
+  auto optd = torch::TensorOptions().dtype(torch::kFloat64)
+                  .device(params.device());
+  auto out = torch::zeros({n_tasks, n_classes, n_classes}, optd);
+  if (W == 0) return out;
+  VoteArgs a;
+  a.params = params.data_ptr<float>();
+  a.weights = nullptr;
+  a.task_id = task_id.data_ptr<int64_t>();
+  a.off = off.data_ptr<int64_t>();
+  a.len = len.data_ptr<int64_t>();
+  a.x = x.data_ptr<float>();
+  a.y = y.data_ptr<int64_t>();
+  a.masks = masks.has_value() ? masks->data_ptr<float>() : nullptr;
+  a.correct = nullptr;
+  a.total = nullptr;
+  a.conf = out.data_ptr<double>();
+  a.task_row = task_row.data_ptr<int64_t>();
+  a.M = (int)params.size(0);
+  a.D = (int)D; a.H = (int)H; a.O = (int)O; a.P = (int)params.size(1);
+  a.kind = (int)kind; a.mode = 0;
+  a.mask_bcast = (masks.has_value() && masks->dim() == 1) ? 1 : 0;
+  hipLaunchKernelGGL(mlp_confusion_kernel, dim3(W), dim3(THREADS), 0,
+                     c10::hip::getCurrentHIPStream(), a);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "mlp_confusion launch");
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("train_fused", &train_fused_hip, "fused batched MLP local training");
   mod.def("eval_tasks", &eval_tasks_hip, "batched MLP accuracy/loss sweep");
   mod.def("apply_aggregate", &apply_aggregate_hip,
           "masked weighted-average model update");
+  mod.def("ens_vote_multi", &ens_vote_multi_hip,
+          "batched weighted ensemble vote (AUE/KUE)");
+  mod.def("confusion_tasks", &confusion_tasks_hip,
+          "batched per-task confusion matrices (KUE)");
   register_cnn(mod);
 }

@@ -19,7 +19,6 @@ from . import hip_loader, mlp_torch
 make_opt_state = mlp_torch.make_opt_state
 forward_logits = mlp_torch.forward_logits
 ens_vote_eval = mlp_torch.ens_vote_eval
-confusion_tasks = mlp_torch.confusion_tasks
 
 _KIND = {"fnn": 0, "lr": 1}
 
@@ -103,6 +102,56 @@ def eval_tasks(spec: MLPSpec, params: torch.Tensor,
                              task_id, win_off, win_len, n_tasks, want_mse,
                              x_mask)
     return out[0], out[1], out[2], (out[3] if want_mse else None)
+
+
+def ens_vote_multi(spec: MLPSpec, params: torch.Tensor,
+                   weights: torch.Tensor, x_arena: torch.Tensor,
+                   y_arena: torch.Tensor, task_id: torch.Tensor,
+                   win_off: torch.Tensor, win_len: torch.Tensor,
+                   n_tasks: int, mode: str = "hard",
+                   masks: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Batched weighted-vote ensemble accuracy in ONE kernel launch
+    (AUE _infer_ens / AUE-PC per-client weights / KUE soft vote —
+    FedAvgEnsAggregatorAue.py:256-283, Kue:234-264). weights [M] or
+    per-task [T, M]."""
+    if not _fits_eval(spec) or spec.o > 64:
+        return mlp_torch.ens_vote_multi(
+            spec, params, weights, x_arena, y_arena, task_id, win_off,
+            win_len, n_tasks, mode=mode, masks=masks)
+    mod = hip_loader.load()
+    W = task_id.shape[0]
+    M = params.shape[0]
+    wt = weights.to(params.device).float()
+    if wt.dim() == 1:
+        wt = wt.unsqueeze(0).expand(W, M)
+    else:
+        wt = wt[task_id]
+    return mod.ens_vote_multi(
+        params.contiguous(), wt.contiguous(), x_arena, y_arena,
+        task_id.contiguous(), win_off.contiguous(), win_len.contiguous(),
+        n_tasks, spec.d, spec.h, spec.o, _KIND[spec.kind],
+        0 if mode == "hard" else 1,
+        masks.contiguous().float() if masks is not None else None)
+
+
+def confusion_tasks(spec: MLPSpec, params: torch.Tensor,
+                    x_arena: torch.Tensor, y_arena: torch.Tensor,
+                    task_row: torch.Tensor, task_id: torch.Tensor,
+                    win_off: torch.Tensor, win_len: torch.Tensor,
+                    n_tasks: int, n_classes: int,
+                    x_mask: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Per-task confusion matrices in ONE kernel launch (KUE kappa,
+    FedAvgEnsAggregatorKue.py:266-303)."""
+    if not _fits_eval(spec) or n_classes != spec.o or spec.o > 64:
+        return mlp_torch.confusion_tasks(
+            spec, params, x_arena, y_arena, task_row, task_id, win_off,
+            win_len, n_tasks, n_classes, x_mask=x_mask)
+    mod = hip_loader.load()
+    return mod.confusion_tasks(
+        params.contiguous(), x_arena, y_arena, task_row.contiguous(),
+        task_id.contiguous(), win_off.contiguous(), win_len.contiguous(),
+        n_tasks, n_classes, spec.d, spec.h, spec.o, _KIND[spec.kind],
+        x_mask.contiguous().float() if x_mask is not None else None)
 
 
 def eval_tasks_stacked(spec: MLPSpec, params: torch.Tensor,

@@ -207,3 +207,80 @@ def test_hip_fused_broadcast_and_partial():
     torch.cuda.synchronize()
     assert torch.equal(glob3[1], glob[1])
     assert (glob3[0] - want[0]).abs().max().item() < 3e-4
+
+
+@requires_gpu
+@pytest.mark.parametrize("mode", ["hard", "soft"])
+@pytest.mark.parametrize("per_task", [False, True])
+def test_hip_vote_matches_torch(mode, per_task):
+    from feddrift_amd.ops import mlp_hip
+    torch.manual_seed(5)
+    spec = spec_for("fnn", 4, 3)
+    n, M, T, W = 1500, 4, 6, 40
+    x = torch.rand(n, 4) * 4
+    y = torch.randint(0, 3, (n,))
+    params = torch.randn(M, spec.n_params) * 0.5
+    task_id = torch.randint(0, T, (W,))
+    off = torch.randint(0, n - 130, (W,))
+    ln = torch.randint(1, 128, (W,))
+    weights = (torch.rand(T, M) if per_task else torch.rand(M))
+    weights[..., 1] = 0.0   # inactive model skipped
+    ref = mlp_torch.ens_vote_multi(spec, params, weights, x, y, task_id,
+                                   off, ln, T, mode=mode)
+    dev = torch.device("cuda:0")
+    got = mlp_hip.ens_vote_multi(
+        spec, params.to(dev), weights.to(dev), x.to(dev), y.to(dev),
+        task_id.to(dev), off.to(dev), ln.to(dev), T, mode=mode)
+    torch.cuda.synchronize()
+    assert torch.equal(got[1].cpu(), ref[1])
+    # vote ties can break differently at fp ulp; require near-exact
+    assert (got[0].cpu() - ref[0]).abs().max() <= 1.0
+
+
+@requires_gpu
+def test_hip_vote_with_masks():
+    from feddrift_amd.ops import mlp_hip
+    torch.manual_seed(6)
+    spec = spec_for("lr", 5, 3)
+    n, M, T, W = 800, 3, 4, 20
+    x = torch.rand(n, 5) * 3
+    y = torch.randint(0, 3, (n,))
+    params = torch.randn(M, spec.n_params) * 0.5
+    task_id = torch.randint(0, T, (W,))
+    off = torch.randint(0, n - 80, (W,))
+    ln = torch.randint(1, 64, (W,))
+    weights = torch.rand(M)
+    masks = (torch.rand(M, 5) > 0.3).float()
+    ref = mlp_torch.ens_vote_multi(spec, params, weights, x, y, task_id,
+                                   off, ln, T, mode="soft", masks=masks)
+    dev = torch.device("cuda:0")
+    got = mlp_hip.ens_vote_multi(
+        spec, params.to(dev), weights.to(dev), x.to(dev), y.to(dev),
+        task_id.to(dev), off.to(dev), ln.to(dev), T, mode="soft",
+        masks=masks.to(dev))
+    torch.cuda.synchronize()
+    assert torch.equal(got[1].cpu(), ref[1])
+    assert (got[0].cpu() - ref[0]).abs().max() <= 1.0
+
+
+@requires_gpu
+def test_hip_confusion_matches_torch():
+    from feddrift_amd.ops import mlp_hip
+    torch.manual_seed(7)
+    spec = spec_for("fnn", 3, 2)
+    n, M, T, W = 1200, 3, 5, 30
+    x = torch.rand(n, 3) * 5
+    y = torch.randint(0, 2, (n,))
+    params = torch.randn(M, spec.n_params) * 0.5
+    task_row = torch.randint(0, M, (W,))
+    task_id = torch.randint(0, T, (W,))
+    off = torch.randint(0, n - 130, (W,))
+    ln = torch.randint(1, 128, (W,))
+    ref = mlp_torch.confusion_tasks(spec, params, x, y, task_row, task_id,
+                                    off, ln, T, 2)
+    dev = torch.device("cuda:0")
+    got = mlp_hip.confusion_tasks(
+        spec, params.to(dev), x.to(dev), y.to(dev), task_row.to(dev),
+        task_id.to(dev), off.to(dev), ln.to(dev), T, 2)
+    torch.cuda.synchronize()
+    assert torch.equal(got.cpu(), ref)
