@@ -674,12 +674,18 @@ class FLJob:
         FedAvgEnsAggregatorSoftCluster.py:64-69)."""
         proto = zoo.create_model(self.cfg.model, self.dataset.class_num,
                                  self.dataset.feature_num)
+        # the reference resets the model's TOP-LEVEL children only
+        # (FedAvgEnsAggregatorSoftCluster.py:66-69): on torchvision
+        # models that means conv1/bn1/fc reset while the Sequential
+        # block stages keep the shared init — PARTIAL divergence. Our
+        # FlatImageModel wrapper adds one level, so unwrap to the
+        # backbone first (children() on the wrapper itself reset
+        # nothing, leaving all K models bit-identical — the round-1 bug;
+        # full modules() recursion over-diverges relative to the
+        # reference and measurably hurts IFCA on the CIFAR config).
+        target = getattr(proto, "backbone", proto)
         for m in range(self.n_models):
-            # recurse modules() (not children()) so wrapped backbones
-            # (FlatImageModel -> ResNet blocks) actually reset; children()
-            # missed every conv/bn inside the wrapper and left all K models
-            # bit-identical on the image paths.
-            for layer in proto.modules():
+            for layer in target.children():
                 if hasattr(layer, "reset_parameters"):
                     layer.reset_parameters()
             self.global_params[m] = self.packer.flatten(
