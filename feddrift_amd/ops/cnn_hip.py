@@ -99,7 +99,8 @@ class CnnHipEngine:
             "wtf": f(G * 9 * 2048),
             "wtd": f(G * 9 * 2048),
             "z1part": f(G * 8 * B * NH),
-            "w2part": f(G * 9 * 8 * 2048),
+            "w2ms": min(32, max(1, 512 // max(G, 1))),
+            "w2part": f(G * min(32, max(1, 512 // max(G, 1))) * 9 * 2048),
             "b2part": f(G * B * 64),
             "grad": f(G, self.P),
         }
@@ -168,7 +169,7 @@ class CnnHipEngine:
                 opt["vmax"] if adam else None,
                 opt["t"] if adam else None,
                 opt["lr"], float(opt.get("wd", 0.0)),
-                p1, p2, seed, g_base, B, self.O)
+                p1, p2, seed, g_base, B, self.O, ws["w2ms"])
         replicas[rows] = work
 
     # -- evaluation -------------------------------------------------------

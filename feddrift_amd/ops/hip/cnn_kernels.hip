@@ -94,7 +94,7 @@ struct CnnArgs {
   float p1, p2;                      // dropout probs
   unsigned long long seed;           // per (round, epoch)
   long long g0;                      // global pair-index base (mask hash)
-  int G, B, E, e, O, P, opt;
+  int G, B, E, e, O, P, opt, w2ms;
 };
 
 #define OPT_SGD 0
@@ -190,9 +190,11 @@ void cnn_w2_reshape(CnnArgs a) {
 }
 
 // conv2 forward as implicit GEMM on f32 MFMA: per (g, b, ptile) block,
-// 64 output pixels x 64 channels, K = 9 taps x 32 ci. A tiles (x1
-// slices) and B tiles (reshaped weights) stage through LDS; z2 lands
-// CHANNELS-LAST in the zz2 buffer for the pool kernel.
+// 64 output pixels x 64 channels, K = 9 taps x 32 ci. The x1 REGION
+// feeding all 9 taps (6 rows x 26 cols x 32 ci, channels-last) stages
+// once per block — ONE barrier, then 288 MFMAs with the B operand
+// (reshaped weights, L2-resident: 74 KB per pair shared by 9*B blocks)
+// loaded straight to registers. z2 lands CHANNELS-LAST in zz2.
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_conv2_fwd_mfma(CnnArgs a) {
   const int pt = blockIdx.x % 9;           // 576 / 64
@@ -204,32 +206,40 @@ void cnn_conv2_fwd_mfma(CnnArgs a) {
   const int wv = tid >> 6;
   const int l = tid & 63;
   const int li = l & 15, lk = l >> 4;
-  __shared__ __attribute__((aligned(16))) float sA[64][C1 + 1];
-  __shared__ __attribute__((aligned(16))) float sB[C1][C2 + 1];
+  // region: x1 rows [r0, r0+6) x 26 cols, channels-last, +1 padded
+  __shared__ __attribute__((aligned(16))) float sR[6 * S1][C1 + 1];
+  const int p0 = pt * 64;
+  const int r0 = p0 / S2;
+  const float* x1 = a.x1 + (((long long)g * a.B + b) * 676
+                            + (long long)r0 * S1) * C1;
+  {
+    // last tile: x1 only has rows r0..25 — rows past it are never read
+    // by the compute (oy <= 25) but must not be FETCHED (OOB)
+    const int nrow = min(6 * S1, (S1 - r0) * S1);
+    const int r8 = tid >> 5, kk = tid & 31;
+    for (int rr = r8; rr < 6 * S1; rr += 8)
+      sR[rr][kk] = (rr < nrow) ? x1[(long long)rr * C1 + kk] : 0.f;
+  }
+  __syncthreads();
   f32x4 acc[4];
 #pragma unroll
   for (int t = 0; t < 4; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
-  const float* x1 = a.x1 + ((long long)g * a.B + b) * X1N;
-  const int r8 = tid >> 5, kk = tid & 31;
+  // this lane's A row: output pixel p = p0 + wv*16 + li
+  const int p = p0 + wv * 16 + li;
+  const int arow = (p / S2 - r0) * S1 + (p - (p / S2) * S2);
+  const float* wt = a.wtf + (long long)g * 9 * 2048;
   for (int kyx = 0; kyx < 9; ++kyx) {
     const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
-    for (int rr = r8; rr < 64; rr += 8) {
-      const int p = pt * 64 + rr;
-      const int oy = p / S2 + ky, ox = p - (p / S2) * S2 + kx;
-      sA[rr][kk] = x1[(oy * S1 + ox) * C1 + kk];
-    }
-    const float* wt = a.wtf + ((long long)g * 9 + kyx) * 2048;
-    for (int cc = r8; cc < C2; cc += 8) sB[kk][cc] = wt[kk * C2 + cc];
-    __syncthreads();
+    const int off = arow + ky * S1 + kx;
+    const float* wk = wt + kyx * 2048;
 #pragma unroll
     for (int ks = 0; ks < C1 / 4; ++ks) {
-      const float av = sA[wv * 16 + li][ks * 4 + lk];
+      const float av = sR[off][ks * 4 + lk];
 #pragma unroll
       for (int ct = 0; ct < 4; ++ct)
         acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-            av, sB[ks * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
+            av, wk[(ks * 4 + lk) * C2 + ct * 16 + li], acc[ct], 0, 0, 0);
     }
-    __syncthreads();
   }
   const float* bias = a.work + (long long)g * a.P + OFF_B2C;
   float* z2 = a.zz2 + ((long long)g * a.B + b) * Z2N;
@@ -239,8 +249,8 @@ void cnn_conv2_fwd_mfma(CnnArgs a) {
     const float bb = bias[co];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int p = pt * 64 + wv * 16 + lk * 4 + r;
-      z2[(long long)p * C2 + co] = acc[ct][r] + bb;
+      const int pp = p0 + wv * 16 + lk * 4 + r;
+      z2[(long long)pp * C2 + co] = acc[ct][r] + bb;
     }
   }
 }
@@ -594,69 +604,78 @@ void cnn_pool_bwd(CnnArgs a) {
 }
 
 // conv2 wgrad as MFMA GEMM: dW[(kyx, ci)][co] = sum_m A[m, ci] dz[m, co]
-// with m = (b, pixel). Blocks (g, kyx, msplit): 4 waves share staged
-// A/dz tiles, each owning 2 of the 8 (ci-tile, co-tile) positions;
-// per-msplit partials land in w2part and reduce deterministically.
-#define W2_KS 8
-
+// with m = (b, pixel). Block per (g, ms): for each owned 64-pixel
+// m-tile, the x1 REGION (feeds all 9 taps) and the dz tile stage ONCE;
+// each wave accumulates 18 of the 72 (kyx, ci-tile, co-tile) outputs
+// (acc stays in registers across the whole m-range). Partials land in
+// w2part[g][ms] and reduce deterministically. The m-split count is a
+// runtime knob (a.w2ms) so small fleets still fill 256 CUs.
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_conv2_wgrad_mfma(CnnArgs a) {
-  const int ms = blockIdx.x % W2_KS;
-  const int rest = blockIdx.x / W2_KS;
-  const int kyx = rest % 9;
-  const int g = rest / 9;
+  const int ms = blockIdx.x % a.w2ms;
+  const int g = blockIdx.x / a.w2ms;
   const int n = step_n(a, g);
-  float* part = a.w2part +
-      (((long long)g * 9 + kyx) * W2_KS + ms) * 2048;
+  float* part = a.w2part + ((long long)g * a.w2ms + ms) * (9 * 2048);
   const int tid = threadIdx.x;
   const int wv = tid >> 6;
   const int l = tid & 63;
   const int li = l & 15, lk = l >> 4;
-  const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
-  __shared__ __attribute__((aligned(16))) float sA[64][C1 + 1];
+  __shared__ __attribute__((aligned(16))) float sR[6 * S1][C1 + 1];
   __shared__ __attribute__((aligned(16))) float sD[64][C2 + 1];
-  f32x4 acc[2];
-  acc[0] = f32x4{0.f, 0.f, 0.f, 0.f};
-  acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+  f32x4 acc[18];
+#pragma unroll
+  for (int t = 0; t < 18; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
   const int r8 = tid >> 5, kk = tid & 31;
-  const long long mtot = (long long)n * (S2 * S2);
-  const int mtiles = (int)((mtot + 63) / 64);
-  for (int mt = ms; mt < mtiles; mt += W2_KS) {
-    // stage 64 m-rows: x1 tap slice + dz slice (channels-last)
+  const long long mtiles = ((long long)n * (S2 * S2) + 63) / 64;
+  for (long long mt = ms; mt < mtiles; mt += a.w2ms) {
+    const long long m0 = mt * 64;
+    const int b = (int)(m0 / (S2 * S2));
+    const int p0 = (int)(m0 - (long long)b * (S2 * S2));
+    // tiles are 64-aligned within a sample (576 % 64 == 0)
+    const int r0 = p0 / S2;
+    const long long gb = (long long)g * a.B + b;
+    const float* x1 = a.x1 + (gb * 676 + (long long)r0 * S1) * C1;
+    const float* dz = a.zz2 + (gb * 576 + p0) * C2;
+    const int nrow = min(6 * S1, (S1 - r0) * S1);
+    for (int rr = r8; rr < 6 * S1; rr += 8)
+      sR[rr][kk] = (rr < nrow) ? x1[(long long)rr * C1 + kk] : 0.f;
     for (int rr = r8; rr < 64; rr += 8) {
-      const long long m = (long long)mt * 64 + rr;
-      const int b = (int)(m / (S2 * S2));
-      const int p = (int)(m - (long long)b * (S2 * S2));
-      const bool ok = m < mtot;
-      const int oy = p / S2 + ky, ox = p - (p / S2) * S2 + kx;
-      const long long gb = (long long)g * a.B + b;
-      sA[rr][kk] = ok ? a.x1[(gb * 676 + oy * S1 + ox) * C1 + kk] : 0.f;
-      sD[rr][kk] = ok ? a.zz2[(gb * 576 + p) * C2 + kk] : 0.f;
-      sD[rr][kk + 32] =
-          ok ? a.zz2[(gb * 576 + p) * C2 + kk + 32] : 0.f;
+      sD[rr][kk] = dz[(long long)rr * C2 + kk];
+      sD[rr][kk + 32] = dz[(long long)rr * C2 + kk + 32];
     }
     __syncthreads();
+    // this lane's A pixel rows for the 16 k-steps: m = km*4 + lk
 #pragma unroll
-    for (int km = 0; km < 16; ++km) {
+    for (int kyx = 0; kyx < 9; ++kyx) {
+      const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
 #pragma unroll
-      for (int t = 0; t < 2; ++t) {
-        const int tile = wv * 2 + t;
-        const int rt = tile >> 2, ct = tile & 3;
-        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-            sA[km * 4 + lk][rt * 16 + li],
-            sD[km * 4 + lk][ct * 16 + li], acc[t], 0, 0, 0);
+      for (int tt = 0; tt < 8; ++tt) {
+        if ((kyx * 8 + tt) % 4 != wv) continue;
+        const int slot = (kyx * 8 + tt) / 4;
+        const int rt = tt >> 2, ct = tt & 3;
+#pragma unroll
+        for (int km = 0; km < 16; ++km) {
+          const int mr = km * 4 + lk;
+          const int q = p0 + mr;
+          const int arow = (q / S2 - r0) * S1 + (q - (q / S2) * S2)
+                           + ky * S1 + kx;
+          acc[slot] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              sR[arow][rt * 16 + li], sD[mr][ct * 16 + li], acc[slot],
+              0, 0, 0);
+        }
       }
     }
     __syncthreads();
   }
 #pragma unroll
-  for (int t = 0; t < 2; ++t) {
-    const int tile = wv * 2 + t;
-    const int rt = tile >> 2, ct = tile & 3;
+  for (int slot = 0; slot < 18; ++slot) {
+    const int kyx = (slot * 4 + wv) / 8;
+    const int tt = (slot * 4 + wv) % 8;
+    const int rt = tt >> 2, ct = tt & 3;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int ci = rt * 16 + lk * 4 + r;
-      part[ci * C2 + ct * 16 + li] = acc[t][r];
+      part[kyx * 2048 + ci * C2 + ct * 16 + li] = acc[slot][r];
     }
   }
 }
@@ -685,17 +704,19 @@ void cnn_conv2_wgrad_reduce(CnnArgs a) {
     const int kyx = r / 2048;
     const int t = r - kyx * 2048;
     const int ci = t / C2, co = t - (t / C2) * C2;
-    const float* part = a.w2part + ((long long)g * 9 + kyx) * W2_KS * 2048
-                        + t;
+    const float* part = a.w2part + (long long)g * a.w2ms * 9 * 2048
+                        + kyx * 2048 + t;
     float s = 0.f;
-#pragma unroll
-    for (int ms = 0; ms < W2_KS; ++ms) s += part[(long long)ms * 2048];
+    for (int ms = 0; ms < a.w2ms; ++ms)
+      s += part[(long long)ms * 9 * 2048];
     a.grad[(long long)g * a.P + OFF_W2C + (co * C1 + ci) * 9 + kyx] = s;
   }
 }
 
-// conv2 dgrad as MFMA GEMM: dx1[m=(b,y,x)][ci] over K = (kyx, co) with
-// zero-padded dz staging at the borders. Blocks (g, b, mtile of 64).
+// conv2 dgrad as MFMA GEMM: dx1[m=(b,y,x)][ci] over K = (kyx, co).
+// The dz REGION feeding all 9 taps stages once per block (6 rows x 28
+// cols with zero borders, channels-last) — ONE barrier, then 288 MFMAs
+// with the B operand (wtd, L2-resident) loaded straight to registers.
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_conv2_dgrad_mfma(CnnArgs a) {
   const int mtiles = 11;                  // ceil(676 / 64)
@@ -708,45 +729,48 @@ void cnn_conv2_dgrad_mfma(CnnArgs a) {
   const int wv = tid >> 6;
   const int l = tid & 63;
   const int li = l & 15, lk = l >> 4;
-  __shared__ __attribute__((aligned(16))) float sA[64][C2 + 1];
-  __shared__ __attribute__((aligned(16))) float sB[C2][C1 + 1];
+  // dz region rows [y0-2, y0+4) x cols [-2, 26), zero-padded borders
+  __shared__ __attribute__((aligned(16))) float sD[6 * 28][C2 + 1];
+  const int p0 = mt * 64;
+  const int y0 = p0 / S1;
+  const float* dz = a.zz2 + ((long long)g * a.B + b) * Z2N;
+  {
+    const int r8 = tid >> 5, kk = tid & 31;
+    for (int rr = r8; rr < 6 * 28; rr += 8) {
+      const int dzrow = rr / 28 + y0 - 2;
+      const int col = rr - (rr / 28) * 28 - 2;
+      const bool ok = dzrow >= 0 && dzrow < S2 && col >= 0 && col < S2;
+      const long long src = ((long long)dzrow * S2 + col) * C2;
+      sD[rr][kk] = ok ? dz[src + kk] : 0.f;
+      sD[rr][kk + 32] = ok ? dz[src + kk + 32] : 0.f;
+    }
+  }
+  __syncthreads();
   f32x4 acc[2];
   acc[0] = f32x4{0.f, 0.f, 0.f, 0.f};
   acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
-  const float* dz = a.zz2 + ((long long)g * a.B + b) * Z2N;
-  const int r8 = tid >> 5, kk = tid & 31;
+  const int p = p0 + wv * 16 + li;        // this lane's x1 pixel
+  const int y = p / S1, x = p - (p / S1) * S1;
+  const float* wt = a.wtd + (long long)g * 9 * 2048;
   for (int kyx = 0; kyx < 9; ++kyx) {
     const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
-    for (int rr = r8; rr < 64; rr += 8) {
-      const int pix = mt * 64 + rr;
-      const int y = pix / S1, x = pix - (pix / S1) * S1;
-      const int oy = y - ky, ox = x - kx;
-      const bool ok = pix < 676 && oy >= 0 && oy < S2 && ox >= 0
-                      && ox < S2;
-      const long long src = ((long long)oy * S2 + ox) * C2;
-      sA[rr][kk] = ok ? dz[src + kk] : 0.f;
-      sA[rr][kk + 32] = ok ? dz[src + kk + 32] : 0.f;
-    }
-    const float* wt = a.wtd + ((long long)g * 9 + kyx) * 2048;
-    for (int cc = r8; cc < C2; cc += 8)
-      sB[cc][kk] = wt[cc * C1 + kk];
-    __syncthreads();
+    const int arow = (y - ky - y0 + 2) * 28 + (x - kx + 2);
+    const float* wk = wt + kyx * 2048;
 #pragma unroll
     for (int ks = 0; ks < C2 / 4; ++ks) {
-      const float av = sA[wv * 16 + li][ks * 4 + lk];
+      const float av = sD[arow][ks * 4 + lk];
 #pragma unroll
       for (int ct = 0; ct < 2; ++ct)
         acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-            av, sB[ks * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
+            av, wk[(ks * 4 + lk) * C1 + ct * 16 + li], acc[ct], 0, 0, 0);
     }
-    __syncthreads();
   }
   float* dx1 = a.dx1 + ((long long)g * a.B + b) * X1N;
 #pragma unroll
   for (int ct = 0; ct < 2; ++ct) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int pix = mt * 64 + wv * 16 + lk * 4 + r;
+      const int pix = p0 + wv * 16 + lk * 4 + r;
       if (pix < 676)
         dx1[(long long)pix * C1 + ct * 16 + li] = acc[ct][r];
     }
@@ -1090,7 +1114,7 @@ void cnn_train_epoch_impl(
     c10::optional<torch::Tensor> m, c10::optional<torch::Tensor> v,
     c10::optional<torch::Tensor> vmax, c10::optional<torch::Tensor> t,
     torch::Tensor lr, double wd, double p1, double p2,
-    int64_t seed, int64_t g0, int64_t B, int64_t O) {
+    int64_t seed, int64_t g0, int64_t B, int64_t O, int64_t w2ms) {
   const int G = rows.size(0);
   if (G == 0) return;
   CnnArgs a;
@@ -1136,6 +1160,7 @@ void cnn_train_epoch_impl(
   a.O = (int)O;
   a.P = (int)work.size(1);
   a.opt = adam ? OPT_ADAM : OPT_SGD;
+  a.w2ms = (int)w2ms;
 
   auto s = c10::hip::getCurrentHIPStream();
   const long long GB = (long long)G * B;
@@ -1161,7 +1186,7 @@ void cnn_train_epoch_impl(
   hipLaunchKernelGGL(cnn_fc1_dgrad, dim3(G * (NF / 256)), dim3(WG), 0, s,
                      a);
   hipLaunchKernelGGL(cnn_pool_bwd, dim3((int)GB), dim3(WG), 0, s, a);
-  hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * 9 * W2_KS), dim3(WG),
+  hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * a.w2ms), dim3(WG),
                      0, s, a);
   L(cnn_conv2_wgrad_reduce, (long long)G * (9 * 2048 + C2));
   hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB * 11), dim3(WG), 0,
