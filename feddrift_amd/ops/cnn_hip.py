@@ -175,7 +175,9 @@ class CnnHipEngine:
     # -- evaluation -------------------------------------------------------
     # bound the pooled-activation staging arena: windows per sweep chunked
     # so slots * NF floats stays within budget
-    EVAL_SLOT_BUDGET = 32768
+    # slots per sweep chunk: bounds the x1e/z2e/a2e staging arenas
+    # (~280 KB per slot across the three)
+    EVAL_SLOT_BUDGET = 8192
 
     def _eval_sweep(self, params, task_row, task_id, win_off, win_len,
                     n_tasks, mode, want_mse=False, x_mask=None,
@@ -205,7 +207,12 @@ class CnnHipEngine:
             x_mask = x_mask[order]
         csum = torch.cumsum(win_len, 0)
         start = 0
-        a2e = z1e = None
+        a2e = z1e = x1e = z2e = None
+        # reshaped conv2 weights for the eval MFMA B-operand:
+        # [M, 9, 32, 64] from the [co][ci][ky][kx] flat slice (L2-hot,
+        # shared across every block of a sweep)
+        wtf_e = params[:, 320:18752].reshape(-1, 64, 32, 9) \
+            .permute(0, 3, 2, 1).contiguous()
         while start < W:
             base = csum[start - 1] if start > 0 else csum.new_zeros(())
             end_idx = int(torch.searchsorted(
@@ -221,12 +228,17 @@ class CnnHipEngine:
             if a2e is None or a2e.shape[0] < slots:
                 a2e = torch.empty(max(slots, 1), NF, device=dev)
                 z1e = torch.empty(max(slots, 1), NH, device=dev)
+                x1e = torch.empty(max(slots, 1), X1N, device=dev)
+                z2e = torch.empty(max(slots, 1), Z2N, device=dev)
             # per-slot metadata
             srow = tr.repeat_interleave(wl)
             stid = ti.repeat_interleave(wl)
+            swin = torch.arange(tr.numel(), device=dev) \
+                .repeat_interleave(wl)
             within = (torch.arange(slots, device=dev)
                       - slot.repeat_interleave(wl))
-            sy = self._y_arena[wo.repeat_interleave(wl) + within]
+            soff = wo.repeat_interleave(wl) + within
+            sy = self._y_arena[soff]
             # fc1 GEMM blocks: runs of equal row, tiled by 64 slots
             blk_row, blk_s0, blk_len = self._fc1_blocks(tr, wl, slot)
             xm = None
@@ -238,8 +250,9 @@ class CnnHipEngine:
                 outp = torch.empty(slots, self.O, device=dev)
             out = self.mod.cnn_eval(
                 params.contiguous(), tr, ti, wo, wl, slot,
-                self._x_arena, self._y_arena, a2e, z1e,
+                self._x_arena, self._y_arena, a2e, z1e, x1e, z2e, wtf_e,
                 blk_row, blk_s0, blk_len, srow, stid, sy.contiguous(),
+                soff.contiguous(), swin.contiguous(),
                 xm, n_tasks, self.O, mode, want_mse, max_len, slots, outp)
             if mode == EV_DUMP:
                 dump_sink(stid, sy, outp)

@@ -905,6 +905,11 @@ struct CnnEvalArgs {
   const int64_t* __restrict__ srow;     // model row per slot
   const int64_t* __restrict__ stid;     // task id per slot
   const int64_t* __restrict__ sy;       // label per slot
+  const int64_t* __restrict__ soff;     // arena sample index per slot
+  const int64_t* __restrict__ swin;     // window index per slot (masks)
+  float* __restrict__ x1e;              // [slots, X1N] channels-last
+  float* __restrict__ z2e;              // [slots, Z2N] channels-last
+  const float* __restrict__ wtf_e;      // [M, 9, 32, 64] reshaped conv2 W
   long long n_slots;
   // mode EV_ACC
   double* __restrict__ correct;         // [T]
@@ -917,61 +922,123 @@ struct CnnEvalArgs {
   int O, P, mode;
 };
 
-// per (window, sample) block: conv1 (LDS x1) + conv2 + pool -> a2e
+// eval conv1: thread per (slot, pixel, ci) grid-stride, channels-last
+// x1 into the eval workspace (mirror of the train conv1 kernel over the
+// slot space)
 extern "C" __global__ __launch_bounds__(WG)
-void cnn_eval_conv(CnnEvalArgs a) {
-  const int w = blockIdx.x;
-  const int i = blockIdx.y;
-  if (i >= (int)a.len[w]) return;
-  const int tid = threadIdx.x;
-  __shared__ __attribute__((aligned(16))) float xin[D_IN];
-  extern __shared__ __attribute__((aligned(16))) float x1[];  // [X1N]
-  const float* wp = a.params + a.task_row[w] * (long long)a.P;
-  const float* xs = a.x + (a.off[w] + i) * D_IN;
-  const float* xm = a.x_mask
-      ? a.x_mask + (a.xm_per_task ? (long long)w * D_IN : 0) : nullptr;
-  for (int d = tid; d < D_IN; d += WG)
-    xin[d] = xm ? xs[d] * xm[d] : xs[d];
-  __syncthreads();
-  for (int e = tid; e < X1N; e += WG) {
-    const int c = e / (S1 * S1);
-    const int p = e - c * S1 * S1;
-    const int oy = p / S1, ox = p - (p / S1) * S1;
+void cnn_eval_conv1(CnnEvalArgs a) {
+  const long long total = a.n_slots * X1N;
+  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
+       q += (long long)gridDim.x * WG) {
+    const long long slot = q / X1N;
+    const int e = (int)(q - slot * X1N);
+    const int pp = e / C1;
+    const int c = e - pp * C1;
+    const int oy = pp / S1, ox = pp - (pp / S1) * S1;
+    const float* wp = a.params + a.srow[slot] * (long long)a.P;
+    const float* xs = a.x + a.soff[slot] * D_IN;
+    const float* xm = a.x_mask
+        ? a.x_mask + (a.xm_per_task ? a.swin[slot] * D_IN : 0) : nullptr;
     float z = wp[OFF_B1C + c];
 #pragma unroll
     for (int ky = 0; ky < 3; ++ky)
 #pragma unroll
-      for (int kx = 0; kx < 3; ++kx)
-        z = fmaf(xin[(oy + ky) * IN_W + ox + kx],
-                 wp[OFF_W1C + c * 9 + ky * 3 + kx], z);
-    x1[e] = z;
+      for (int kx = 0; kx < 3; ++kx) {
+        const int xi = (oy + ky) * IN_W + ox + kx;
+        float xv = xs[xi];
+        if (xm) xv *= xm[xi];
+        z = fmaf(xv, wp[OFF_W1C + c * 9 + ky * 3 + kx], z);
+      }
+    a.x1e[q] = z;
+  }
+}
+
+// eval conv2 as the region-staged MFMA GEMM (same structure as the
+// train kernel, indexed by slot with per-slot model rows)
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_eval_conv2_mfma(CnnEvalArgs a) {
+  const int pt = blockIdx.x % 9;
+  const long long slot = blockIdx.x / 9;
+  if (slot >= a.n_slots) return;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sR[6 * S1][C1 + 1];
+  const int p0 = pt * 64;
+  const int r0 = p0 / S2;
+  const float* x1 = a.x1e + slot * X1N + (long long)r0 * S1 * C1;
+  {
+    const int nrow = min(6 * S1, (S1 - r0) * S1);
+    const int r8 = tid >> 5, kk = tid & 31;
+    for (int rr = r8; rr < 6 * S1; rr += 8)
+      sR[rr][kk] = (rr < nrow) ? x1[(long long)rr * C1 + kk] : 0.f;
   }
   __syncthreads();
-  float* out = a.a2e + (a.slot[w] + i) * (long long)NF;
-  for (int e = tid; e < NF; e += WG) {
-    const int c = e / (SP * SP);
-    const int p = e - c * SP * SP;
-    const int py = p / SP, px = p - (p / SP) * SP;
-    float best = -1e30f;
+  f32x4 acc[4];
 #pragma unroll
-    for (int dy = 0; dy < 2; ++dy)
+  for (int t = 0; t < 4; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+  const int p = p0 + wv * 16 + li;
+  const int arow = (p / S2 - r0) * S1 + (p - (p / S2) * S2);
+  const long long row = a.srow[slot];
+  const float* wt = a.wtf_e + row * (9 * 2048);
+  for (int kyx = 0; kyx < 9; ++kyx) {
+    const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+    const int off = arow + ky * S1 + kx;
+    const float* wk = wt + kyx * 2048;
 #pragma unroll
-      for (int dx = 0; dx < 2; ++dx) {
-        const int oy = 2 * py + dy, ox = 2 * px + dx;
-        float z = wp[OFF_B2C + c];
-        for (int ci = 0; ci < C1; ++ci) {
-          const float* xc = x1 + ci * S1 * S1 + oy * S1 + ox;
-          const float* wc = wp + OFF_W2C + (c * C1 + ci) * 9;
+    for (int ks = 0; ks < C1 / 4; ++ks) {
+      const float av = sR[off][ks * 4 + lk];
 #pragma unroll
-          for (int ky = 0; ky < 3; ++ky)
-#pragma unroll
-            for (int kx = 0; kx < 3; ++kx)
-              z = fmaf(xc[ky * S1 + kx], wc[ky * 3 + kx], z);
-        }
-        best = fmaxf(best, z);
-      }
-    out[e] = best;
+      for (int ct = 0; ct < 4; ++ct)
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            av, wk[(ks * 4 + lk) * C2 + ct * 16 + li], acc[ct], 0, 0, 0);
+    }
   }
+  const float* bias = a.params + row * (long long)a.P + OFF_B2C;
+  float* z2 = a.z2e + slot * Z2N;
+#pragma unroll
+  for (int ct = 0; ct < 4; ++ct) {
+    const int co = ct * 16 + li;
+    const float bb = bias[co];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int pp = p0 + wv * 16 + lk * 4 + r;
+      z2[(long long)pp * C2 + co] = acc[ct][r] + bb;
+    }
+  }
+}
+
+// eval maxpool, block per slot: channels-last z2 reads coalesced via
+// LDS row staging, pooled a2e assembled in LDS and stored in one
+// coalesced pass (torch-flatten layout; no dropout, no argmax at eval)
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_eval_pool(CnnEvalArgs a) {
+  const long long slot = blockIdx.x;
+  if (slot >= a.n_slots) return;
+  const int tid = threadIdx.x;
+  __shared__ __attribute__((aligned(16))) float srow_[2 * S2 * C2];
+  __shared__ __attribute__((aligned(16))) float sa2[NF];
+  const float* z2 = a.z2e + slot * Z2N;
+  for (int py = 0; py < SP; ++py) {
+    for (int q = tid; q < 2 * S2 * C2; q += WG)
+      srow_[q] = z2[(long long)(2 * py) * S2 * C2 + q];
+    __syncthreads();
+    for (int q = tid; q < C2 * SP; q += WG) {
+      const int c = q / SP;
+      const int px = q - c * SP;
+      float best = -1e30f;
+#pragma unroll
+      for (int dy = 0; dy < 2; ++dy)
+#pragma unroll
+        for (int dx = 0; dx < 2; ++dx)
+          best = fmaxf(best, srow_[(dy * S2 + 2 * px + dx) * C2 + c]);
+      sa2[c * (SP * SP) + py * SP + px] = best;
+    }
+    __syncthreads();
+  }
+  float* out = a.a2e + slot * (long long)NF;
+  for (int e = tid; e < NF; e += WG) out[e] = sa2[e];
 }
 
 // fc1 eval as an MFMA tile GEMM: z1e[slot, h] = relu(a2e[slot, :] @
@@ -1206,8 +1273,10 @@ torch::Tensor cnn_eval(
     torch::Tensor params, torch::Tensor task_row, torch::Tensor task_id,
     torch::Tensor off, torch::Tensor len, torch::Tensor slot,
     torch::Tensor x, torch::Tensor y, torch::Tensor a2e, torch::Tensor z1e,
+    torch::Tensor x1e, torch::Tensor z2e, torch::Tensor wtf_e,
     torch::Tensor blk_row, torch::Tensor blk_s0, torch::Tensor blk_len,
     torch::Tensor srow, torch::Tensor stid, torch::Tensor sy,
+    torch::Tensor soff, torch::Tensor swin,
     c10::optional<torch::Tensor> x_mask, int64_t n_tasks, int64_t O,
     int64_t mode, bool want_mse, int64_t max_len, int64_t n_slots,
     c10::optional<torch::Tensor> outp) {
@@ -1242,6 +1311,11 @@ torch::Tensor cnn_eval(
   a.srow = srow.data_ptr<int64_t>();
   a.stid = stid.data_ptr<int64_t>();
   a.sy = sy.data_ptr<int64_t>();
+  a.soff = soff.data_ptr<int64_t>();
+  a.swin = swin.data_ptr<int64_t>();
+  a.x1e = x1e.data_ptr<float>();
+  a.z2e = z2e.data_ptr<float>();
+  a.wtf_e = wtf_e.data_ptr<float>();
   a.n_slots = (long long)n_slots;
   double* base = (mode == EV_ACC) ? out.data_ptr<double>() : nullptr;
   a.correct = base;
@@ -1255,13 +1329,17 @@ torch::Tensor cnn_eval(
   a.mode = (int)mode;
 
   auto s = c10::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(cnn_eval_conv, dim3(W, (int)max_len), dim3(WG),
-                     X1N * sizeof(float), s, a);
+  hipLaunchKernelGGL(cnn_eval_conv1, dim3(grid_for(n_slots * X1N)),
+                     dim3(WG), 0, s, a);
+  hipLaunchKernelGGL(cnn_eval_conv2_mfma, dim3((int)(n_slots * 9)),
+                     dim3(WG), 0, s, a);
+  hipLaunchKernelGGL(cnn_eval_pool, dim3((int)n_slots), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_eval_fc1_mfma, dim3((int)blk_row.size(0)),
                      dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_eval_head, dim3(grid_for(n_slots)), dim3(WG), 0,
                      s, a);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "cnn_eval launch");
+  (void)max_len;
   return out;
 }
 
