@@ -228,6 +228,7 @@ void cnn_conv2_fwd_mfma(CnnArgs a) {
   const int p = p0 + wv * 16 + li;
   const int arow = (p / S2 - r0) * S1 + (p - (p / S2) * S2);
   const float* wt = a.wtf + (long long)g * 9 * 2048;
+#pragma unroll
   for (int kyx = 0; kyx < 9; ++kyx) {
     const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
     const int off = arow + ky * S1 + kx;
@@ -316,8 +317,8 @@ void cnn_fc1_fwd_mfma(CnnArgs a) {
   const int wv = tid >> 6;
   const int l = tid & 63;
   const int li = l & 15, lk = l >> 4;
-  __shared__ __attribute__((aligned(16))) float sA[64][EVAL_BK + 1];
-  __shared__ __attribute__((aligned(16))) float sB[EVAL_BK][NH + 1];
+  __shared__ __attribute__((aligned(16))) float sA[2][64][EVAL_BK + 1];
+  __shared__ __attribute__((aligned(16))) float sB[2][EVAL_BK][NH + 1];
   f32x4 acc[8];
 #pragma unroll
   for (int t = 0; t < 8; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
@@ -326,22 +327,39 @@ void cnn_fc1_fwd_mfma(CnnArgs a) {
   const int mlen = min(64, n - mt * 64);
   const int r8 = tid >> 5, kk = tid & 31;
   const int k_lo = ks * (NF / FC1_KS), k_hi = (ks + 1) * (NF / FC1_KS);
+  float ra[8], rb[16];
+#define FC_LOAD(k0)                                                     \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j) {                        \
+    const int rr = r8 + j * 8;                                           \
+    ra[j] = (rr < mlen) ? a2[(long long)rr * NF + (k0) + kk] : 0.f;      \
+  }                                                                      \
+  _Pragma("unroll") for (int j = 0; j < 16; ++j)                         \
+    rb[j] = wp[(long long)(r8 + j * 8) * NF + (k0) + kk];
+#define FC_WRITE(buf)                                                   \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j)                          \
+    sA[buf][r8 + j * 8][kk] = ra[j];                                     \
+  _Pragma("unroll") for (int j = 0; j < 16; ++j)                         \
+    sB[buf][kk][r8 + j * 8] = rb[j];
+  FC_LOAD(k_lo);
+  FC_WRITE(0);
+  __syncthreads();
+  int cur = 0;
   for (int k0 = k_lo; k0 < k_hi; k0 += EVAL_BK) {
-    for (int rr = r8; rr < 64; rr += 8)
-      sA[rr][kk] = (rr < mlen) ? a2[(long long)rr * NF + k0 + kk] : 0.f;
-    for (int hh = r8; hh < NH; hh += 8)
-      sB[kk][hh] = wp[(long long)hh * NF + k0 + kk];
-    __syncthreads();
+    if (k0 + EVAL_BK < k_hi) { FC_LOAD(k0 + EVAL_BK); }
 #pragma unroll
     for (int kq = 0; kq < EVAL_BK / 4; ++kq) {
-      const float av = sA[wv * 16 + li][kq * 4 + lk];
+      const float av = sA[cur][wv * 16 + li][kq * 4 + lk];
 #pragma unroll
       for (int ct = 0; ct < 8; ++ct)
         acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-            av, sB[kq * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
+            av, sB[cur][kq * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
     }
+    if (k0 + EVAL_BK < k_hi) { FC_WRITE(cur ^ 1); }
     __syncthreads();
+    cur ^= 1;
   }
+#undef FC_LOAD
+#undef FC_WRITE
 #pragma unroll
   for (int ct = 0; ct < 8; ++ct) {
 #pragma unroll
@@ -752,6 +770,7 @@ void cnn_conv2_dgrad_mfma(CnnArgs a) {
   const int p = p0 + wv * 16 + li;        // this lane's x1 pixel
   const int y = p / S1, x = p - (p / S1) * S1;
   const float* wt = a.wtd + (long long)g * 9 * 2048;
+#pragma unroll
   for (int kyx = 0; kyx < 9; ++kyx) {
     const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
     const int arow = (y - ky - y0 + 2) * 28 + (x - kx + 2);
@@ -922,34 +941,36 @@ struct CnnEvalArgs {
   int O, P, mode;
 };
 
-// eval conv1: thread per (slot, pixel, ci) grid-stride, channels-last
-// x1 into the eval workspace (mirror of the train conv1 kernel over the
-// slot space)
+// eval conv1: block per slot; the masked input stages in LDS once and
+// each thread produces channels-last x1 elements (9 fused MACs each)
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_eval_conv1(CnnEvalArgs a) {
-  const long long total = a.n_slots * X1N;
-  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
-       q += (long long)gridDim.x * WG) {
-    const long long slot = q / X1N;
-    const int e = (int)(q - slot * X1N);
+  const long long slot = blockIdx.x;
+  if (slot >= a.n_slots) return;
+  const int tid = threadIdx.x;
+  __shared__ __attribute__((aligned(16))) float xin[D_IN];
+  __shared__ __attribute__((aligned(16))) float wc[288 + C1];
+  const float* wp = a.params + a.srow[slot] * (long long)a.P;
+  const float* xs = a.x + a.soff[slot] * D_IN;
+  const float* xm = a.x_mask
+      ? a.x_mask + (a.xm_per_task ? a.swin[slot] * D_IN : 0) : nullptr;
+  for (int d = tid; d < D_IN; d += WG)
+    xin[d] = xm ? xs[d] * xm[d] : xs[d];
+  for (int i = tid; i < 288 + C1; i += WG) wc[i] = wp[OFF_W1C + i];
+  __syncthreads();
+  float* out = a.x1e + slot * X1N;
+  for (int e = tid; e < X1N; e += WG) {
     const int pp = e / C1;
     const int c = e - pp * C1;
     const int oy = pp / S1, ox = pp - (pp / S1) * S1;
-    const float* wp = a.params + a.srow[slot] * (long long)a.P;
-    const float* xs = a.x + a.soff[slot] * D_IN;
-    const float* xm = a.x_mask
-        ? a.x_mask + (a.xm_per_task ? a.swin[slot] * D_IN : 0) : nullptr;
-    float z = wp[OFF_B1C + c];
+    float z = wc[288 + c];
 #pragma unroll
     for (int ky = 0; ky < 3; ++ky)
 #pragma unroll
-      for (int kx = 0; kx < 3; ++kx) {
-        const int xi = (oy + ky) * IN_W + ox + kx;
-        float xv = xs[xi];
-        if (xm) xv *= xm[xi];
-        z = fmaf(xv, wp[OFF_W1C + c * 9 + ky * 3 + kx], z);
-      }
-    a.x1e[q] = z;
+      for (int kx = 0; kx < 3; ++kx)
+        z = fmaf(xin[(oy + ky) * IN_W + ox + kx],
+                 wc[c * 9 + ky * 3 + kx], z);
+    out[e] = z;
   }
 }
 
@@ -982,6 +1003,7 @@ void cnn_eval_conv2_mfma(CnnEvalArgs a) {
   const int arow = (p / S2 - r0) * S1 + (p - (p / S2) * S2);
   const long long row = a.srow[slot];
   const float* wt = a.wtf_e + row * (9 * 2048);
+#pragma unroll
   for (int kyx = 0; kyx < 9; ++kyx) {
     const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
     const int off = arow + ky * S1 + kx;
@@ -1056,38 +1078,50 @@ void cnn_eval_fc1_mfma(CnnEvalArgs a) {
   const int tid = threadIdx.x;
   const int wv = tid >> 6;
   const int l = tid & 63;
-  const int li = l & 15;        // fragment row/col index
-  const int lk = l >> 4;        // fragment k index (0..3)
-
-  __shared__ __attribute__((aligned(16))) float sA[64][EVAL_BK + 1];
-  __shared__ __attribute__((aligned(16))) float sB[EVAL_BK][NH + 1];
-
+  const int li = l & 15;
+  const int lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sA[2][64][EVAL_BK + 1];
+  __shared__ __attribute__((aligned(16))) float sB[2][EVAL_BK][NH + 1];
   f32x4 acc[8];
 #pragma unroll
   for (int t = 0; t < 8; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
-
   const float* wp = a.params + row * (long long)a.P + OFF_W1F;
-  const int r8 = tid >> 5;       // 8 stager rows per pass
-  const int kk = tid & 31;
+  const int r8 = tid >> 5, kk = tid & 31;
+  // register staging: 8 A rows + 16 B rows per thread per tile
+  float ra[8], rb[16];
+#define EV_LOAD(k0)                                                     \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j) {                        \
+    const int rr = r8 + j * 8;                                           \
+    ra[j] = (rr < mlen) ? a.a2e[(s0 + rr) * (long long)NF + (k0) + kk]   \
+                        : 0.f;                                           \
+  }                                                                      \
+  _Pragma("unroll") for (int j = 0; j < 16; ++j)                         \
+    rb[j] = wp[(long long)(r8 + j * 8) * NF + (k0) + kk];
+#define EV_WRITE(buf)                                                   \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j)                          \
+    sA[buf][r8 + j * 8][kk] = ra[j];                                     \
+  _Pragma("unroll") for (int j = 0; j < 16; ++j)                         \
+    sB[buf][kk][r8 + j * 8] = rb[j];
+  EV_LOAD(0);
+  EV_WRITE(0);
+  __syncthreads();
+  int cur = 0;
   for (int k0 = 0; k0 < NF; k0 += EVAL_BK) {
-    for (int rr = r8; rr < 64; rr += 8)
-      sA[rr][kk] = (rr < mlen)
-          ? a.a2e[(s0 + rr) * (long long)NF + k0 + kk] : 0.f;
-    for (int hh = r8; hh < NH; hh += 8)
-      sB[kk][hh] = wp[(long long)hh * NF + k0 + kk];
-    __syncthreads();
+    if (k0 + EVAL_BK < NF) { EV_LOAD(k0 + EVAL_BK); }
 #pragma unroll
-    for (int ks = 0; ks < EVAL_BK / 4; ++ks) {
-      const float av = sA[wv * 16 + li][ks * 4 + lk];
+    for (int kq = 0; kq < EVAL_BK / 4; ++kq) {
+      const float av = sA[cur][wv * 16 + li][kq * 4 + lk];
 #pragma unroll
-      for (int ct = 0; ct < 8; ++ct) {
-        const float bv = sB[ks * 4 + lk][ct * 16 + li];
-        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc[ct],
-                                                       0, 0, 0);
-      }
+      for (int ct = 0; ct < 8; ++ct)
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            av, sB[cur][kq * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
     }
+    if (k0 + EVAL_BK < NF) { EV_WRITE(cur ^ 1); }
     __syncthreads();
+    cur ^= 1;
   }
+#undef EV_LOAD
+#undef EV_WRITE
   // epilogue: bias + relu -> z1e[slot, h]
   const float* bias = a.params + row * (long long)a.P + OFF_B1F;
 #pragma unroll
@@ -1329,8 +1363,8 @@ torch::Tensor cnn_eval(
   a.mode = (int)mode;
 
   auto s = c10::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(cnn_eval_conv1, dim3(grid_for(n_slots * X1N)),
-                     dim3(WG), 0, s, a);
+  hipLaunchKernelGGL(cnn_eval_conv1, dim3((int)n_slots), dim3(WG), 0, s,
+                     a);
   hipLaunchKernelGGL(cnn_eval_conv2_mfma, dim3((int)(n_slots * 9)),
                      dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_eval_pool, dim3((int)n_slots), dim3(WG), 0, s, a);
