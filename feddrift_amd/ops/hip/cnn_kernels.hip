@@ -228,7 +228,6 @@ void cnn_conv2_fwd_mfma(CnnArgs a) {
   const int p = p0 + wv * 16 + li;
   const int arow = (p / S2 - r0) * S1 + (p - (p / S2) * S2);
   const float* wt = a.wtf + (long long)g * 9 * 2048;
-#pragma unroll
   for (int kyx = 0; kyx < 9; ++kyx) {
     const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
     const int off = arow + ky * S1 + kx;
@@ -770,7 +769,6 @@ void cnn_conv2_dgrad_mfma(CnnArgs a) {
   const int p = p0 + wv * 16 + li;        // this lane's x1 pixel
   const int y = p / S1, x = p - (p / S1) * S1;
   const float* wt = a.wtd + (long long)g * 9 * 2048;
-#pragma unroll
   for (int kyx = 0; kyx < 9; ++kyx) {
     const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
     const int arow = (y - ky - y0 + 2) * 28 + (x - kx + 2);
@@ -1003,7 +1001,6 @@ void cnn_eval_conv2_mfma(CnnEvalArgs a) {
   const int arow = (p / S2 - r0) * S1 + (p - (p / S2) * S2);
   const long long row = a.srow[slot];
   const float* wt = a.wtf_e + row * (9 * 2048);
-#pragma unroll
   for (int kyx = 0; kyx < 9; ++kyx) {
     const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
     const int off = arow + ky * S1 + kx;
