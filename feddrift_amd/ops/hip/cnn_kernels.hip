@@ -137,36 +137,38 @@ __device__ __forceinline__ long long step_o(const CnnArgs& a, int g) {
 // forward
 // ---------------------------------------------------------------------------
 
-// conv1: one thread per (g, b, y, x, c1) output element, grid-stride.
-// x1 is CHANNELS-LAST [G, B, 26*26, 32] so the conv2 MFMA stagers read
-// contiguous 32-float ci slices per (ky, kx) tap.
+// conv1 forward, block per (g, b): the masked input and the conv1
+// weights stage in LDS once; each thread emits channels-last x1
+// elements (9 fused MACs each). x1 is CHANNELS-LAST [G, B, 676, 32] so
+// the conv2 MFMA stagers read contiguous ci slices.
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_conv1_fwd(CnnArgs a) {
-  const long long total = (long long)a.G * a.B * X1N;
-  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
-       q += (long long)gridDim.x * WG) {
-    const int g = (int)(q / ((long long)a.B * X1N));
-    const long long r = q - (long long)g * a.B * X1N;
-    const int b = (int)(r / X1N);
-    if (b >= step_n(a, g)) continue;
-    const int e = (int)(r - (long long)b * X1N);
-    const int p = e / C1;                  // pixel (channels-last)
-    const int c = e - p * C1;
-    const int oy = p / S1, ox = p - (p / S1) * S1;
-    const float* w = a.work + (long long)g * a.P;
-    const float* xs = a.x + (step_o(a, g) + b) * D_IN;
-    const float* xm = a.x_mask ? a.x_mask + (long long)g * D_IN : nullptr;
-    float z = w[OFF_B1C + c];
+  const int g = blockIdx.x / a.B;
+  const int b = blockIdx.x - g * a.B;
+  if (b >= step_n(a, g)) return;
+  const int tid = threadIdx.x;
+  __shared__ __attribute__((aligned(16))) float xin[D_IN];
+  __shared__ __attribute__((aligned(16))) float wc[288 + C1];
+  const float* w = a.work + (long long)g * a.P;
+  const float* xs = a.x + (step_o(a, g) + b) * D_IN;
+  const float* xm = a.x_mask ? a.x_mask + (long long)g * D_IN : nullptr;
+  for (int d = tid; d < D_IN; d += WG)
+    xin[d] = xm ? xs[d] * xm[d] : xs[d];
+  for (int i = tid; i < 288 + C1; i += WG) wc[i] = w[OFF_W1C + i];
+  __syncthreads();
+  float* out = a.x1 + ((long long)g * a.B + b) * X1N;
+  for (int e = tid; e < X1N; e += WG) {
+    const int pp = e / C1;
+    const int c = e - pp * C1;
+    const int oy = pp / S1, ox = pp - (pp / S1) * S1;
+    float z = wc[288 + c];
 #pragma unroll
     for (int ky = 0; ky < 3; ++ky)
 #pragma unroll
-      for (int kx = 0; kx < 3; ++kx) {
-        const int xi = (oy + ky) * IN_W + ox + kx;
-        float xv = xs[xi];
-        if (xm) xv *= xm[xi];
-        z = fmaf(xv, w[OFF_W1C + c * 9 + ky * 3 + kx], z);
-      }
-    a.x1[q] = z;
+      for (int kx = 0; kx < 3; ++kx)
+        z = fmaf(xin[(oy + ky) * IN_W + ox + kx],
+                 wc[c * 9 + ky * 3 + kx], z);
+    out[e] = z;
   }
 }
 
@@ -851,38 +853,57 @@ void cnn_conv1_wgrad_reduce(CnnArgs a) {
   }
 }
 
-// optimizer update: thread per (g, p) grid-stride; pairs with n == 0 skip
+// optimizer update: float4-vectorized grid-stride over (g, p/4) with a
+// scalar tail; pairs with n == 0 skip (reference skips the step)
+__device__ __forceinline__ void opt_one(const CnnArgs& a, int g,
+                                        long long q, long long row,
+                                        float lr_, int tnew, float bc1,
+                                        float bc2) {
+  const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
+  float wv = a.work[q];
+  const float gr0 = a.grad[q];
+  if (a.opt == OPT_SGD) {
+    a.work[q] = wv - lr_ * gr0;
+    return;
+  }
+  const long long gp = row * a.P + (q - (long long)g * a.P);
+  const float gr = gr0 + a.wd * wv;
+  const float mn = b1 * a.m[gp] + (1.f - b1) * gr;
+  const float vn = b2 * a.v[gp] + (1.f - b2) * gr * gr;
+  a.m[gp] = mn;
+  a.v[gp] = vn;
+  const float vm = fmaxf(a.vmax[gp], vn);
+  a.vmax[gp] = vm;
+  a.work[q] = wv - lr_ * (mn / bc1) / (sqrtf(vm / bc2) + eps);
+  (void)tnew;
+}
+
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_opt_step(CnnArgs a) {
-  const long long total = (long long)a.G * a.P;
-  const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
-  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
-       q += (long long)gridDim.x * WG) {
-    const int g = (int)(q / a.P);
+  const long long pv = a.P / 4;          // float4 body; tail scalar
+  const long long total = (long long)a.G * pv;
+  for (long long qv = (long long)blockIdx.x * WG + threadIdx.x;
+       qv < total; qv += (long long)gridDim.x * WG) {
+    const int g = (int)(qv / pv);
     if (step_n(a, g) == 0) continue;
-    const long long p = q - (long long)g * a.P;
     const long long row = a.rows[g];
     const float lr_ = a.lr[row];
-    float wv = a.work[q];
-    const float gr0 = a.grad[q];
-    if (a.opt == OPT_SGD) {
-      a.work[q] = wv - lr_ * gr0;
-      continue;
+    int tnew = 0;
+    float bc1 = 1.f, bc2 = 1.f;
+    if (a.opt == OPT_ADAM) {
+      tnew = a.t[row] + 1;
+      bc1 = 1.f - powf(0.9f, (float)tnew);
+      bc2 = 1.f - powf(0.999f, (float)tnew);
     }
-    const int tnew = a.t[row] + 1;  // tick kernel commits after
-    // precise powf: __powf's fast-math error lands ~1e-5 off the torch
-    // reference through the bias corrections (measured on the box)
-    const float bc1 = 1.f - powf(b1, (float)tnew);
-    const float bc2 = 1.f - powf(b2, (float)tnew);
-    const long long gp = row * a.P + p;
-    const float gr = gr0 + a.wd * wv;
-    const float mn = b1 * a.m[gp] + (1.f - b1) * gr;
-    const float vn = b2 * a.v[gp] + (1.f - b2) * gr * gr;
-    a.m[gp] = mn;
-    a.v[gp] = vn;
-    const float vm = fmaxf(a.vmax[gp], vn);
-    a.vmax[gp] = vm;
-    a.work[q] = wv - lr_ * (mn / bc1) / (sqrtf(vm / bc2) + eps);
+    const long long q0 = (long long)g * a.P + (qv - (long long)g * pv) * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      opt_one(a, g, q0 + j, row, lr_, tnew, bc1, bc2);
+    // tail elements handled by the last vector thread of each pair row
+    if (qv - (long long)g * pv == pv - 1)
+      for (long long q = (long long)g * a.P + pv * 4;
+           q < (long long)(g + 1) * a.P; ++q)
+        opt_one(a, g, q, row, lr_, tnew, bc1, bc2);
   }
 }
 
@@ -1267,7 +1288,7 @@ void cnn_train_epoch_impl(
   const int mtiles = (a.B + 63) / 64;
   // forward
   L(cnn_w2_reshape, (long long)G * C2 * C1 * 9);
-  L(cnn_conv1_fwd, GB * X1N);
+  hipLaunchKernelGGL(cnn_conv1_fwd, dim3((int)GB), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_conv2_fwd_mfma, dim3((int)GB * 9), dim3(WG), 0,
                      s, a);
   hipLaunchKernelGGL(cnn_pool_fwd, dim3((int)GB), dim3(WG), 0, s, a);
