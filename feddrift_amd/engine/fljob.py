@@ -454,8 +454,14 @@ class FLJob:
             and not robust   # clipping happens between train and aggregate
         if self._partial is None or self._partial.shape[0] != K:
             self._partial = torch.zeros(K, P + 1, device=dev)
+            self._totals = torch.zeros(K, device=dev)
+            self._partial_clean = True
         if hip:
-            self._partial.zero_()
+            # the apply kernel drains partial to zero on its way out, so
+            # steady-state rounds skip this fill entirely
+            if not getattr(self, "_partial_clean", False):
+                self._partial.zero_()
+            self._partial_clean = False
         if plan.rows.size == 0:
             self._partial_fused = hip
             if not hip and not self.is_module_path:
@@ -582,7 +588,9 @@ class FLJob:
         if plain and not self.is_module_path and \
                 self.backend is not ops.mlp_torch:
             from ..ops import mlp_hip
-            mlp_hip.apply_aggregate(self.global_params, partial, mask_t)
+            totals = mlp_hip.apply_aggregate(self.global_params, partial,
+                                             mask_t, self._totals)
+            self._partial_clean = True
             return totals
         upd = totals > 0
         if mask_t is not None:

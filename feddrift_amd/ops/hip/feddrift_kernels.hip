@@ -567,19 +567,31 @@ static bool launch_small(const TrainArgs& args, int G, hipStream_t stream) {
 
 // apply the aggregated average: models with positive total weight get
 // partial/total, others keep their parameters (skip rules:
-// FedAvgEnsAggregatorSoftCluster.py:151-169)
+// FedAvgEnsAggregatorSoftCluster.py:151-169). The kernel also drains
+// the partial buffer to zero for the next round's fused accumulation
+// (saves a separate fill launch per round) and parks the per-model
+// totals in `totals` for the host-side weight checks.
 extern "C" __global__ __launch_bounds__(THREADS)
 void apply_aggregate_kernel(float* __restrict__ global_params,
-                            const float* __restrict__ partial,
+                            float* __restrict__ partial,
                             const unsigned char* __restrict__ mask,
+                            float* __restrict__ totals,
                             int K, int P) {
   const int m = blockIdx.y;
   const float tot = partial[(int64_t)m * (P + 1) + P];
-  if (tot <= 0.f || (mask && !mask[m])) return;
-  const float inv = 1.0f / tot;
+  const bool upd = tot > 0.f && !(mask && !mask[m]);
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    totals[m] = tot;
+    partial[(int64_t)m * (P + 1) + P] = 0.f;
+  }
+  const float inv = upd ? 1.0f / tot : 0.f;
   for (int p = blockIdx.x * THREADS + threadIdx.x; p < P;
-       p += gridDim.x * THREADS)
-    global_params[(int64_t)m * P + p] = partial[(int64_t)m * (P + 1) + p] * inv;
+       p += gridDim.x * THREADS) {
+    if (upd)
+      global_params[(int64_t)m * P + p] =
+          partial[(int64_t)m * (P + 1) + p] * inv;
+    partial[(int64_t)m * (P + 1) + p] = 0.f;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1053,7 +1065,8 @@ torch::Tensor eval_tasks_hip(
 }
 
 void apply_aggregate_hip(torch::Tensor global_params, torch::Tensor partial,
-                         c10::optional<torch::Tensor> mask) {
+                         c10::optional<torch::Tensor> mask,
+                         torch::Tensor totals) {
   const int K = global_params.size(0);
   const int P = global_params.size(1);
   const int bx = std::min(64, (P + THREADS - 1) / THREADS);
@@ -1063,6 +1076,7 @@ void apply_aggregate_hip(torch::Tensor global_params, torch::Tensor partial,
                      partial.data_ptr<float>(),
                      mask.has_value() ? mask->data_ptr<unsigned char>()
                                       : nullptr,
+                     totals.data_ptr<float>(),
                      K, P);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "apply_aggregate launch");
 }

@@ -85,8 +85,18 @@ def train_fused(spec: MLPSpec, params_all: torch.Tensor, rows: torch.Tensor,
 
 
 def apply_aggregate(global_params: torch.Tensor, partial: torch.Tensor,
-                    mask: Optional[torch.Tensor] = None) -> None:
-    hip_loader.load().apply_aggregate(global_params, partial, mask)
+                    mask: Optional[torch.Tensor] = None,
+                    totals_out: Optional[torch.Tensor] = None
+                    ) -> torch.Tensor:
+    """Masked weighted-average update; ALSO zero-drains `partial` for the
+    next round's fused accumulation and parks the per-model totals in
+    totals_out (allocated if not given). Returns totals_out."""
+    if totals_out is None:
+        totals_out = torch.empty(global_params.shape[0],
+                                 device=global_params.device)
+    hip_loader.load().apply_aggregate(global_params, partial, mask,
+                                      totals_out)
+    return totals_out
 
 
 def eval_tasks(spec: MLPSpec, params: torch.Tensor,
