@@ -193,20 +193,28 @@ def test_hip_fused_broadcast_and_partial():
     assert (reps - reps_ref).abs().max().item() < 5e-5
     assert (partial - part_ref).abs().max().item() < 3e-4
 
-    # apply: average models, verify against manual
+    # apply: average models, verify against manual; the kernel DRAINS
+    # partial to zero on its way out (next round's fused accumulation
+    # starts clean without a separate fill) and parks totals
     glob2 = glob.clone()
-    mlp_hip.apply_aggregate(glob2, partial, None)
+    part2 = partial.clone()
+    totals = mlp_hip.apply_aggregate(glob2, part2, None)
     torch.cuda.synchronize()
     want = part_ref[:, :P] / part_ref[:, P:P + 1]
     assert (glob2 - want).abs().max().item() < 3e-4
+    assert (totals - part_ref[:, P]).abs().max().item() < 1e-4
+    assert part2.abs().max().item() == 0.0, "partial must be drained"
 
-    # masked apply keeps masked-out rows
+    # masked apply keeps masked-out rows (fresh partial: the buffer is
+    # consumed by each apply)
     glob3 = glob.clone()
+    part3 = partial.clone()
     mask = torch.tensor([1, 0, 1], dtype=torch.uint8, device=dev)
-    mlp_hip.apply_aggregate(glob3, partial, mask)
+    mlp_hip.apply_aggregate(glob3, part3, mask)
     torch.cuda.synchronize()
     assert torch.equal(glob3[1], glob[1])
     assert (glob3[0] - want[0]).abs().max().item() < 3e-4
+    assert part3.abs().max().item() == 0.0
 
 
 @requires_gpu
