@@ -577,16 +577,18 @@ void apply_aggregate_kernel(float* __restrict__ global_params,
                             const unsigned char* __restrict__ mask,
                             float* __restrict__ totals,
                             int K, int P) {
-  const int m = blockIdx.y;
+  // ONE block per model: every thread reads the weight total before
+  // anything zeroes it (a multi-block grid would race the drain of the
+  // [m, P] slot against sibling blocks' read of it)
+  const int m = blockIdx.x;
   const float tot = partial[(int64_t)m * (P + 1) + P];
   const bool upd = tot > 0.f && !(mask && !mask[m]);
-  if (blockIdx.x == 0 && threadIdx.x == 0) {
+  if (threadIdx.x == 0) {
     totals[m] = tot;
     partial[(int64_t)m * (P + 1) + P] = 0.f;
   }
   const float inv = upd ? 1.0f / tot : 0.f;
-  for (int p = blockIdx.x * THREADS + threadIdx.x; p < P;
-       p += gridDim.x * THREADS) {
+  for (int p = threadIdx.x; p < P; p += THREADS) {
     if (upd)
       global_params[(int64_t)m * P + p] =
           partial[(int64_t)m * (P + 1) + p] * inv;
@@ -1069,8 +1071,7 @@ void apply_aggregate_hip(torch::Tensor global_params, torch::Tensor partial,
                          torch::Tensor totals) {
   const int K = global_params.size(0);
   const int P = global_params.size(1);
-  const int bx = std::min(64, (P + THREADS - 1) / THREADS);
-  hipLaunchKernelGGL(apply_aggregate_kernel, dim3(bx, K), dim3(THREADS), 0,
+  hipLaunchKernelGGL(apply_aggregate_kernel, dim3(K), dim3(THREADS), 0,
                      c10::hip::getCurrentHIPStream(),
                      global_params.data_ptr<float>(),
                      partial.data_ptr<float>(),
