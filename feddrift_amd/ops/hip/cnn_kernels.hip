@@ -663,10 +663,17 @@ void cnn_conv2_wgrad_mfma(CnnArgs a) {
       sD[rr][kk + 32] = dz[(long long)rr * C2 + kk + 32];
     }
     __syncthreads();
-    // this lane's A pixel rows for the 16 k-steps: m = km*4 + lk
+    // this lane's A pixel rows for the 16 k-steps (m = km*4 + lk),
+    // hoisted out of the tap loops (one div per k-step, not 18)
+    int parow[16];
+#pragma unroll
+    for (int km = 0; km < 16; ++km) {
+      const int q = p0 + km * 4 + lk;
+      parow[km] = (q / S2 - r0) * S1 + (q - (q / S2) * S2);
+    }
 #pragma unroll
     for (int kyx = 0; kyx < 9; ++kyx) {
-      const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+      const int kyoff = (kyx / 3) * S1 + (kyx - (kyx / 3) * 3);
 #pragma unroll
       for (int tt = 0; tt < 8; ++tt) {
         if ((kyx * 8 + tt) % 4 != wv) continue;
@@ -674,13 +681,9 @@ void cnn_conv2_wgrad_mfma(CnnArgs a) {
         const int rt = tt >> 2, ct = tt & 3;
 #pragma unroll
         for (int km = 0; km < 16; ++km) {
-          const int mr = km * 4 + lk;
-          const int q = p0 + mr;
-          const int arow = (q / S2 - r0) * S1 + (q - (q / S2) * S2)
-                           + ky * S1 + kx;
           acc[slot] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-              sR[arow][rt * 16 + li], sD[mr][ct * 16 + li], acc[slot],
-              0, 0, 0);
+              sR[parow[km] + kyoff][rt * 16 + li],
+              sD[km * 4 + lk][ct * 16 + li], acc[slot], 0, 0, 0);
         }
       }
     }

@@ -320,3 +320,15 @@ def test_world2_secure_agg_matches_plain(tmp_path):
         os.environ.pop("FD_SECURE", None)
     assert abs(r_plain[0] - r_sec[0]) < 5e-3, (r_plain, r_sec)
     assert np.allclose(r_plain, r_sec, atol=0.02), (r_plain, r_sec)
+
+
+def test_world8_with_idle_ranks(tmp_path):
+    """World 8 over 6 clients: two ranks own NO clients — their
+    zero-contribution collectives, empty plans and empty eval shards
+    must still agree with the single-process run (the driver's 8-GPU
+    scaling bench has exactly this shape when clients < ranks)."""
+    data = _write_data(tmp_path)
+    r1 = _run(1, data, str(tmp_path / "w1"), 29625)
+    r8 = _run(8, data, str(tmp_path / "w8"), 29626)
+    assert abs(r1[0] - r8[0]) < 1e-6, (r1, r8)
+    assert np.allclose(r1, r8, atol=0.02), (r1, r8)
