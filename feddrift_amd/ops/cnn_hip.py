@@ -51,8 +51,9 @@ class CnnHipEngine:
 
     # workspace budget (bytes) for the per-epoch activation/grad arena;
     # pairs are chunked so the arena fits (chunks are numerically exact:
-    # pairs are independent)
-    WS_BUDGET = 4 << 30
+    # pairs are independent). Sized for 288 GB of HBM3E: bigger chunks =
+    # fewer launch sequences per round at FEMNIST-scale fleets.
+    WS_BUDGET = 24 << 30
 
     def __init__(self, template: torch.nn.Module, packer, device):
         assert is_cnn_dropout(template)
@@ -110,7 +111,7 @@ class CnnHipEngine:
     def _chunk_pairs(self, B: int) -> int:
         per_pair = (2 * X1N + 3 * NF + Z2N + 2 * NH + 64) * B * 4 \
             + NF * B + self.P * 4
-        return max(1, min(256, int(self.WS_BUDGET // max(1, per_pair))))
+        return max(1, min(2048, int(self.WS_BUDGET // max(1, per_pair))))
 
     def train(self, global_params: torch.Tensor, replicas: torch.Tensor,
               plan, opt: Dict, x_arena: torch.Tensor, y_arena: torch.Tensor,
