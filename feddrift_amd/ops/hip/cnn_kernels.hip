@@ -490,9 +490,11 @@ void cnn_fc2_dgrad(CnnArgs a) {
   }
 }
 
-// fc1 wgrad: block per (g, jtile of 128 columns). dz1 reads come from
-// global (L2-hot, ~50 KB per pair); a2 column tiles stage through LDS in
-// batch chunks; exclusive-owner writes, no atomics.
+// fc1 wgrad as MFMA GEMM: dW1f[h][j] = sum_b dz1[b][h] a2[b][j] —
+// block per (g, jtile of 128 columns), 64 (h-tile, j-tile) outputs
+// split over 4 waves, K = batch staged in 16-sample chunks (both
+// operands come in [b][.] layout, so the A fragment reads the staged
+// dz1 tile transposed in place). Exclusive-owner writes.
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_fc1_wgrad(CnnArgs a) {
   const int jt = blockIdx.x % (NF / 128);
@@ -500,34 +502,49 @@ void cnn_fc1_wgrad(CnnArgs a) {
   const int n = step_n(a, g);
   if (n == 0) return;
   const int tid = threadIdx.x;
-  __shared__ __attribute__((aligned(16))) float sa2[96][129];
-  // thread owns (h0.., j): j fixed per thread, 64 h's
-  const int j = tid & 127;
-  const int h0 = (tid >> 7) * 64;       // 0 or 64
-  float acc[64];
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sDZ[16][NH + 1];
+  __shared__ __attribute__((aligned(16))) float sA2[16][129];
+  f32x4 acc[16];
 #pragma unroll
-  for (int t = 0; t < 64; ++t) acc[t] = 0.f;
-  for (int b0 = 0; b0 < n; b0 += 96) {
-    const int bc = min(96, n - b0);
-    for (int q = tid; q < bc * 128; q += WG) {
-      const int bb = q >> 7;
-      sa2[bb][q & 127] =
-          a.a2[((long long)g * a.B + b0 + bb) * NF + jt * 128 + (q & 127)];
+  for (int t = 0; t < 16; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+  const float* dz1 = a.dz1 + (long long)g * a.B * NH;
+  const float* a2 = a.a2 + (long long)g * a.B * NF + jt * 128;
+  const int j = tid & 127;
+  const int b2 = tid >> 7;                // 2 stager rows per pass
+  for (int b0 = 0; b0 < n; b0 += 16) {
+    for (int bb = b2; bb < 16; bb += 2) {
+      const bool ok = b0 + bb < n;
+      sDZ[bb][j] = ok ? dz1[(long long)(b0 + bb) * NH + j] : 0.f;
+      sA2[bb][j] = ok ? a2[(long long)(b0 + bb) * NF + j] : 0.f;
     }
     __syncthreads();
-    const float* d1 = a.dz1 + ((long long)g * a.B + b0) * NH;
-    for (int bb = 0; bb < bc; ++bb) {
-      const float av = sa2[bb][j];
-      const float* dr = d1 + (long long)bb * NH + h0;
 #pragma unroll
-      for (int t = 0; t < 64; ++t) acc[t] = fmaf(dr[t], av, acc[t]);
+    for (int ks = 0; ks < 4; ++ks) {
+#pragma unroll
+      for (int t = 0; t < 16; ++t) {
+        const int tile = wv * 16 + t;
+        const int rt = tile >> 3, ct = tile & 7;
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            sDZ[ks * 4 + lk][rt * 16 + li],
+            sA2[ks * 4 + lk][ct * 16 + li], acc[t], 0, 0, 0);
+      }
     }
     __syncthreads();
   }
   float* gr = a.grad + (long long)g * a.P + OFF_W1F;
 #pragma unroll
-  for (int t = 0; t < 64; ++t)
-    gr[(long long)(h0 + t) * NF + jt * 128 + j] = acc[t];
+  for (int t = 0; t < 16; ++t) {
+    const int tile = wv * 16 + t;
+    const int rt = tile >> 3, ct = tile & 7;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int h = rt * 16 + lk * 4 + r;
+      gr[(long long)h * NF + jt * 128 + ct * 16 + li] = acc[t][r];
+    }
+  }
 }
 
 // fc1 bias grad: thread per (g, h)
@@ -547,36 +564,57 @@ void cnn_fc1_bias_grad(CnnArgs a) {
   }
 }
 
-// fc1 dgrad: block per (g, jtile of 256); dz1 tile staged in LDS, W1f
-// columns streamed once per 16-sample tile (B/16 re-reads, L2-resident)
+// fc1 dgrad as MFMA GEMM: da2[b][j] = sum_h dz1[b][h] W1f[h][j] —
+// block per (g, b-tile of 64, jtile of 128), K = 128 in 32-deep chunks
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_fc1_dgrad(CnnArgs a) {
-  const int jt = blockIdx.x % (NF / 256);
-  const int g = blockIdx.x / (NF / 256);
+  const int jt = blockIdx.x % (NF / 128);
+  const int rest = blockIdx.x / (NF / 128);
+  const int mtiles = (a.B + 63) / 64;
+  const int mt = rest % mtiles;
+  const int g = rest / mtiles;
   const int n = step_n(a, g);
-  if (n == 0) return;
+  if (mt * 64 >= n) return;
   const int tid = threadIdx.x;
-  const int j = jt * 256 + tid;
-  __shared__ __attribute__((aligned(16))) float sd1[16][NH];
-  const float* w = a.work + (long long)g * a.P + OFF_W1F;
-  for (int b0 = 0; b0 < n; b0 += 16) {
-    const int bc = min(16, n - b0);
-    for (int q = tid; q < bc * NH; q += WG)
-      sd1[q >> 7][q & 127] =
-          a.dz1[((long long)g * a.B + b0 + (q >> 7)) * NH + (q & 127)];
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sA[64][33];
+  __shared__ __attribute__((aligned(16))) float sB[32][129];
+  f32x4 acc[8];
+#pragma unroll
+  for (int t = 0; t < 8; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+  const int mlen = min(64, n - mt * 64);
+  const float* dz1 = a.dz1 + ((long long)g * a.B + mt * 64) * NH;
+  const float* w = a.work + (long long)g * a.P + OFF_W1F + jt * 128;
+  const int r8 = tid >> 5, kk = tid & 31;
+  const int j = tid & 127, h2 = tid >> 7;
+  for (int k0 = 0; k0 < NH; k0 += 32) {
+    for (int rr = r8; rr < 64; rr += 8)
+      sA[rr][kk] = (rr < mlen)
+          ? dz1[(long long)rr * NH + k0 + kk] : 0.f;
+    for (int hh = h2; hh < 32; hh += 2)
+      sB[hh][j] = w[(long long)(k0 + hh) * NF + j];
     __syncthreads();
-    float acc[16];
 #pragma unroll
-    for (int t = 0; t < 16; ++t) acc[t] = 0.f;
-    for (int h = 0; h < NH; ++h) {
-      const float wv = w[(long long)h * NF + j];
+    for (int ks = 0; ks < 8; ++ks) {
+      const float av = sA[wv * 16 + li][ks * 4 + lk];
 #pragma unroll
-      for (int t = 0; t < 16; ++t)
-        acc[t] = fmaf(sd1[t][h], wv, acc[t]);
+      for (int ct = 0; ct < 8; ++ct)
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            av, sB[ks * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
     }
-    for (int t = 0; t < bc; ++t)
-      a.da2[((long long)g * a.B + b0 + t) * NF + j] = acc[t];
     __syncthreads();
+  }
+  float* da2 = a.da2 + ((long long)g * a.B + mt * 64) * NF + jt * 128;
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = wv * 16 + lk * 4 + r;
+      if (row < mlen)
+        da2[(long long)row * NF + ct * 16 + li] = acc[ct][r];
+    }
   }
 }
 
@@ -873,33 +911,16 @@ void cnn_conv1_wgrad_reduce(CnnArgs a) {
   }
 }
 
-// optimizer update: float4-vectorized grid-stride over (g, p/4) with a
-// scalar tail; pairs with n == 0 skip (reference skips the step)
-__device__ __forceinline__ void opt_one(const CnnArgs& a, int g,
-                                        long long q, long long row,
-                                        float lr_, int tnew, float bc1,
-                                        float bc2) {
-  const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
-  float wv = a.work[q];
-  const float gr0 = a.grad[q];
-  if (a.opt == OPT_SGD) {
-    a.work[q] = wv - lr_ * gr0;
-    return;
-  }
-  const long long gp = row * a.P + (q - (long long)g * a.P);
-  const float gr = gr0 + a.wd * wv;
-  const float mn = b1 * a.m[gp] + (1.f - b1) * gr;
-  const float vn = b2 * a.v[gp] + (1.f - b2) * gr * gr;
-  a.m[gp] = mn;
-  a.v[gp] = vn;
-  const float vm = fmaxf(a.vmax[gp], vn);
-  a.vmax[gp] = vm;
-  a.work[q] = wv - lr_ * (mn / bc1) / (sqrtf(vm / bc2) + eps);
-  (void)tnew;
-}
+// optimizer update: vector grid-stride over (g, p/4) with a scalar
+// tail. The SGD/Adam branch is hoisted OUT of the element loop and the
+// state rows use alignment-4 float4s (rows are only dword-aligned at
+// odd P), so the compiler emits merged dwordx4 traffic instead of the
+// 28 scalar loads the per-element branch produced.
+typedef float f4u __attribute__((ext_vector_type(4), aligned(4)));
 
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_opt_step(CnnArgs a) {
+  const float b1 = 0.9f, b2 = 0.999f, eps = 1e-8f;
   const long long pv = a.P / 4;          // float4 body; tail scalar
   const long long total = (long long)a.G * pv;
   for (long long qv = (long long)blockIdx.x * WG + threadIdx.x;
@@ -908,22 +929,57 @@ void cnn_opt_step(CnnArgs a) {
     if (step_n(a, g) == 0) continue;
     const long long row = a.rows[g];
     const float lr_ = a.lr[row];
-    int tnew = 0;
-    float bc1 = 1.f, bc2 = 1.f;
-    if (a.opt == OPT_ADAM) {
-      tnew = a.t[row] + 1;
-      bc1 = 1.f - powf(0.9f, (float)tnew);
-      bc2 = 1.f - powf(0.999f, (float)tnew);
-    }
-    const long long q0 = (long long)g * a.P + (qv - (long long)g * pv) * 4;
+    const long long pp = (qv - (long long)g * pv) * 4;
+    const long long q0 = (long long)g * a.P + pp;
+    const bool tail = (qv - (long long)g * pv == pv - 1);
+    if (a.opt == OPT_SGD) {
+      f4u wv = *(const f4u*)(a.work + q0);
+      const f4u gr = *(const f4u*)(a.grad + q0);
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
-      opt_one(a, g, q0 + j, row, lr_, tnew, bc1, bc2);
-    // tail elements handled by the last vector thread of each pair row
-    if (qv - (long long)g * pv == pv - 1)
-      for (long long q = (long long)g * a.P + pv * 4;
-           q < (long long)(g + 1) * a.P; ++q)
-        opt_one(a, g, q, row, lr_, tnew, bc1, bc2);
+      for (int j = 0; j < 4; ++j) wv[j] -= lr_ * gr[j];
+      *(f4u*)(a.work + q0) = wv;
+      if (tail)
+        for (long long q = q0 + 4; q < (long long)(g + 1) * a.P; ++q)
+          a.work[q] -= lr_ * a.grad[q];
+      continue;
+    }
+    const int tnew = a.t[row] + 1;       // tick kernel commits after
+    const float bc1 = 1.f - powf(b1, (float)tnew);
+    const float bc2 = 1.f - powf(b2, (float)tnew);
+    const long long s0 = row * a.P + pp;
+    f4u wv = *(const f4u*)(a.work + q0);
+    const f4u gr0 = *(const f4u*)(a.grad + q0);
+    f4u mo = *(const f4u*)(a.m + s0);
+    f4u vo = *(const f4u*)(a.v + s0);
+    f4u vm = *(const f4u*)(a.vmax + s0);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float gr = gr0[j] + a.wd * wv[j];
+      const float mn = b1 * mo[j] + (1.f - b1) * gr;
+      const float vn = b2 * vo[j] + (1.f - b2) * gr * gr;
+      mo[j] = mn;
+      vo[j] = vn;
+      const float v2 = fmaxf(vm[j], vn);
+      vm[j] = v2;
+      wv[j] -= lr_ * (mn / bc1) / (sqrtf(v2 / bc2) + eps);
+    }
+    *(f4u*)(a.m + s0) = mo;
+    *(f4u*)(a.v + s0) = vo;
+    *(f4u*)(a.vmax + s0) = vm;
+    *(f4u*)(a.work + q0) = wv;
+    if (tail) {
+      for (long long q = q0 + 4; q < (long long)(g + 1) * a.P; ++q) {
+        const long long gp = row * a.P + (q - (long long)g * a.P);
+        const float gr = a.grad[q] + a.wd * a.work[q];
+        const float mn = b1 * a.m[gp] + (1.f - b1) * gr;
+        const float vn = b2 * a.v[gp] + (1.f - b2) * gr * gr;
+        a.m[gp] = mn;
+        a.v[gp] = vn;
+        const float v2 = fmaxf(a.vmax[gp], vn);
+        a.vmax[gp] = v2;
+        a.work[q] -= lr_ * (mn / bc1) / (sqrtf(v2 / bc2) + eps);
+      }
+    }
   }
 }
 
@@ -1322,8 +1378,8 @@ void cnn_train_epoch_impl(
   hipLaunchKernelGGL(cnn_fc1_wgrad, dim3(G * (NF / 128)), dim3(WG), 0, s,
                      a);
   L(cnn_fc1_bias_grad, (long long)G * NH);
-  hipLaunchKernelGGL(cnn_fc1_dgrad, dim3(G * (NF / 256)), dim3(WG), 0, s,
-                     a);
+  hipLaunchKernelGGL(cnn_fc1_dgrad, dim3(G * mtiles * (NF / 128)),
+                     dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_pool_bwd, dim3((int)GB), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * a.w2ms), dim3(WG),
                      0, s, a);
