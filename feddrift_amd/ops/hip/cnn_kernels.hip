@@ -677,32 +677,56 @@ void cnn_conv2_wgrad_mfma(CnnArgs a) {
   const int wv = tid >> 6;
   const int l = tid & 63;
   const int li = l & 15, lk = l >> 4;
-  __shared__ __attribute__((aligned(16))) float sR[6 * S1][C1 + 1];
-  __shared__ __attribute__((aligned(16))) float sD[64][C2 + 1];
+  __shared__ __attribute__((aligned(16))) float sR[2][6 * S1][C1 + 1];
+  __shared__ __attribute__((aligned(16))) float sD[2][64][C2 + 1];
   f32x4 acc[18];
 #pragma unroll
   for (int t = 0; t < 18; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
   const int r8 = tid >> 5, kk = tid & 31;
   const long long mtiles = ((long long)n * (S2 * S2) + 63) / 64;
+  // register-staged double buffering: 20 region values + 16 dz values
+  // per thread per tile; tile t+w2ms's loads issue before tile t's
+  // MFMAs, the LDS writes land after
+  float rr_[20], rd_[16];
+  int cur = 0;
+#define WG_LOAD(MT)                                                     \
+  {                                                                      \
+    const long long m0_ = (MT) * 64;                                     \
+    const int b_ = (int)(m0_ / (S2 * S2));                               \
+    const int p0_ = (int)(m0_ - (long long)b_ * (S2 * S2));              \
+    const int r0_ = p0_ / S2;                                            \
+    const long long gb_ = (long long)g * a.B + b_;                       \
+    const float* x1_ = a.x1 + (gb_ * 676 + (long long)r0_ * S1) * C1;    \
+    const float* dz_ = a.zz2 + (gb_ * 576 + p0_) * C2;                   \
+    const int nrow_ = min(6 * S1, (S1 - r0_) * S1);                      \
+    _Pragma("unroll") for (int jj = 0; jj < 20; ++jj) {                  \
+      const int rr2 = r8 + jj * 8;                                       \
+      rr_[jj] = (rr2 < nrow_)                                            \
+          ? x1_[(long long)rr2 * C1 + kk] : 0.f;                         \
+    }                                                                    \
+    _Pragma("unroll") for (int jj = 0; jj < 8; ++jj) {                   \
+      rd_[jj] = dz_[(long long)(r8 + jj * 8) * C2 + kk];                 \
+      rd_[jj + 8] = dz_[(long long)(r8 + jj * 8) * C2 + kk + 32];        \
+    }                                                                    \
+  }
+#define WG_WRITE(buf)                                                   \
+  _Pragma("unroll") for (int jj = 0; jj < 20; ++jj) {                    \
+    const int rr2 = r8 + jj * 8;                                         \
+    if (rr2 < 6 * S1) sR[buf][rr2][kk] = rr_[jj];                        \
+  }                                                                      \
+  _Pragma("unroll") for (int jj = 0; jj < 8; ++jj) {                     \
+    sD[buf][r8 + jj * 8][kk] = rd_[jj];                                  \
+    sD[buf][r8 + jj * 8][kk + 32] = rd_[jj + 8];                         \
+  }
+  if (ms < mtiles) {
+    WG_LOAD(ms);
+    WG_WRITE(0);
+  }
+  __syncthreads();
   for (long long mt = ms; mt < mtiles; mt += a.w2ms) {
-    const long long m0 = mt * 64;
-    const int b = (int)(m0 / (S2 * S2));
-    const int p0 = (int)(m0 - (long long)b * (S2 * S2));
-    // tiles are 64-aligned within a sample (576 % 64 == 0)
+    const int p0 = (int)((mt * 64) % (S2 * S2));
     const int r0 = p0 / S2;
-    const long long gb = (long long)g * a.B + b;
-    const float* x1 = a.x1 + (gb * 676 + (long long)r0 * S1) * C1;
-    const float* dz = a.zz2 + (gb * 576 + p0) * C2;
-    const int nrow = min(6 * S1, (S1 - r0) * S1);
-    for (int rr = r8; rr < 6 * S1; rr += 8)
-      sR[rr][kk] = (rr < nrow) ? x1[(long long)rr * C1 + kk] : 0.f;
-    for (int rr = r8; rr < 64; rr += 8) {
-      sD[rr][kk] = dz[(long long)rr * C2 + kk];
-      sD[rr][kk + 32] = dz[(long long)rr * C2 + kk + 32];
-    }
-    __syncthreads();
-    // this lane's A pixel rows for the 16 k-steps (m = km*4 + lk),
-    // hoisted out of the tap loops (one div per k-step, not 18)
+    if (mt + a.w2ms < mtiles) { WG_LOAD(mt + a.w2ms); }
     int parow[16];
 #pragma unroll
     for (int km = 0; km < 16; ++km) {
@@ -720,13 +744,17 @@ void cnn_conv2_wgrad_mfma(CnnArgs a) {
 #pragma unroll
         for (int km = 0; km < 16; ++km) {
           acc[slot] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-              sR[parow[km] + kyoff][rt * 16 + li],
-              sD[km * 4 + lk][ct * 16 + li], acc[slot], 0, 0, 0);
+              sR[cur][parow[km] + kyoff][rt * 16 + li],
+              sD[cur][km * 4 + lk][ct * 16 + li], acc[slot], 0, 0, 0);
         }
       }
     }
+    if (mt + a.w2ms < mtiles) { WG_WRITE(cur ^ 1); }
     __syncthreads();
+    cur ^= 1;
   }
+#undef WG_LOAD
+#undef WG_WRITE
 #pragma unroll
   for (int slot = 0; slot < 18; ++slot) {
     const int kyx = (slot * 4 + wv) / 8;
