@@ -267,14 +267,25 @@ void cnn_pool_fwd(CnnArgs a) {
   const int b = blockIdx.x - g * a.B;
   if (b >= step_n(a, g)) return;
   const int tid = threadIdx.x;
-  __shared__ __attribute__((aligned(16))) float srow[2 * S2 * C2];
+  __shared__ __attribute__((aligned(16))) float srow[2][2 * S2 * C2];
   __shared__ __attribute__((aligned(16))) float sa2[NF];
   __shared__ unsigned char spidx[NF];
   const float* z2 = a.zz2 + ((long long)g * a.B + b) * Z2N;
+  // double-buffered row-pair staging: the next pooled row's z2 loads
+  // issue before this row's max pass, land after it
+  float rst[12];
+#define PF_LOAD(py)                                                     \
+  _Pragma("unroll") for (int jj = 0; jj < 12; ++jj)                      \
+    rst[jj] = z2[(long long)(2 * (py)) * S2 * C2 + tid + jj * WG];
+#define PF_WRITE(buf)                                                   \
+  _Pragma("unroll") for (int jj = 0; jj < 12; ++jj)                      \
+    srow[buf][tid + jj * WG] = rst[jj];
+  PF_LOAD(0);
+  PF_WRITE(0);
+  __syncthreads();
+  int cur = 0;
   for (int py = 0; py < SP; ++py) {
-    for (int q = tid; q < 2 * S2 * C2; q += WG)
-      srow[q] = z2[(long long)(2 * py) * S2 * C2 + q];
-    __syncthreads();
+    if (py + 1 < SP) { PF_LOAD(py + 1); }
     for (int q = tid; q < C2 * SP; q += WG) {
       const int c = q / SP;
       const int px = q - c * SP;
@@ -284,14 +295,18 @@ void cnn_pool_fwd(CnnArgs a) {
       for (int dy = 0; dy < 2; ++dy)
 #pragma unroll
         for (int dx = 0; dx < 2; ++dx) {
-          const float z = srow[(dy * S2 + 2 * px + dx) * C2 + c];
+          const float z = srow[cur][(dy * S2 + 2 * px + dx) * C2 + c];
           if (z > best) { best = z; arg = dy * 2 + dx; }
         }
       sa2[c * (SP * SP) + py * SP + px] = best;
       spidx[c * (SP * SP) + py * SP + px] = (unsigned char)arg;
     }
+    if (py + 1 < SP) { PF_WRITE(cur ^ 1); }
     __syncthreads();
+    cur ^= 1;
   }
+#undef PF_LOAD
+#undef PF_WRITE
   float* a2o = a.a2 + ((long long)g * a.B + b) * NF;
   unsigned char* po = a.pidx + ((long long)g * a.B + b) * NF;
   for (int e = tid; e < NF; e += WG) {
