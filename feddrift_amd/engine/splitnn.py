@@ -51,3 +51,28 @@ class SplitNN:
                  y: torch.Tensor) -> float:
         logits = self.server_model(self.client_models[client_idx](x))
         return float((logits.argmax(-1) == y).float().mean().item())
+
+
+class SplitNNRelay(SplitNN):
+    """The reference's relay protocol (fedml_api/distributed/split_nn):
+    clients form a ring and pass a turn token; the active client runs
+    `epochs_per_node` local epochs against the SHARED server-side model
+    (activations up, cut-layer gradients down), then a validation pass,
+    then hands the token to node_right (client.py:12-14,
+    client_manager.py semaphore flow). One lap = every client's turn."""
+
+    def run_lap(self, data, epochs_per_node: int = 1,
+                batch_size: int = 32) -> dict:
+        losses = {}
+        n_clients = len(self.client_models)
+        for turn in range(n_clients):       # ring order 0 -> 1 -> ... -> 0
+            x, y = data[turn]
+            n = x.shape[0]
+            loss_sum = steps = 0
+            for _ in range(epochs_per_node):
+                for i in range(0, n, batch_size):
+                    loss_sum += self.train_step(turn, x[i:i + batch_size],
+                                                y[i:i + batch_size])
+                    steps += 1
+            losses[turn] = loss_sum / max(steps, 1)
+        return losses
