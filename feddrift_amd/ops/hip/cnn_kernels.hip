@@ -817,14 +817,14 @@ void cnn_conv2_wgrad_reduce(CnnArgs a) {
 }
 
 // conv2 dgrad as MFMA GEMM: dx1[m=(b,y,x)][ci] over K = (kyx, co).
-// The dz REGION feeding all 9 taps stages once per block (6 rows x 28
-// cols with zero borders, channels-last) — then per tap the weight
-// B-tile cycles through a double-buffered LDS slab shared by all 4
-// waves (register-staged split: next tap's loads issue before the
-// current tap's MFMAs, written after).
+// 128-pixel tiles (2 row-tiles per wave) halve the staged bytes per
+// MFMA vs 64-pixel tiles: the dz REGION feeding all 9 taps (8 rows x
+// 28 cols, zero-padded borders) stages once per block — ONE barrier,
+// then 576 MFMAs per wave with the B operand (wtd, L2-resident)
+// loaded straight to registers.
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_conv2_dgrad_mfma(CnnArgs a) {
-  const int mtiles = 11;                  // ceil(676 / 64)
+  const int mtiles = 6;                   // ceil(676 / 128)
   const int mt = blockIdx.x % mtiles;
   const int gb = blockIdx.x / mtiles;
   const int g = gb / a.B;
@@ -834,22 +834,14 @@ void cnn_conv2_dgrad_mfma(CnnArgs a) {
   const int wv = tid >> 6;
   const int l = tid & 63;
   const int li = l & 15, lk = l >> 4;
-  __shared__ __attribute__((aligned(16))) float sD[6 * 28][C2 + 1];
-  __shared__ __attribute__((aligned(16))) float sW[2][C2][C1 + 1];
-  const int p0 = mt * 64;
+  // dz region rows [y0-2, y0+6) x cols [-2, 26), zero-padded borders
+  __shared__ __attribute__((aligned(16))) float sD[8 * 28][C2 + 1];
+  const int p0 = mt * 128;
   const int y0 = p0 / S1;
   const float* dz = a.zz2 + ((long long)g * a.B + b) * Z2N;
-  const float* wt = a.wtd + (long long)g * 9 * 2048;
-  const int r8 = tid >> 5, kk = tid & 31;
-  float rw[8];
-#define DG_LOADW(kyx)                                                   \
-  _Pragma("unroll") for (int j = 0; j < 8; ++j)                          \
-    rw[j] = wt[(kyx) * 2048 + (r8 + j * 8) * C1 + kk];
-#define DG_WRITEW(buf)                                                  \
-  _Pragma("unroll") for (int j = 0; j < 8; ++j)                          \
-    sW[buf][r8 + j * 8][kk] = rw[j];
   {
-    for (int rr = r8; rr < 6 * 28; rr += 8) {
+    const int r8 = tid >> 5, kk = tid & 31;
+    for (int rr = r8; rr < 8 * 28; rr += 8) {
       const int dzrow = rr / 28 + y0 - 2;
       const int col = rr - (rr / 28) * 28 - 2;
       const bool ok = dzrow >= 0 && dzrow < S2 && col >= 0 && col < S2;
@@ -858,43 +850,50 @@ void cnn_conv2_dgrad_mfma(CnnArgs a) {
       sD[rr][kk + 32] = ok ? dz[src + kk + 32] : 0.f;
     }
   }
-  DG_LOADW(0);
-  DG_WRITEW(0);
   __syncthreads();
-  f32x4 acc[2];
-  acc[0] = f32x4{0.f, 0.f, 0.f, 0.f};
-  acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
-  const int p = p0 + wv * 16 + li;        // this lane's x1 pixel
-  const int y = p / S1, x = p - (p / S1) * S1;
-  int cur = 0;
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int rt = 0; rt < 2; ++rt)
+#pragma unroll
+    for (int ct = 0; ct < 2; ++ct)
+      acc[rt][ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+  // this lane's two x1 pixels (row-tiles wv*32 + {0,16} + li)
+  int py[2], px[2];
+#pragma unroll
+  for (int rt = 0; rt < 2; ++rt) {
+    const int p = p0 + wv * 32 + rt * 16 + li;
+    py[rt] = p / S1;
+    px[rt] = p - (p / S1) * S1;
+  }
+  const float* wt = a.wtd + (long long)g * 9 * 2048;
   for (int kyx = 0; kyx < 9; ++kyx) {
-    if (kyx + 1 < 9) { DG_LOADW(kyx + 1); }
     const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
-    const int arow = (y - ky - y0 + 2) * 28 + (x - kx + 2);
+    const float* wk = wt + kyx * 2048;
 #pragma unroll
     for (int ks = 0; ks < C2 / 4; ++ks) {
-      const float av = sD[arow][ks * 4 + lk];
 #pragma unroll
-      for (int ct = 0; ct < 2; ++ct)
-        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-            av, sW[cur][ks * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
+      for (int rt = 0; rt < 2; ++rt) {
+        const int arow = (py[rt] - ky - y0 + 2) * 28 + (px[rt] - kx + 2);
+        const float av = sD[arow][ks * 4 + lk];
+#pragma unroll
+        for (int ct = 0; ct < 2; ++ct)
+          acc[rt][ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              av, wk[(ks * 4 + lk) * C1 + ct * 16 + li], acc[rt][ct],
+              0, 0, 0);
+      }
     }
-    if (kyx + 1 < 9) { DG_WRITEW(cur ^ 1); }
-    __syncthreads();
-    cur ^= 1;
   }
-#undef DG_LOADW
-#undef DG_WRITEW
   float* dx1 = a.dx1 + ((long long)g * a.B + b) * X1N;
 #pragma unroll
-  for (int ct = 0; ct < 2; ++ct) {
+  for (int rt = 0; rt < 2; ++rt)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int pix = p0 + wv * 16 + lk * 4 + r;
-      if (pix < 676)
-        dx1[(long long)pix * C1 + ct * 16 + li] = acc[ct][r];
-    }
-  }
+    for (int ct = 0; ct < 2; ++ct)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int pix = p0 + wv * 32 + rt * 16 + lk * 4 + r;
+        if (pix < 676)
+          dx1[(long long)pix * C1 + ct * 16 + li] = acc[rt][ct][r];
+      }
 }
 
 // conv1 wgrad, stage 1: per-(g, b) partials into the dz2 scratch region
@@ -1427,7 +1426,7 @@ void cnn_train_epoch_impl(
   hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * a.w2ms), dim3(WG),
                      0, s, a);
   L(cnn_conv2_wgrad_reduce, (long long)G * (9 * 2048 + C2));
-  hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB * 11), dim3(WG), 0,
+  hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB * 6), dim3(WG), 0,
                      s, a);
   hipLaunchKernelGGL(cnn_conv1_wgrad_part, dim3(G * (int)B), dim3(WG), 0,
                      s, a);
