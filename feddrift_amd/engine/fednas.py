@@ -11,7 +11,7 @@ machinery testable; the search loop is model-agnostic.
 
 from __future__ import annotations
 
-from typing import List, Tuple
+from typing import List
 
 import torch
 import torch.nn.functional as F

@@ -10,7 +10,7 @@ cut at the split layer) is the same.
 
 from __future__ import annotations
 
-from typing import List, Tuple
+from typing import List
 
 import torch
 import torch.nn.functional as F

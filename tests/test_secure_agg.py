@@ -4,7 +4,6 @@ with secure_agg=1 must reproduce the unmasked run to fp-roundoff, while
 the pre-reduce partials (what a rank exposes on the wire) differ."""
 
 import numpy as np
-import pytest
 import torch
 
 from feddrift_amd.comm import Communicator
