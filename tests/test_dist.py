@@ -332,3 +332,13 @@ def test_world8_with_idle_ranks(tmp_path):
     r8 = _run(8, data, str(tmp_path / "w8"), 29626)
     assert abs(r1[0] - r8[0]) < 1e-6, (r1, r8)
     assert np.allclose(r1, r8, atol=0.02), (r1, r8)
+
+
+def test_sequential_runs_reuse_port(tmp_path):
+    """Two back-to-back torchrun launches on the SAME master port (the
+    driver reuses its port across the N=1,2,4,8 scaling runs; a stale
+    TIME_WAIT socket must not break the rendezvous)."""
+    data = _write_data(tmp_path)
+    r_a = _run(2, data, str(tmp_path / "pa"), 29650)
+    r_b = _run(2, data, str(tmp_path / "pb"), 29650)
+    assert np.allclose(r_a, r_b, atol=1e-12), (r_a, r_b)
