@@ -430,29 +430,42 @@ class SoftClusterState:
 
     def cluster_cfl(self, hooks: EngineHooks, curr_iter: int, round_idx: int,
                     weight_updates_by_model: Dict[int, List[np.ndarray]],
-                    clients_by_model: Dict[int, np.ndarray]) -> bool:
+                    clients_by_model: Dict[int, np.ndarray],
+                    stats_by_model: Optional[Dict[int, tuple]] = None
+                    ) -> bool:
         """One CFL round: binary split check per active cluster.
 
         weight_updates_by_model[m] = flattened (local - global) updates of
         the clients (with data) in cluster m, same order as
-        clients_by_model[m] (reference :1159-1223).
+        clients_by_model[m] (reference :1159-1223). The engine may pass
+        stats_by_model[m] = (max_norm, mean_norm, sims) precomputed on
+        the GPU (norms + the pairwise cosine matrix are O(n^2 * P) — at
+        ResNet scale that is a GEMM, not host numpy work); the raw
+        updates are then not consulted at all.
         """
         did_split = False
         for model_idx, clients in clients_by_model.items():
-            updates = weight_updates_by_model[model_idx]
-            if not updates:
-                continue
-            stack = np.stack(updates)
-            norms = np.linalg.norm(stack, axis=1)
-            max_norm = float(np.max(norms))
-            mean_norm = float(np.linalg.norm(np.mean(stack, axis=0)))
+            if stats_by_model is not None:
+                if model_idx not in stats_by_model:
+                    continue
+                max_norm, mean_norm, sims_pre = stats_by_model[model_idx]
+            else:
+                updates = weight_updates_by_model[model_idx]
+                if not updates:
+                    continue
+                stack = np.stack(updates)
+                norms = np.linalg.norm(stack, axis=1)
+                max_norm = float(np.max(norms))
+                mean_norm = float(np.linalg.norm(np.mean(stack, axis=0)))
+                sims_pre = None
             if mean_norm > self.cfl_norm:
                 self.cfl_norm = mean_norm
                 self.cfl_eps1 = self.cfl_norm / 10.0
                 self.cfl_eps2 = 6 * self.cfl_eps1
             else:
                 if mean_norm < self.cfl_eps1 and max_norm > self.cfl_eps2:
-                    sims = self._pairwise_cos(stack)
+                    sims = sims_pre if sims_pre is not None \
+                        else self._pairwise_cos(stack)
                     cl1, cl2 = self._bipartition(sims)
                     alpha_cross = max(max(sims[i, j] for j in cl2)
                                       for i in cl1)

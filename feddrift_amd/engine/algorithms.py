@@ -379,26 +379,37 @@ class SoftClusterAlgo(AlgoBase):
         rows_per_chunk = max(1, (256 << 20) // max(4 * P, 1))
         for i0 in range(0, len(pairs), rows_per_chunk):
             job.comm.all_reduce_(deltas[i0:i0 + rows_per_chunk])
-        dall = deltas.cpu().numpy()
-        updates: Dict[tuple, np.ndarray] = {
-            (int(client_idx[w]), m): dall[i]
-            for i, (w, m) in enumerate(pairs)}
+        row_of = {(int(client_idx[w]), m): i
+                  for i, (w, m) in enumerate(pairs)}
         models_in_use = [m for m in range(K)
                          if np.any(st.get_weights()[job.curr_iter][m] > 0)]
         clients_by_model = {
             m: np.nonzero(st.get_weights()[job.curr_iter][m])[0]
             for m in models_in_use}
-        upd_by_model = {
-            m: [updates[(c, m)] for c in clients_by_model[m]
-                if (c, m) in updates]
-            for m in models_in_use}
         # keep client lists aligned with the updates actually present
         clients_aligned = {
             m: np.array([c for c in clients_by_model[m]
-                         if (c, m) in updates])
+                         if (c, m) in row_of])
             for m in models_in_use}
+        # norm + pairwise-cosine statistics on the DEVICE (O(n^2 * P)
+        # GEMM — host numpy at ResNet scale was the split check's cost);
+        # only the tiny per-cluster matrices cross to the host
+        stats = {}
+        for m in models_in_use:
+            cl = clients_aligned[m]
+            if len(cl) == 0:
+                continue
+            idx = torch.as_tensor([row_of[(int(c), m)] for c in cl],
+                                  device=job.device)
+            D = deltas[idx]
+            norms = D.norm(dim=1)
+            mean_norm = float(D.mean(dim=0).norm())
+            sims = (D @ D.T) / (torch.outer(norms, norms) + 1e-12)
+            stats[m] = (float(norms.max()), mean_norm,
+                        sims.cpu().numpy())
         return st.cluster_cfl(Hooks(job), job.curr_iter, round_idx + 1,
-                              upd_by_model, clients_aligned)
+                              None, clients_aligned,
+                              stats_by_model=stats)
 
     def test(self, job: FLJob, round_idx: int) -> None:
         cfg = job.cfg
