@@ -69,7 +69,8 @@ def main():
 
     results = {}
     for algo, arg in ALGOS:
-        if a.algos and a.algos not in algo:
+        if a.algos and not any(tok and tok in f"{algo}:{arg}"
+                               for tok in a.algos.split(",")):
             continue
         name = f"{algo}:{arg}" if arg else algo
         log_dir = os.path.join(a.data_dir, "run_" + name.replace(":", "_"))
