@@ -817,16 +817,22 @@ void cnn_conv2_wgrad_reduce(CnnArgs a) {
 }
 
 // conv2 dgrad as MFMA GEMM: dx1[m=(b,y,x)][ci] over K = (kyx, co).
-// 128-pixel tiles (2 row-tiles per wave) halve the staged bytes per
-// MFMA vs 64-pixel tiles: the dz REGION feeding all 9 taps (8 rows x
-// 28 cols, zero-padded borders) stages once per block — ONE barrier,
-// then 576 MFMAs per wave with the B operand (wtd, L2-resident)
-// loaded straight to registers.
+// ONE block per (g, b) walks all 11 64-pixel tiles with EVERYTHING
+// LDS-resident via global_load_lds (async DMA): the full 74 KB weight
+// tensor stages once in the prologue, and the 37 KB dz region
+// double-buffers so the next tile streams while the current tile's
+// 288 MFMAs issue (counted-wait + raw barriers; __syncthreads or any
+// ordinary global load in the loop would drain the DMA queue — guide
+// T3/T4 and the §5 mixing-load-kinds trap). The region image is linear
+// (glds requirement): bank spread comes from a 4-float channel
+// rotation keyed on the row-col index, applied identically at the DMA
+// source address and at read time; out-of-bounds taps resolve by
+// operand predication instead of staged zero borders.
+#define DG_RC (6 * S2)              // 144 row-cols per tile region
+
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_conv2_dgrad_mfma(CnnArgs a) {
-  const int mtiles = 6;                   // ceil(676 / 128)
-  const int mt = blockIdx.x % mtiles;
-  const int gb = blockIdx.x / mtiles;
+  const int gb = blockIdx.x;
   const int g = gb / a.B;
   const int b = gb - g * a.B;
   if (b >= step_n(a, g)) return;
@@ -834,66 +840,101 @@ void cnn_conv2_dgrad_mfma(CnnArgs a) {
   const int wv = tid >> 6;
   const int l = tid & 63;
   const int li = l & 15, lk = l >> 4;
-  // dz region rows [y0-2, y0+6) x cols [-2, 26), zero-padded borders
-  __shared__ __attribute__((aligned(16))) float sD[8 * 28][C2 + 1];
-  const int p0 = mt * 128;
-  const int y0 = p0 / S1;
+  // carve: sB [9*2048] then sD [2][DG_RC][64]
+  extern __shared__ __attribute__((aligned(16))) float lds_[];
+  float* sB = lds_;
+  float* sD = lds_ + 9 * 2048;
   const float* dz = a.zz2 + ((long long)g * a.B + b) * Z2N;
-  {
-    const int r8 = tid >> 5, kk = tid & 31;
-    for (int rr = r8; rr < 8 * 28; rr += 8) {
-      const int dzrow = rr / 28 + y0 - 2;
-      const int col = rr - (rr / 28) * 28 - 2;
-      const bool ok = dzrow >= 0 && dzrow < S2 && col >= 0 && col < S2;
-      const long long src = ((long long)dzrow * S2 + col) * C2;
-      sD[rr][kk] = ok ? dz[src + kk] : 0.f;
-      sD[rr][kk + 32] = ok ? dz[src + kk + 32] : 0.f;
-    }
-  }
-  __syncthreads();
-  f32x4 acc[2][2];
-#pragma unroll
-  for (int rt = 0; rt < 2; ++rt)
-#pragma unroll
-    for (int ct = 0; ct < 2; ++ct)
-      acc[rt][ct] = f32x4{0.f, 0.f, 0.f, 0.f};
-  // this lane's two x1 pixels (row-tiles wv*32 + {0,16} + li)
-  int py[2], px[2];
-#pragma unroll
-  for (int rt = 0; rt < 2; ++rt) {
-    const int p = p0 + wv * 32 + rt * 16 + li;
-    py[rt] = p / S1;
-    px[rt] = p - (p / S1) * S1;
-  }
   const float* wt = a.wtd + (long long)g * 9 * 2048;
-  for (int kyx = 0; kyx < 9; ++kyx) {
-    const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
-    const float* wk = wt + kyx * 2048;
+  float* dx1 = a.dx1 + ((long long)g * a.B + b) * X1N;
+
+  // prologue: the whole wtd[9][64][32] tensor -> LDS (18 glds per wave)
+  {
+    const int base = wv * (18 * 256);       // floats
 #pragma unroll
-    for (int ks = 0; ks < C2 / 4; ++ks) {
+    for (int gq = 0; gq < 18; ++gq)
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)
+              (wt + base + gq * 256 + l * 4),
+          (__attribute__((address_space(3))) unsigned int*)
+              (sB + base + gq * 256),
+          16, 0, 0);
+  }
+
+  // each wave DMAs rows {wv, wv + 4 partial} of the 6-row region: the
+  // 144 row-cols split as 36 glds (4 arows each) -> 9 per wave; invalid
+  // rows load from a clamped source row and are masked at use
+#define DG_ISSUE(mt, buf)                                                \
+  {                                                                      \
+    const int y0_ = (mt) * 64 / S1;                                      \
+    _Pragma("unroll") for (int gq = 0; gq < 9; ++gq) {                   \
+      const int arow0_ = (wv * 9 + gq) * 4;                              \
+      const int arow_l = arow0_ + (l >> 4);                              \
+      const int row_ = arow_l / S2;                                      \
+      const int col_ = arow_l - row_ * S2;                               \
+      int dzrow_ = y0_ - 2 + row_;  /* region = dz rows [y0-2, y0+4) */  \
+      dzrow_ = dzrow_ < 0 ? 0 : (dzrow_ > S2 - 1 ? S2 - 1 : dzrow_);     \
+      const int chp_ = (l & 15) * 4;                                     \
+      const int chs_ = (chp_ - 4 * arow_l) & 63;                         \
+      __builtin_amdgcn_global_load_lds(                                  \
+          (const __attribute__((address_space(1))) unsigned int*)        \
+              (dz + ((long long)dzrow_ * S2 + col_) * C2 + chs_),        \
+          (__attribute__((address_space(3))) unsigned int*)              \
+              (sD + (buf) * (DG_RC * 64) + arow0_ * 64),                 \
+          16, 0, 0);                                                     \
+    }                                                                    \
+  }
+
+  DG_ISSUE(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  int cur = 0;
+  for (int mt = 0; mt < 11; ++mt) {
+    if (mt + 1 < 11) { DG_ISSUE(mt + 1, cur ^ 1); }
+    const int p0 = mt * 64;
+    const int y0 = p0 / S1;
+    const int p = p0 + wv * 16 + li;        // this lane's x1 pixel
+    const int y = p / S1, x = p - (p / S1) * S1;
+    f32x4 acc[2];
+    acc[0] = f32x4{0.f, 0.f, 0.f, 0.f};
+    acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+    const float* base = sD + cur * (DG_RC * 64);
+    for (int kyx = 0; kyx < 9; ++kyx) {
+      const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+      const float* wk = sB + kyx * 2048;
+      const int oy = y - ky, ox = x - kx;
+      // region = dz rows [y0-2, y0+4): every VALID tap of this tile's
+      // pixels lands inside (oy in [y-2, y], y in [y0, y0+2])
+      const bool ok = oy >= 0 && oy < S2 && ox >= 0 && ox < S2;
+      const int arow = ok ? (oy - y0 + 2) * S2 + ox : 0;
+      const int rot = 4 * arow;
+      const float* rowp = base + arow * 64;
 #pragma unroll
-      for (int rt = 0; rt < 2; ++rt) {
-        const int arow = (py[rt] - ky - y0 + 2) * 28 + (px[rt] - kx + 2);
-        const float av = sD[arow][ks * 4 + lk];
+      for (int ks = 0; ks < C2 / 4; ++ks) {
+        const int co = ks * 4 + lk;
+        float av = rowp[(co + rot) & 63];
+        av = ok ? av : 0.f;
 #pragma unroll
         for (int ct = 0; ct < 2; ++ct)
-          acc[rt][ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-              av, wk[(ks * 4 + lk) * C1 + ct * 16 + li], acc[rt][ct],
-              0, 0, 0);
+          acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              av, wk[co * C1 + ct * 16 + li], acc[ct], 0, 0, 0);
       }
     }
-  }
-  float* dx1 = a.dx1 + ((long long)g * a.B + b) * X1N;
-#pragma unroll
-  for (int rt = 0; rt < 2; ++rt)
 #pragma unroll
     for (int ct = 0; ct < 2; ++ct)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int pix = p0 + wv * 32 + rt * 16 + lk * 4 + r;
+        const int pix = p0 + wv * 16 + lk * 4 + r;
         if (pix < 676)
-          dx1[(long long)pix * C1 + ct * 16 + li] = acc[rt][ct][r];
+          dx1[(long long)pix * C1 + ct * 16 + li] = acc[ct][r];
       }
+    if (mt + 1 < 11) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+#undef DG_ISSUE
 }
 
 // conv1 wgrad, stage 1: per-(g, b) partials into the dz2 scratch region
@@ -1426,8 +1467,8 @@ void cnn_train_epoch_impl(
   hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * a.w2ms), dim3(WG),
                      0, s, a);
   L(cnn_conv2_wgrad_reduce, (long long)G * (9 * 2048 + C2));
-  hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB * 6), dim3(WG), 0,
-                     s, a);
+  hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB), dim3(WG),
+                     (9 * 2048 + 2 * DG_RC * 64) * sizeof(float), s, a);
   hipLaunchKernelGGL(cnn_conv1_wgrad_part, dim3(G * (int)B), dim3(WG), 0,
                      s, a);
   L(cnn_conv1_wgrad_reduce, (long long)G * 320);
