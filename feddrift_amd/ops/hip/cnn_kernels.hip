@@ -692,62 +692,70 @@ void cnn_conv2_wgrad_mfma(CnnArgs a) {
   const int wv = tid >> 6;
   const int l = tid & 63;
   const int li = l & 15, lk = l >> 4;
-  __shared__ __attribute__((aligned(16))) float sR[2][6 * S1][C1 + 1];
-  __shared__ __attribute__((aligned(16))) float sD[2][64][C2 + 1];
+  // all staging by async DMA (global_load_lds): both operands are read
+  // stride-1 per lane group, so LINEAR LDS images need no padding or
+  // swizzle. Region [160 rows x 32 ci] (rows 156..159 are never read),
+  // dz tile [64 m x 64 co]; both double-buffered with counted waits +
+  // raw barriers so the next m-tile streams under this tile's MFMAs.
+  // ONE shared object (a second __shared__ makes hipcc drain vmcnt(0)
+  // before every ds_read beside an in-flight glds — guide §5 trap (a))
+  __shared__ __attribute__((aligned(16))) float ldsw[2 * 160 * C1
+                                                     + 2 * 64 * C2];
+  float (*sR)[160 * C1] = (float (*)[160 * C1])ldsw;
+  float (*sD)[64 * C2] = (float (*)[64 * C2])(ldsw + 2 * 160 * C1);
   f32x4 acc[18];
 #pragma unroll
   for (int t = 0; t < 18; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
-  const int r8 = tid >> 5, kk = tid & 31;
   const long long mtiles = ((long long)n * (S2 * S2) + 63) / 64;
-  // register-staged double buffering: 20 region values + 16 dz values
-  // per thread per tile; tile t+w2ms's loads issue before tile t's
-  // MFMAs, the LDS writes land after
-  float rr_[20], rd_[16];
-  int cur = 0;
-#define WG_LOAD(MT)                                                     \
+  // per wave per tile: 5 region glds + 4 dz glds (1 KiB each)
+#define WG_ISSUE(MT, BUF)                                                \
   {                                                                      \
     const long long m0_ = (MT) * 64;                                     \
     const int b_ = (int)(m0_ / (S2 * S2));                               \
     const int p0_ = (int)(m0_ - (long long)b_ * (S2 * S2));              \
     const int r0_ = p0_ / S2;                                            \
+    const int nrow_ = (S1 - r0_) * S1;                                   \
     const long long gb_ = (long long)g * a.B + b_;                       \
     const float* x1_ = a.x1 + (gb_ * 676 + (long long)r0_ * S1) * C1;    \
     const float* dz_ = a.zz2 + (gb_ * 576 + p0_) * C2;                   \
-    const int nrow_ = min(6 * S1, (S1 - r0_) * S1);                      \
-    _Pragma("unroll") for (int jj = 0; jj < 20; ++jj) {                  \
-      const int rr2 = r8 + jj * 8;                                       \
-      rr_[jj] = (rr2 < nrow_)                                            \
-          ? x1_[(long long)rr2 * C1 + kk] : 0.f;                         \
+    _Pragma("unroll") for (int gq = 0; gq < 5; ++gq) {                   \
+      const int f0_ = (wv * 5 + gq) * 256;      /* region floats */      \
+      int srow_ = (f0_ + l * 4) / C1;           /* source arow */        \
+      srow_ = srow_ < nrow_ ? srow_ : nrow_ - 1;                         \
+      const int sch_ = (f0_ + l * 4) & 31;                               \
+      __builtin_amdgcn_global_load_lds(                                  \
+          (const __attribute__((address_space(1))) unsigned int*)        \
+              (x1_ + (long long)srow_ * C1 + sch_),                      \
+          (__attribute__((address_space(3))) unsigned int*)              \
+              (&sR[BUF][f0_]),                                           \
+          16, 0, 0);                                                     \
     }                                                                    \
-    _Pragma("unroll") for (int jj = 0; jj < 8; ++jj) {                   \
-      rd_[jj] = dz_[(long long)(r8 + jj * 8) * C2 + kk];                 \
-      rd_[jj + 8] = dz_[(long long)(r8 + jj * 8) * C2 + kk + 32];        \
+    _Pragma("unroll") for (int gq = 0; gq < 4; ++gq) {                   \
+      const int f0_ = (wv * 4 + gq) * 256;                               \
+      __builtin_amdgcn_global_load_lds(                                  \
+          (const __attribute__((address_space(1))) unsigned int*)        \
+              (dz_ + f0_ + l * 4),                                       \
+          (__attribute__((address_space(3))) unsigned int*)              \
+              (&sD[BUF][f0_]),                                           \
+          16, 0, 0);                                                     \
     }                                                                    \
   }
-#define WG_WRITE(buf)                                                   \
-  _Pragma("unroll") for (int jj = 0; jj < 20; ++jj) {                    \
-    const int rr2 = r8 + jj * 8;                                         \
-    if (rr2 < 6 * S1) sR[buf][rr2][kk] = rr_[jj];                        \
-  }                                                                      \
-  _Pragma("unroll") for (int jj = 0; jj < 8; ++jj) {                     \
-    sD[buf][r8 + jj * 8][kk] = rd_[jj];                                  \
-    sD[buf][r8 + jj * 8][kk + 32] = rd_[jj + 8];                         \
-  }
-  if (ms < mtiles) {
-    WG_LOAD(ms);
-    WG_WRITE(0);
-  }
-  __syncthreads();
+  int cur = 0;
+  if (ms < mtiles) { WG_ISSUE(ms, 0); }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
   for (long long mt = ms; mt < mtiles; mt += a.w2ms) {
     const int p0 = (int)((mt * 64) % (S2 * S2));
     const int r0 = p0 / S2;
-    if (mt + a.w2ms < mtiles) { WG_LOAD(mt + a.w2ms); }
+    if (mt + a.w2ms < mtiles) { WG_ISSUE(mt + a.w2ms, cur ^ 1); }
     int parow[16];
 #pragma unroll
     for (int km = 0; km < 16; ++km) {
       const int q = p0 + km * 4 + lk;
       parow[km] = (q / S2 - r0) * S1 + (q - (q / S2) * S2);
     }
+    const float* bR = sR[cur];
+    const float* bD = sD[cur];
 #pragma unroll
     for (int kyx = 0; kyx < 9; ++kyx) {
       const int kyoff = (kyx / 3) * S1 + (kyx - (kyx / 3) * 3);
@@ -759,17 +767,19 @@ void cnn_conv2_wgrad_mfma(CnnArgs a) {
 #pragma unroll
         for (int km = 0; km < 16; ++km) {
           acc[slot] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-              sR[cur][parow[km] + kyoff][rt * 16 + li],
-              sD[cur][km * 4 + lk][ct * 16 + li], acc[slot], 0, 0, 0);
+              bR[(parow[km] + kyoff) * C1 + rt * 16 + li],
+              bD[(km * 4 + lk) * C2 + ct * 16 + li], acc[slot],
+              0, 0, 0);
         }
       }
     }
-    if (mt + a.w2ms < mtiles) { WG_WRITE(cur ^ 1); }
-    __syncthreads();
+    if (mt + a.w2ms < mtiles) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
     cur ^= 1;
   }
-#undef WG_LOAD
-#undef WG_WRITE
+#undef WG_ISSUE
 #pragma unroll
   for (int slot = 0; slot < 18; ++slot) {
     const int kyx = (slot * 4 + wv) / 8;
