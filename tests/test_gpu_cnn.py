@@ -266,3 +266,33 @@ def test_cnn_dropout_statistics():
     # params should still be in the same ballpark (scaled masks)
     assert (reps - gp[torch.as_tensor(plan.rows, device=DEV) % K]
             ).abs().max().item() < 1.0
+
+
+def test_cnn_engine_selected_on_gpu(tmp_path):
+    """On a GPU box the cnn model MUST run on CnnHipEngine (the
+    hand-written kernel path) — a silent fallback to the vmap/MIOpen
+    engine would pass numerics while abandoning the native path."""
+    from feddrift_amd.comm import Communicator
+    from feddrift_amd.config import Config
+    from feddrift_amd.data.generators import sample_mnist
+    from feddrift_amd.data.loader import DriftDataset
+    from feddrift_amd.engine.fljob import FLJob
+    from feddrift_amd.eval.metrics import MetricLogger
+    ds = DriftDataset(data_dir="/nonexistent", dataset="MNIST",
+                      num_client=2)
+    rng = np.random.default_rng(0)
+    for c in range(2):
+        for t in range(3):
+            arr = sample_mnist(40, 0, rng)
+            ds.store.put(c, t, arr[:, :-1], arr[:, -1])
+    cfg = Config(model="cnn", dataset="MNIST", data_dir="/nonexistent",
+                 client_num_in_total=2, client_num_per_round=2,
+                 batch_size=20, epochs=1, comm_round=1,
+                 total_train_iteration=2, curr_train_iteration=1,
+                 concept_num=2, concept_drift_algo="softcluster",
+                 concept_drift_algo_arg="mmacc_06", bench_mode=1,
+                 report_client=0, log_dir=str(tmp_path))
+    job = FLJob(cfg, Communicator(),
+                MetricLogger(enabled=False, to_file=False), dataset=ds)
+    assert isinstance(job.mod_engine, CnnHipEngine), type(job.mod_engine)
+    assert job.device.type == "cuda"

@@ -272,3 +272,34 @@ def test_cfl_split_on_module_path(tmp_path):
     job = FLJob(cfg, comm, logger, dataset=ds)
     job.run()
     assert np.isfinite(logger.mean("Test/Acc"))
+
+
+def test_cnn_hip_engine_not_selected_on_cpu(tmp_path):
+    """CPU runs stay on the vmap engine (the HIP kernels need a GPU);
+    use_hip_kernels='never' must also force the torch path on any
+    device — the escape hatch for A/B runs."""
+    import numpy as np
+    from feddrift_amd.comm import Communicator
+    from feddrift_amd.config import Config
+    from feddrift_amd.data.generators import sample_mnist
+    from feddrift_amd.data.loader import DriftDataset
+    from feddrift_amd.engine.fljob import FLJob
+    from feddrift_amd.eval.metrics import MetricLogger
+    from feddrift_amd.ops.module_vmap import VmapEngine
+    ds = DriftDataset(data_dir="/nonexistent", dataset="MNIST",
+                      num_client=2)
+    rng = np.random.default_rng(0)
+    for c in range(2):
+        for t in range(3):
+            arr = sample_mnist(30, 0, rng)
+            ds.store.put(c, t, arr[:, :-1], arr[:, -1])
+    cfg = Config(model="cnn", dataset="MNIST", data_dir="/nonexistent",
+                 client_num_in_total=2, client_num_per_round=2,
+                 batch_size=15, epochs=1, comm_round=1,
+                 total_train_iteration=2, curr_train_iteration=1,
+                 concept_num=2, concept_drift_algo="softcluster",
+                 concept_drift_algo_arg="mmacc_06", bench_mode=1,
+                 report_client=0, log_dir=str(tmp_path))
+    job = FLJob(cfg, Communicator(device=__import__("torch").device("cpu")),
+                MetricLogger(enabled=False, to_file=False), dataset=ds)
+    assert isinstance(job.mod_engine, VmapEngine), type(job.mod_engine)
