@@ -120,6 +120,11 @@ class CnnHipEngine:
             return
         B = int(plan.step_len.max()) if plan.step_len.size else 0
         if B == 0:
+            # nothing trains (reference skips n==0 batches) but the round
+            # still BROADCASTS: replicas get the global model, like vmap
+            rows = torch.as_tensor(plan.rows, dtype=torch.int64,
+                                   device=self.device)
+            replicas[rows] = global_params[rows % n_models]
             return
         chunk = self._chunk_pairs(B)
         if n_pairs > chunk:
@@ -145,6 +150,13 @@ class CnnHipEngine:
         B = int(plan.step_len.max())
         model_of = rows % n_models
         work = global_params[model_of].clone()          # [G, P]
+        if B == 0:
+            # a chunk whose every pair has zero-length steps throughout:
+            # kernel grids would be dim3(0) (invalid launch). Broadcast
+            # only — matches the unchunked kernel, which skips len-0
+            # steps per pair but still returns the staged weights.
+            replicas[rows] = work
+            return
         ws = self._workspace(G, B)
         step_off = torch.as_tensor(plan.step_off, dtype=torch.int64,
                                    device=dev)

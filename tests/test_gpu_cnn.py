@@ -186,6 +186,36 @@ def test_cnn_train_chunked_pairs():
     assert torch.equal(reps_a, reps_b)
 
 
+def test_cnn_train_zero_work_chunk():
+    """A pair whose step_len is 0 in EVERY epoch, isolated in its own
+    chunk (WS_BUDGET=1 -> one pair per chunk): the chunk's batch max is
+    0, which must broadcast-only, not launch dim3(0) grids (regression:
+    hipErrorInvalidConfiguration found by fuzz trial 28)."""
+    proto, packer, gp, x, y, plan, K, G, P = make_setup(seed=9, E=2)
+    plan.step_len[1, :] = 0          # pair 1 does no work at all
+    hipE, vmapE = fresh_engines(proto, packer)
+    hipE.WS_BUDGET = 1               # force one-pair chunks
+    res = {}
+    for name, eng in (("hip", hipE), ("vmap", vmapE)):
+        reps = torch.zeros(G, P, device=DEV)
+        opt = eng.make_opt_state("sgd", G, 0.03, 0.0)
+        eng.train(gp.clone(), reps, plan, opt, x, y, K)
+        torch.cuda.synchronize()
+        res[name] = reps.clone()
+    err = (res["hip"] - res["vmap"]).abs().max().item()
+    assert err < 1e-4, err
+    # the zero-work pair carries the broadcast model exactly
+    assert torch.equal(res["hip"][1], gp[1 % K])
+    # all-zero plan: broadcast every pair, train nothing
+    plan.step_len[:, :] = 0
+    reps = torch.full((G, P), 7.0, device=DEV)
+    opt = hipE.make_opt_state("sgd", G, 0.03, 0.0)
+    hipE.train(gp.clone(), reps, plan, opt, x, y, K)
+    torch.cuda.synchronize()
+    rows = torch.as_tensor(plan.rows, device=DEV)
+    assert torch.equal(reps, gp[rows % K])
+
+
 def _mk_tasks(rng, n_tasks, n_win, N, K, max_len=40):
     task_row, task_id, off, ln = [], [], [], []
     for w in range(n_win):
