@@ -947,6 +947,83 @@ void cnn_conv2_dgrad_mfma(CnnArgs a) {
 #undef DG_ISSUE
 }
 
+// conv2 dgrad v2 — the fwd-shaped decomposition: ONE block per
+// (g, b, mtile of 11), mirroring cnn_conv2_fwd_mfma. The block's 6-row
+// dz region (37 KB, zero-bordered, +1-padded) stages once with ordinary
+// loads and ONE barrier; the B operand (wtd, 74 KB per pair) reads
+// straight from L2 inside the MFMA loop exactly like fwd's wtf. v1
+// keeps everything (weights + region) LDS-resident via the glds DMA
+// pipeline, which is elegant but needs 144 KB of LDS -> ONE block
+// (4 waves) per CU, and its inter-tile barrier + vmcnt waits run
+// unhidden (PMC: 57% parked, profiles/pmc5_summary.json). This shape
+// fits 4 blocks/CU, so parked time overlaps across blocks.
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_conv2_dgrad_mfma2(CnnArgs a) {
+  const int mt = blockIdx.x % 11;
+  const int gb = blockIdx.x / 11;
+  const int g = gb / a.B;
+  const int b = gb - g * a.B;
+  if (b >= step_n(a, g)) return;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  // row stride C2+4 = 68 === 4 (mod 32): the MFMA-loop read
+  // sD[arow][ks*4+lk] with arow consecutive in li then hits bank
+  // 4*li + lk -- an exact 2-way spread (the 64-lane minimum)
+  __shared__ __attribute__((aligned(16))) float sD[6 * S2][C2 + 4];
+  const float* dz = a.zz2 + ((long long)g * a.B + b) * Z2N;
+  const float* wt = a.wtd + (long long)g * 9 * 2048;
+  float* dx1 = a.dx1 + ((long long)g * a.B + b) * X1N;
+  const int p0 = mt * 64;
+  const int y0 = p0 / S1;          // first dx1 pixel row of this tile
+  // region = dz rows [y0-2, y0+4): every valid tap of pixels y in
+  // [y0, y0+2] lands inside; rows outside [0, S2) stage as zeros
+  for (int e = tid; e < 6 * S2 * C2; e += WG) {
+    const int rr = e / (S2 * C2);
+    const int rem = e - rr * S2 * C2;
+    const int cc = rem / C2;
+    const int ch = rem - cc * C2;
+    const int dzrow = y0 - 2 + rr;
+    sD[rr * S2 + cc][ch] =
+        (dzrow >= 0 && dzrow < S2)
+            ? dz[((long long)dzrow * S2 + cc) * C2 + ch] : 0.f;
+  }
+  __syncthreads();
+  const int p = p0 + wv * 16 + li;        // this lane's dx1 pixel
+  const int y = p / S1, x = p - (p / S1) * S1;
+  f32x4 acc[2];
+  acc[0] = f32x4{0.f, 0.f, 0.f, 0.f};
+  acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int kyx = 0; kyx < 9; ++kyx) {
+    const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+    const int oy = y - ky, ox = x - kx;
+    // x borders (S1=26 > S2=24) still need operand predication; the
+    // staged y borders are already zero
+    const bool ok = oy >= 0 && oy < S2 && ox >= 0 && ox < S2;
+    const int arow = ok ? (oy - y0 + 2) * S2 + ox : 0;
+    const float* wk = wt + kyx * 2048;
+#pragma unroll
+    for (int ks = 0; ks < C2 / 4; ++ks) {
+      const int co = ks * 4 + lk;
+      float av = sD[arow][co];
+      av = ok ? av : 0.f;
+#pragma unroll
+      for (int ct = 0; ct < 2; ++ct)
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+            av, wk[co * C1 + ct * 16 + li], acc[ct], 0, 0, 0);
+    }
+  }
+#pragma unroll
+  for (int ct = 0; ct < 2; ++ct)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int pix = p0 + wv * 16 + lk * 4 + r;
+      if (pix < 676)
+        dx1[(long long)pix * C1 + ct * 16 + li] = acc[ct][r];
+    }
+}
+
 // conv1 wgrad, stage 1: per-(g, b) partials into the dz2 scratch region
 // (reused: dz2 is [G, B, 64] and conv1 has 288+32=320 grad entries, so
 // partials use their own ws buffer c1part [G, B, 320]). Deterministic
@@ -1477,8 +1554,20 @@ void cnn_train_epoch_impl(
   hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * a.w2ms), dim3(WG),
                      0, s, a);
   L(cnn_conv2_wgrad_reduce, (long long)G * (9 * 2048 + C2));
-  hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB), dim3(WG),
-                     (9 * 2048 + 2 * DG_RC * 64) * sizeof(float), s, a);
+  // dgrad variant switch (A/B-measured; see kernel comments):
+  // FEDDRIFT_DGRAD=1 forces the all-LDS glds pipeline (v1),
+  // FEDDRIFT_DGRAD=2 the fwd-shaped L2-operand split (v2, default:
+  // measured 463 -> 408 us on the config-3 probe, same-box A/B)
+  static const int dgrad_v = [] {
+    const char* e = getenv("FEDDRIFT_DGRAD");
+    return e ? atoi(e) : 2;
+  }();
+  if (dgrad_v == 1)
+    hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB), dim3(WG),
+                       (9 * 2048 + 2 * DG_RC * 64) * sizeof(float), s, a);
+  else
+    hipLaunchKernelGGL(cnn_conv2_dgrad_mfma2, dim3((int)GB * 11), dim3(WG),
+                       0, s, a);
   hipLaunchKernelGGL(cnn_conv1_wgrad_part, dim3(G * (int)B), dim3(WG), 0,
                      s, a);
   L(cnn_conv1_wgrad_reduce, (long long)G * 320);
