@@ -1554,14 +1554,17 @@ void cnn_train_epoch_impl(
   hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * a.w2ms), dim3(WG),
                      0, s, a);
   L(cnn_conv2_wgrad_reduce, (long long)G * (9 * 2048 + C2));
-  // dgrad variant switch (A/B-measured; see kernel comments):
-  // FEDDRIFT_DGRAD=1 forces the all-LDS glds pipeline (v1),
-  // FEDDRIFT_DGRAD=2 the fwd-shaped L2-operand split (v2, default:
-  // measured 463 -> 408 us on the config-3 probe, same-box A/B)
-  static const int dgrad_v = [] {
+  // dgrad variant switch, A/B-measured on both ends of the fleet-size
+  // axis: the fwd-shaped v2 (4 blocks/CU) wins at small fleets where
+  // occupancy is the bound (config-3 probe: 463 -> 408 us, train
+  // 8.61 -> 8.29 ms/round), the all-LDS glds pipeline v1 wins at scale
+  // where its zero-idle DMA staging is the bound (config-5, 3400
+  // clients: 5.69 vs 5.97 s/round). FEDDRIFT_DGRAD=1|2 forces either.
+  static const int dgrad_env = [] {
     const char* e = getenv("FEDDRIFT_DGRAD");
-    return e ? atoi(e) : 2;
+    return e ? atoi(e) : 0;
   }();
+  const int dgrad_v = dgrad_env ? dgrad_env : (G <= 128 ? 2 : 1);
   if (dgrad_v == 1)
     hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB), dim3(WG),
                        (9 * 2048 + 2 * DG_RC * 64) * sizeof(float), s, a);
