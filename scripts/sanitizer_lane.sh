@@ -32,6 +32,12 @@ if [ ${#args[@]} -eq 0 ]; then
 fi
 python -m pytest "${args[@]}"
 rc=$?
+if [ $rc -eq 0 ]; then
+  # the conv2-dgrad variant the host would NOT pick at test-sized fleets
+  # (the glds pipeline, chosen at scale) gets its own parity pass
+  FEDDRIFT_DGRAD=1 python -m pytest tests/test_gpu_cnn.py -x -q
+  rc=$?
+fi
 echo "sanitizer lane exit: $rc (0 = no kernel faulted under" \
      "XNACK-precise, uncached, serialized execution)"
 exit $rc
