@@ -413,11 +413,29 @@ void cnn_fc1_act(CnnArgs a) {
 
 // fc2 + softmax (in-graph model output s) + CE-on-s gradient -> dz2
 // dL/ds = (softmax(s) - onehot(y)) / n; dL/dz2 = s*(dL/ds - sum(dL/ds*s))
+//
+// ONE WAVE per sample: lane l accumulates logit l (O <= 64 always), the
+// a1 row loads once into two coalesced registers per lane, and both
+// softmaxes run as 6-step shuffle reductions — no per-thread O-arrays
+// (the previous thread-per-sample form kept z2/s/ds[64] in scratch:
+// 133 us/dispatch of spill traffic at the config-3 probe shape).
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int m = 1; m < 64; m <<= 1) v += __shfl_xor(v, m, 64);
+  return v;
+}
+__device__ __forceinline__ float wave_max(float v) {
+#pragma unroll
+  for (int m = 1; m < 64; m <<= 1) v = fmaxf(v, __shfl_xor(v, m, 64));
+  return v;
+}
+
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_head_fwd(CnnArgs a) {
   const long long total = (long long)a.G * a.B;
-  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
-       q += (long long)gridDim.x * WG) {
+  const int l = threadIdx.x & 63;
+  for (long long q = (long long)blockIdx.x * (WG / 64) + (threadIdx.x >> 6);
+       q < total; q += (long long)gridDim.x * (WG / 64)) {
     const int g = (int)(q / a.B);
     const int b = (int)(q - (long long)g * a.B);
     const int n = step_n(a, g);
@@ -426,33 +444,28 @@ void cnn_head_fwd(CnnArgs a) {
     const float* w = a.work + (long long)g * a.P;
     const float* a1 = a.a1 + q * NH;
     const int yi = (int)a.y[step_o(a, g) + b];
-    float z2[64], s[64];
-    float zmax = -1e30f;
+    const float r0 = a1[l], r1 = a1[64 + l];
+    // lane o collects logit o: each o-round is two coalesced weight
+    // loads + two fmas + a wave reduce
+    float zlane = 0.f;
     for (int o = 0; o < a.O; ++o) {
-      float z = w[OFF_W2F + (long long)a.O * NH + o];  // bias after W2f
       const float* wo = w + OFF_W2F + (long long)o * NH;
-      for (int h = 0; h < NH; ++h) z = fmaf(a1[h], wo[h], z);
-      z2[o] = z;
-      zmax = fmaxf(zmax, z);
+      const float v = wave_sum(fmaf(r0, wo[l], r1 * wo[64 + l]));
+      if (l == o) zlane = v;
     }
-    float zsum = 0.f;
-    for (int o = 0; o < a.O; ++o) { s[o] = __expf(z2[o] - zmax); zsum += s[o]; }
-    for (int o = 0; o < a.O; ++o) s[o] /= zsum;
+    const bool live = l < a.O;
+    float z = live ? zlane + w[OFF_W2F + (long long)a.O * NH + l] : -1e30f;
+    const float zmax = wave_max(z);
+    float e = live ? __expf(z - zmax) : 0.f;
+    const float s = e / wave_sum(e);               // model output s[l]
     // CE(log_softmax(s), y): q2 = softmax(s)
-    float smax = -1e30f;
-    for (int o = 0; o < a.O; ++o) smax = fmaxf(smax, s[o]);
-    float ssum = 0.f;
-    for (int o = 0; o < a.O; ++o) ssum += __expf(s[o] - smax);
-    float dot = 0.f;
-    float ds[64];
-    for (int o = 0; o < a.O; ++o) {
-      float q2 = __expf(s[o] - smax) / ssum;
-      if (o == yi) q2 -= 1.f;
-      ds[o] = q2 * inv_n;
-      dot += ds[o] * s[o];
-    }
-    for (int o = 0; o < a.O; ++o)
-      a.dz2[q * a.O + o] = s[o] * (ds[o] - dot);
+    const float smax = wave_max(live ? s : -1e30f);
+    float e2 = live ? __expf(s - smax) : 0.f;
+    float q2 = e2 / wave_sum(e2);
+    if (l == yi) q2 -= 1.f;
+    const float ds = q2 * inv_n;
+    const float dot = wave_sum(live ? ds * s : 0.f);
+    if (live) a.dz2[q * a.O + l] = s * (ds - dot);
   }
 }
 
@@ -1402,54 +1415,62 @@ void cnn_eval_fc1_mfma(CnnEvalArgs a) {
   }
 }
 
-// head: per-slot fc2 + in-graph softmax + mode tail (thread per slot)
+// head: per-slot fc2 + in-graph softmax + mode tail. ONE WAVE per slot
+// (same shape as cnn_head_fwd): lane o owns output o, the z1 row loads
+// once into two coalesced registers, softmax/argmax run as shuffle
+// reductions, atomic tails issue from lane 0.
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_eval_head(CnnEvalArgs a) {
-  for (long long slot = (long long)blockIdx.x * WG + threadIdx.x;
-       slot < a.n_slots; slot += (long long)gridDim.x * WG) {
+  const int l = threadIdx.x & 63;
+  for (long long slot = (long long)blockIdx.x * (WG / 64)
+                        + (threadIdx.x >> 6);
+       slot < a.n_slots; slot += (long long)gridDim.x * (WG / 64)) {
     const long long row = a.srow[slot];
     const float* wp = a.params + row * (long long)a.P;
     const float* z1 = a.z1e + slot * NH;
-    float s[64];
-    float zmax = -1e30f;
+    const float r0 = z1[l], r1 = z1[64 + l];
+    float zlane = 0.f;
     for (int o = 0; o < a.O; ++o) {
       const float* wo = wp + OFF_W2F + (long long)o * NH;
-      float z = wp[OFF_W2F + (long long)a.O * NH + o];
-      for (int h = 0; h < NH; ++h) z = fmaf(z1[h], wo[h], z);
-      s[o] = z;
-      zmax = fmaxf(zmax, z);
+      const float v = wave_sum(fmaf(r0, wo[l], r1 * wo[64 + l]));
+      if (l == o) zlane = v;
     }
-    float zsum = 0.f;
-    for (int o = 0; o < a.O; ++o) {
-      s[o] = __expf(s[o] - zmax);
-      zsum += s[o];
+    const bool live = l < a.O;
+    float z = live ? zlane + wp[OFF_W2F + (long long)a.O * NH + l]
+                   : -1e30f;
+    const float zmax = wave_max(z);
+    float e = live ? __expf(z - zmax) : 0.f;
+    const float s = e / wave_sum(e);   // the model OUTPUT (in-graph softmax)
+    // argmax, LOWEST index on ties (the sequential scan used strict >)
+    float bv = live ? s : -1e30f;
+    int bi = live ? l : 64;
+#pragma unroll
+    for (int m = 1; m < 64; m <<= 1) {
+      const float ov = __shfl_xor(bv, m, 64);
+      const int oi = __shfl_xor(bi, m, 64);
+      if (ov > bv || (ov == bv && oi < bi)) { bv = ov; bi = oi; }
     }
-    int best = 0;
-    float bv = -1e30f;
-    for (int o = 0; o < a.O; ++o) {
-      s[o] /= zsum;  // the model OUTPUT (in-graph softmax)
-      if (s[o] > bv) { bv = s[o]; best = o; }
-    }
+    const int best = bi;
     const long long tsk = a.stid[slot];
     const int yi = (int)a.sy[slot];
     if (a.mode == EV_DUMP) {
-      for (int o = 0; o < a.O; ++o)
-        a.outp[slot * (long long)a.O + o] = s[o];
+      if (live) a.outp[slot * (long long)a.O + l] = s;
     } else if (a.mode == EV_CONF) {
-      atomicAdd(&a.conf[(tsk * a.O + yi) * a.O + best], 1.0);
+      if (l == 0) atomicAdd(&a.conf[(tsk * a.O + yi) * a.O + best], 1.0);
     } else {
       // CE / mse on the softmax OUTPUT (double-softmax quirk)
-      float smax = -1e30f;
-      for (int o = 0; o < a.O; ++o) smax = fmaxf(smax, s[o]);
-      float ssum = 0.f;
-      for (int o = 0; o < a.O; ++o) ssum += __expf(s[o] - smax);
+      const float smax = wave_max(live ? s : -1e30f);
+      const float ssum = wave_sum(live ? __expf(s - smax) : 0.f);
       const float lse = logf(ssum) + smax;
-      atomicAdd(&a.correct[tsk], (double)((best == yi) ? 1.f : 0.f));
-      atomicAdd(&a.total[tsk], 1.0);
-      atomicAdd(&a.loss[tsk], (double)(lse - s[yi]));
-      if (a.mse) {
-        const float pt = __expf(s[yi] - lse);
-        atomicAdd(&a.mse[tsk], (double)((1.f - pt) * (1.f - pt)));
+      const float syi = __shfl(s, yi, 64);
+      if (l == 0) {
+        atomicAdd(&a.correct[tsk], (double)((best == yi) ? 1.f : 0.f));
+        atomicAdd(&a.total[tsk], 1.0);
+        atomicAdd(&a.loss[tsk], (double)(lse - syi));
+        if (a.mse) {
+          const float pt = __expf(syi - lse);
+          atomicAdd(&a.mse[tsk], (double)((1.f - pt) * (1.f - pt)));
+        }
       }
     }
   }
@@ -1541,7 +1562,7 @@ void cnn_train_epoch_impl(
   hipLaunchKernelGGL(cnn_fc1_fwd_mfma, dim3(G * mtiles * FC1_KS), dim3(WG),
                      0, s, a);
   L(cnn_fc1_act, GB * NH);
-  L(cnn_head_fwd, GB);
+  L(cnn_head_fwd, GB * 64);   // one wave per sample
   // backward
   L(cnn_fc2_wgrad, (long long)G * O * NH);
   L(cnn_fc2_dgrad, GB * NH);
@@ -1649,7 +1670,7 @@ torch::Tensor cnn_eval(
   hipLaunchKernelGGL(cnn_eval_pool, dim3((int)n_slots), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_eval_fc1_mfma, dim3((int)blk_row.size(0)),
                      dim3(WG), 0, s, a);
-  hipLaunchKernelGGL(cnn_eval_head, dim3(grid_for(n_slots)), dim3(WG), 0,
+  hipLaunchKernelGGL(cnn_eval_head, dim3(grid_for(n_slots * 64)), dim3(WG), 0,
                      s, a);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "cnn_eval launch");
   (void)max_len;
