@@ -1316,13 +1316,24 @@ void cnn_eval_pool(CnnEvalArgs a) {
   const long long slot = blockIdx.x;
   if (slot >= a.n_slots) return;
   const int tid = threadIdx.x;
-  __shared__ __attribute__((aligned(16))) float srow_[2 * S2 * C2];
+  __shared__ __attribute__((aligned(16))) float srow_[2][2 * S2 * C2];
   __shared__ __attribute__((aligned(16))) float sa2[NF];
   const float* z2 = a.z2e + slot * Z2N;
+  // double-buffered row-pair staging (same pattern as cnn_pool_fwd):
+  // the next pooled row's z2 loads issue before this row's max pass
+  float rst[12];
+#define PE_LOAD(py)                                                     \
+  _Pragma("unroll") for (int jj = 0; jj < 12; ++jj)                      \
+    rst[jj] = z2[(long long)(2 * (py)) * S2 * C2 + tid + jj * WG];
+#define PE_WRITE(buf)                                                   \
+  _Pragma("unroll") for (int jj = 0; jj < 12; ++jj)                      \
+    srow_[buf][tid + jj * WG] = rst[jj];
+  PE_LOAD(0);
+  PE_WRITE(0);
+  __syncthreads();
+  int cur = 0;
   for (int py = 0; py < SP; ++py) {
-    for (int q = tid; q < 2 * S2 * C2; q += WG)
-      srow_[q] = z2[(long long)(2 * py) * S2 * C2 + q];
-    __syncthreads();
+    if (py + 1 < SP) { PE_LOAD(py + 1); }
     for (int q = tid; q < C2 * SP; q += WG) {
       const int c = q / SP;
       const int px = q - c * SP;
@@ -1331,11 +1342,15 @@ void cnn_eval_pool(CnnEvalArgs a) {
       for (int dy = 0; dy < 2; ++dy)
 #pragma unroll
         for (int dx = 0; dx < 2; ++dx)
-          best = fmaxf(best, srow_[(dy * S2 + 2 * px + dx) * C2 + c]);
+          best = fmaxf(best, srow_[cur][(dy * S2 + 2 * px + dx) * C2 + c]);
       sa2[c * (SP * SP) + py * SP + px] = best;
     }
+    if (py + 1 < SP) { PE_WRITE(cur ^ 1); }
     __syncthreads();
+    cur ^= 1;
   }
+#undef PE_LOAD
+#undef PE_WRITE
   float* out = a.a2e + slot * (long long)NF;
   for (int e = tid; e < NF; e += WG) out[e] = sa2[e];
 }
