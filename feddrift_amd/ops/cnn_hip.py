@@ -219,7 +219,7 @@ class CnnHipEngine:
             x_mask = x_mask[order]
         csum = torch.cumsum(win_len, 0)
         start = 0
-        a2e = z1e = x1e = z2e = None
+        a2e = z1e = z1pe = x1e = z2e = None
         # reshaped conv2 weights for the eval MFMA B-operand:
         # [M, 9, 32, 64] from the [co][ci][ky][kx] flat slice (L2-hot,
         # shared across every block of a sweep)
@@ -240,6 +240,7 @@ class CnnHipEngine:
             if a2e is None or a2e.shape[0] < slots:
                 a2e = torch.empty(max(slots, 1), NF, device=dev)
                 z1e = torch.empty(max(slots, 1), NH, device=dev)
+                z1pe = torch.empty(max(slots, 1), 8 * NH, device=dev)
                 x1e = torch.empty(max(slots, 1), X1N, device=dev)
                 z2e = torch.empty(max(slots, 1), Z2N, device=dev)
             # per-slot metadata
@@ -262,7 +263,8 @@ class CnnHipEngine:
                 outp = torch.empty(slots, self.O, device=dev)
             out = self.mod.cnn_eval(
                 params.contiguous(), tr, ti, wo, wl, slot,
-                self._x_arena, self._y_arena, a2e, z1e, x1e, z2e, wtf_e,
+                self._x_arena, self._y_arena, a2e, z1e, z1pe, x1e, z2e,
+                wtf_e,
                 blk_row, blk_s0, blk_len, srow, stid, sy.contiguous(),
                 soff.contiguous(), swin.contiguous(),
                 xm, n_tasks, self.O, mode, want_mse, max_len, slots, outp)

@@ -1194,6 +1194,7 @@ struct CnnEvalArgs {
   int xm_per_task;
   float* __restrict__ a2e;              // [slots, NF]
   float* __restrict__ z1e;              // [slots, NH] (fc1 output)
+  float* __restrict__ z1pe;             // [slots, KS, NH] fc1 K-split partials
   // fc1 GEMM block metadata (slots grouped by model row, <=64 per block)
   const int64_t* __restrict__ blk_row;
   const int64_t* __restrict__ blk_s0;
@@ -1355,15 +1356,19 @@ void cnn_eval_pool(CnnEvalArgs a) {
   for (int e = tid; e < NF; e += WG) out[e] = sa2[e];
 }
 
-// fc1 eval as an MFMA tile GEMM: z1e[slot, h] = relu(a2e[slot, :] @
-// W1f[row(slot)]^T + b). Slots are grouped by model row (host sorts the
-// windows), each block owns up to 64 consecutive slots of one row:
-// 4 waves x 16 rows x 128 cols, f32-input MFMA (v_mfma_f32_16x16x4_f32 —
-// exact f32, the gfx950 f32 matrix path), BK=32 LDS staging with +1
-// padding against bank conflicts.
+// fc1 eval as an MFMA tile GEMM with the SAME K-split as the training
+// fc1 (the eval grid is otherwise tiny: ~4000 slots = 63 blocks for
+// 256 CUs — per-wave PMC showed healthy 37% MFMA busy but only a
+// quarter of the chip occupied): block (blk, ks) computes the partial
+// z1 for up to 64 consecutive same-row slots x 128 h over K-range
+// [ks*NF/KS, ...), partials land in z1pe and cnn_eval_fc1_act folds
+// them with bias + relu. 4 waves x 16 rows x 128 cols, f32-input MFMA,
+// BK=32 double-buffered register-staged LDS tiles with +1 padding.
+#define EVAL_FC1_KS 8
 extern "C" __global__ __launch_bounds__(WG)
 void cnn_eval_fc1_mfma(CnnEvalArgs a) {
-  const int blk = blockIdx.x;
+  const int ks = blockIdx.x % EVAL_FC1_KS;
+  const int blk = blockIdx.x / EVAL_FC1_KS;
   const long long row = a.blk_row[blk];
   const long long s0 = a.blk_s0[blk];
   const int mlen = (int)a.blk_len[blk];
@@ -1379,6 +1384,8 @@ void cnn_eval_fc1_mfma(CnnEvalArgs a) {
   for (int t = 0; t < 8; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
   const float* wp = a.params + row * (long long)a.P + OFF_W1F;
   const int r8 = tid >> 5, kk = tid & 31;
+  const int k_lo = ks * (NF / EVAL_FC1_KS);
+  const int k_hi = (ks + 1) * (NF / EVAL_FC1_KS);
   // register staging: 8 A rows + 16 B rows per thread per tile
   float ra[8], rb[16];
 #define EV_LOAD(k0)                                                     \
@@ -1394,12 +1401,12 @@ void cnn_eval_fc1_mfma(CnnEvalArgs a) {
     sA[buf][r8 + j * 8][kk] = ra[j];                                     \
   _Pragma("unroll") for (int j = 0; j < 16; ++j)                         \
     sB[buf][kk][r8 + j * 8] = rb[j];
-  EV_LOAD(0);
+  EV_LOAD(k_lo);
   EV_WRITE(0);
   __syncthreads();
   int cur = 0;
-  for (int k0 = 0; k0 < NF; k0 += EVAL_BK) {
-    if (k0 + EVAL_BK < NF) { EV_LOAD(k0 + EVAL_BK); }
+  for (int k0 = k_lo; k0 < k_hi; k0 += EVAL_BK) {
+    if (k0 + EVAL_BK < k_hi) { EV_LOAD(k0 + EVAL_BK); }
 #pragma unroll
     for (int kq = 0; kq < EVAL_BK / 4; ++kq) {
       const float av = sA[cur][wv * 16 + li][kq * 4 + lk];
@@ -1408,14 +1415,13 @@ void cnn_eval_fc1_mfma(CnnEvalArgs a) {
         acc[ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(
             av, sB[cur][kq * 4 + lk][ct * 16 + li], acc[ct], 0, 0, 0);
     }
-    if (k0 + EVAL_BK < NF) { EV_WRITE(cur ^ 1); }
+    if (k0 + EVAL_BK < k_hi) { EV_WRITE(cur ^ 1); }
     __syncthreads();
     cur ^= 1;
   }
 #undef EV_LOAD
 #undef EV_WRITE
-  // epilogue: bias + relu -> z1e[slot, h]
-  const float* bias = a.params + row * (long long)a.P + OFF_B1F;
+  // epilogue: partials out (bias + relu fold in cnn_eval_fc1_act)
 #pragma unroll
   for (int ct = 0; ct < 8; ++ct) {
 #pragma unroll
@@ -1423,10 +1429,27 @@ void cnn_eval_fc1_mfma(CnnEvalArgs a) {
       const int mrow = wv * 16 + lk * 4 + r;
       if (mrow < mlen) {
         const int h = ct * 16 + li;
-        const float z = acc[ct][r] + bias[h];
-        a.z1e[(s0 + mrow) * (long long)NH + h] = z > 0.f ? z : 0.f;
+        a.z1pe[((s0 + mrow) * EVAL_FC1_KS + ks) * (long long)NH + h] =
+            acc[ct][r];
       }
     }
+  }
+}
+
+// fold the eval fc1 K-split partials + bias, apply relu -> z1e
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_eval_fc1_act(CnnEvalArgs a) {
+  const long long total = a.n_slots * NH;
+  for (long long q = (long long)blockIdx.x * WG + threadIdx.x; q < total;
+       q += (long long)gridDim.x * WG) {
+    const long long slot = q / NH;
+    const int h = (int)(q - slot * NH);
+    const float* wp = a.params + a.srow[slot] * (long long)a.P;
+    float z = wp[OFF_B1F + h];
+#pragma unroll
+    for (int ks = 0; ks < EVAL_FC1_KS; ++ks)
+      z += a.z1pe[(slot * EVAL_FC1_KS + ks) * (long long)NH + h];
+    a.z1e[q] = z > 0.f ? z : 0.f;
   }
 }
 
@@ -1622,6 +1645,7 @@ torch::Tensor cnn_eval(
     torch::Tensor params, torch::Tensor task_row, torch::Tensor task_id,
     torch::Tensor off, torch::Tensor len, torch::Tensor slot,
     torch::Tensor x, torch::Tensor y, torch::Tensor a2e, torch::Tensor z1e,
+    torch::Tensor z1pe,
     torch::Tensor x1e, torch::Tensor z2e, torch::Tensor wtf_e,
     torch::Tensor blk_row, torch::Tensor blk_s0, torch::Tensor blk_len,
     torch::Tensor srow, torch::Tensor stid, torch::Tensor sy,
@@ -1654,6 +1678,7 @@ torch::Tensor cnn_eval(
   a.xm_per_task = x_mask.has_value() && x_mask->dim() == 2 ? 1 : 0;
   a.a2e = a2e.data_ptr<float>();
   a.z1e = z1e.data_ptr<float>();
+  a.z1pe = z1pe.data_ptr<float>();
   a.blk_row = blk_row.data_ptr<int64_t>();
   a.blk_s0 = blk_s0.data_ptr<int64_t>();
   a.blk_len = blk_len.data_ptr<int64_t>();
@@ -1683,7 +1708,10 @@ torch::Tensor cnn_eval(
   hipLaunchKernelGGL(cnn_eval_conv2_mfma, dim3((int)(n_slots * 9)),
                      dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_eval_pool, dim3((int)n_slots), dim3(WG), 0, s, a);
-  hipLaunchKernelGGL(cnn_eval_fc1_mfma, dim3((int)blk_row.size(0)),
+  hipLaunchKernelGGL(cnn_eval_fc1_mfma,
+                     dim3((int)blk_row.size(0) * EVAL_FC1_KS),
+                     dim3(WG), 0, s, a);
+  hipLaunchKernelGGL(cnn_eval_fc1_act, dim3(grid_for(n_slots * NH)),
                      dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_eval_head, dim3(grid_for(n_slots * 64)), dim3(WG), 0,
                      s, a);
