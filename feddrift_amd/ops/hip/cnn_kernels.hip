@@ -1037,6 +1037,90 @@ void cnn_conv2_dgrad_mfma2(CnnArgs a) {
     }
 }
 
+// conv2 dgrad v3 — v2 with 128-pixel tiles: each wave owns TWO
+// 16-pixel M-fragments, so every weight load feeds two MFMAs (the B
+// operand is the per-MFMA global-issue cost in v2) and each wave runs
+// four independent accumulator chains. Region = 8 dz rows (52 KB,
+// 3 blocks/CU).
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_conv2_dgrad_mfma3(CnnArgs a) {
+  const int mt = blockIdx.x % 6;          // 676 / 128 -> 6 tiles
+  const int gb = blockIdx.x / 6;
+  const int g = gb / a.B;
+  const int b = gb - g * a.B;
+  if (b >= step_n(a, g)) return;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sD[8 * S2][C2 + 4];
+  const float* dz = a.zz2 + ((long long)g * a.B + b) * Z2N;
+  const float* wt = a.wtd + (long long)g * 9 * 2048;
+  float* dx1 = a.dx1 + ((long long)g * a.B + b) * X1N;
+  const int p0 = mt * 128;
+  const int y0 = p0 / S1;
+  // region = dz rows [y0-2, y0+6): pixels y in [y0, y0+5], taps reach
+  // oy in [y0-2, y0+5]; rows outside [0, S2) stage as zeros
+  for (int e = tid; e < 8 * S2 * C2; e += WG) {
+    const int rr = e / (S2 * C2);
+    const int rem = e - rr * S2 * C2;
+    const int cc = rem / C2;
+    const int ch = rem - cc * C2;
+    const int dzrow = y0 - 2 + rr;
+    sD[rr * S2 + cc][ch] =
+        (dzrow >= 0 && dzrow < S2)
+            ? dz[((long long)dzrow * S2 + cc) * C2 + ch] : 0.f;
+  }
+  __syncthreads();
+  const int p_a = p0 + wv * 32 + li;       // M-fragment 0
+  const int p_b = p_a + 16;                // M-fragment 1
+  const int ya = p_a / S1, xa = p_a - ya * S1;
+  const int yb = p_b / S1, xb = p_b - yb * S1;
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+    for (int ct = 0; ct < 2; ++ct) acc[mf][ct] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int kyx = 0; kyx < 9; ++kyx) {
+    const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+    const int oya = ya - ky, oxa = xa - kx;
+    const int oyb = yb - ky, oxb = xb - kx;
+    const bool oka = oya >= 0 && oya < S2 && oxa >= 0 && oxa < S2;
+    const bool okb = oyb >= 0 && oyb < S2 && oxb >= 0 && oxb < S2;
+    const int ra = oka ? (oya - y0 + 2) * S2 + oxa : 0;
+    const int rb = okb ? (oyb - y0 + 2) * S2 + oxb : 0;
+    const float* wk = wt + kyx * 2048;
+#pragma unroll
+    for (int ks = 0; ks < C2 / 4; ++ks) {
+      const int co = ks * 4 + lk;
+      float av0 = sD[ra][co];
+      float av1 = sD[rb][co];
+      av0 = oka ? av0 : 0.f;
+      av1 = okb ? av1 : 0.f;
+      const float wk0 = wk[co * C1 + li];
+      const float wk1 = wk[co * C1 + 16 + li];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(av0, wk0,
+                                                       acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(av0, wk1,
+                                                       acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(av1, wk0,
+                                                       acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(av1, wk1,
+                                                       acc[1][1], 0, 0, 0);
+    }
+  }
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+    for (int ct = 0; ct < 2; ++ct)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int pix = p0 + wv * 32 + mf * 16 + lk * 4 + r;
+        if (pix < 676)
+          dx1[(long long)pix * C1 + ct * 16 + li] = acc[mf][ct][r];
+      }
+}
+
 // conv1 wgrad, stage 1: per-(g, b) partials into the dz2 scratch region
 // (reused: dz2 is [G, B, 64] and conv1 has 288+32=320 grad entries, so
 // partials use their own ws buffer c1part [G, B, 320]). Deterministic
@@ -1613,20 +1697,23 @@ void cnn_train_epoch_impl(
   hipLaunchKernelGGL(cnn_conv2_wgrad_mfma, dim3(G * a.w2ms), dim3(WG),
                      0, s, a);
   L(cnn_conv2_wgrad_reduce, (long long)G * (9 * 2048 + C2));
-  // dgrad variant switch, A/B-measured on both ends of the fleet-size
-  // axis: the fwd-shaped v2 (4 blocks/CU) wins at small fleets where
-  // occupancy is the bound (config-3 probe: 463 -> 408 us, train
-  // 8.61 -> 8.29 ms/round), the all-LDS glds pipeline v1 wins at scale
-  // where its zero-idle DMA staging is the bound (config-5, 3400
-  // clients: 5.69 vs 5.97 s/round). FEDDRIFT_DGRAD=1|2 forces either.
+  // dgrad variant switch, A/B-measured at both ends of the fleet-size
+  // axis (FEDDRIFT_DGRAD=1|2|3 forces one): v3 (128-pixel tiles, every
+  // weight load shared by two M-fragments, 4 MFMA chains/wave) matches
+  // v2 at small fleets (config-3 train 8.0 ms/round both) and beats
+  // both the fwd-shaped v2 (5.97) and the all-LDS glds pipeline v1
+  // (5.6-5.7) at 3400-client scale: 5.2 s/round -> default everywhere.
   static const int dgrad_env = [] {
     const char* e = getenv("FEDDRIFT_DGRAD");
     return e ? atoi(e) : 0;
   }();
-  const int dgrad_v = dgrad_env ? dgrad_env : (G <= 128 ? 2 : 1);
+  const int dgrad_v = dgrad_env ? dgrad_env : 3;
   if (dgrad_v == 1)
     hipLaunchKernelGGL(cnn_conv2_dgrad_mfma, dim3((int)GB), dim3(WG),
                        (9 * 2048 + 2 * DG_RC * 64) * sizeof(float), s, a);
+  else if (dgrad_v == 3)
+    hipLaunchKernelGGL(cnn_conv2_dgrad_mfma3, dim3((int)GB * 6), dim3(WG),
+                       0, s, a);
   else
     hipLaunchKernelGGL(cnn_conv2_dgrad_mfma2, dim3((int)GB * 11), dim3(WG),
                        0, s, a);
