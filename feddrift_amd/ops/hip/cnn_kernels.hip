@@ -257,6 +257,82 @@ void cnn_conv2_fwd_mfma(CnnArgs a) {
   }
 }
 
+// conv2 forward, 128-pixel tiles (the dgrad-v3 trick applied to fwd):
+// each wave owns TWO 16-pixel M-fragments, so every weight load feeds
+// two MFMAs (fwd was issue-stalled 0.65 with one global per MFMA) and
+// each wave runs eight accumulator chains. Region = 8 input rows,
+// 27.5 KB LDS.
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_conv2_fwd_mfma2(CnnArgs a) {
+  const int pt = blockIdx.x % 5;           // ceil(576 / 128)
+  const int gb = blockIdx.x / 5;
+  const int g = gb / a.B;
+  const int b = gb - g * a.B;
+  if (b >= step_n(a, g)) return;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sR[8 * S1][C1 + 1];
+  const int p0 = pt * 128;
+  const int r0 = p0 / S2;
+  const float* x1 = a.x1 + (((long long)g * a.B + b) * 676
+                            + (long long)r0 * S1) * C1;
+  {
+    const int nrow = min(8 * S1, (S1 - r0) * S1);
+    const int r8 = tid >> 5, kk = tid & 31;
+    for (int rr = r8; rr < 8 * S1; rr += 8)
+      sR[rr][kk] = (rr < nrow) ? x1[(long long)rr * C1 + kk] : 0.f;
+  }
+  __syncthreads();
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+    for (int t = 0; t < 4; ++t) acc[mf][t] = {0.f, 0.f, 0.f, 0.f};
+  const int p_a = p0 + wv * 32 + li;
+  const int p_b = p_a + 16;
+  const int arow_a = (p_a / S2 - r0) * S1 + (p_a - (p_a / S2) * S2);
+  const int arow_b = (p_b / S2 - r0) * S1 + (p_b - (p_b / S2) * S2);
+  const float* wt = a.wtf + (long long)g * 9 * 2048;
+  for (int kyx = 0; kyx < 9; ++kyx) {
+    const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+    const int offa = arow_a + ky * S1 + kx;
+    const int offb = arow_b + ky * S1 + kx;
+    const float* wk = wt + kyx * 2048;
+#pragma unroll
+    for (int ks = 0; ks < C1 / 4; ++ks) {
+      const float av0 = sR[offa][ks * 4 + lk];
+      const float av1 = sR[offb][ks * 4 + lk];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        const float wv_ = wk[(ks * 4 + lk) * C2 + ct * 16 + li];
+        acc[0][ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(av0, wv_,
+                                                          acc[0][ct],
+                                                          0, 0, 0);
+        acc[1][ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(av1, wv_,
+                                                          acc[1][ct],
+                                                          0, 0, 0);
+      }
+    }
+  }
+  const float* bias = a.work + (long long)g * a.P + OFF_B2C;
+  float* z2 = a.zz2 + ((long long)g * a.B + b) * Z2N;
+#pragma unroll
+  for (int ct = 0; ct < 4; ++ct) {
+    const int co = ct * 16 + li;
+    const float bb = bias[co];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int pp = p0 + wv * 32 + mf * 16 + lk * 4 + r;
+        if (pp < Z2N / C2)
+          z2[(long long)pp * C2 + co] = acc[mf][ct][r] + bb;
+      }
+  }
+}
+
 // maxpool + dropout1 over the channels-last z2, block per (g, b):
 // z2 rows stage through LDS (coalesced channels-last reads), the pooled
 // a2/pidx tiles assemble in LDS and store with one coalesced pass in the
@@ -1393,6 +1469,76 @@ void cnn_eval_conv2_mfma(CnnEvalArgs a) {
   }
 }
 
+// eval conv2, 128-pixel tiles (see cnn_conv2_fwd_mfma2)
+extern "C" __global__ __launch_bounds__(WG)
+void cnn_eval_conv2_mfma2(CnnEvalArgs a) {
+  const int pt = blockIdx.x % 5;
+  const long long slot = blockIdx.x / 5;
+  if (slot >= a.n_slots) return;
+  const int tid = threadIdx.x;
+  const int wv = tid >> 6;
+  const int l = tid & 63;
+  const int li = l & 15, lk = l >> 4;
+  __shared__ __attribute__((aligned(16))) float sR[8 * S1][C1 + 1];
+  const int p0 = pt * 128;
+  const int r0 = p0 / S2;
+  const float* x1 = a.x1e + slot * X1N + (long long)r0 * S1 * C1;
+  {
+    const int nrow = min(8 * S1, (S1 - r0) * S1);
+    const int r8 = tid >> 5, kk = tid & 31;
+    for (int rr = r8; rr < 8 * S1; rr += 8)
+      sR[rr][kk] = (rr < nrow) ? x1[(long long)rr * C1 + kk] : 0.f;
+  }
+  __syncthreads();
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+    for (int t = 0; t < 4; ++t) acc[mf][t] = {0.f, 0.f, 0.f, 0.f};
+  const int p_a = p0 + wv * 32 + li;
+  const int p_b = p_a + 16;
+  const int arow_a = (p_a / S2 - r0) * S1 + (p_a - (p_a / S2) * S2);
+  const int arow_b = (p_b / S2 - r0) * S1 + (p_b - (p_b / S2) * S2);
+  const long long row = a.srow[slot];
+  const float* wt = a.wtf_e + row * (9 * 2048);
+  for (int kyx = 0; kyx < 9; ++kyx) {
+    const int ky = kyx / 3, kx = kyx - (kyx / 3) * 3;
+    const int offa = arow_a + ky * S1 + kx;
+    const int offb = arow_b + ky * S1 + kx;
+    const float* wk = wt + kyx * 2048;
+#pragma unroll
+    for (int ks = 0; ks < C1 / 4; ++ks) {
+      const float av0 = sR[offa][ks * 4 + lk];
+      const float av1 = sR[offb][ks * 4 + lk];
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        const float wv_ = wk[(ks * 4 + lk) * C2 + ct * 16 + li];
+        acc[0][ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(av0, wv_,
+                                                          acc[0][ct],
+                                                          0, 0, 0);
+        acc[1][ct] = __builtin_amdgcn_mfma_f32_16x16x4f32(av1, wv_,
+                                                          acc[1][ct],
+                                                          0, 0, 0);
+      }
+    }
+  }
+  const float* bias = a.params + row * (long long)a.P + OFF_B2C;
+  float* z2 = a.z2e + slot * Z2N;
+#pragma unroll
+  for (int ct = 0; ct < 4; ++ct) {
+    const int co = ct * 16 + li;
+    const float bb = bias[co];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int pp = p0 + wv * 32 + mf * 16 + lk * 4 + r;
+        if (pp < Z2N / C2)
+          z2[(long long)pp * C2 + co] = acc[mf][ct][r] + bb;
+      }
+  }
+}
+
 // eval maxpool, block per slot: channels-last z2 reads coalesced via
 // LDS row staging, pooled a2e assembled in LDS and stored in one
 // coalesced pass (torch-flatten layout; no dropout, no argmax at eval)
@@ -1602,6 +1748,16 @@ void cnn_eval_head(CnnEvalArgs a) {
 // host side
 // ---------------------------------------------------------------------------
 
+// conv2 fwd/eval variant: 64-pixel tiles (1) vs 128-pixel shared-load
+// tiles (2); FEDDRIFT_CONV2FWD overrides
+static bool fwd_v2() {
+  static const int v = [] {
+    const char* e = getenv("FEDDRIFT_CONV2FWD");
+    return e ? atoi(e) : 1;
+  }();
+  return v == 2;
+}
+
 static int grid_for(long long total) {
   long long b = (total + WG - 1) / WG;
   if (b > 16384) b = 16384;          // grid-stride beyond
@@ -1678,8 +1834,12 @@ void cnn_train_epoch_impl(
   // forward
   L(cnn_w2_reshape, (long long)G * C2 * C1 * 9);
   hipLaunchKernelGGL(cnn_conv1_fwd, dim3((int)GB), dim3(WG), 0, s, a);
-  hipLaunchKernelGGL(cnn_conv2_fwd_mfma, dim3((int)GB * 9), dim3(WG), 0,
-                     s, a);
+  if (fwd_v2())
+    hipLaunchKernelGGL(cnn_conv2_fwd_mfma2, dim3((int)GB * 5), dim3(WG), 0,
+                       s, a);
+  else
+    hipLaunchKernelGGL(cnn_conv2_fwd_mfma, dim3((int)GB * 9), dim3(WG), 0,
+                       s, a);
   hipLaunchKernelGGL(cnn_pool_fwd, dim3((int)GB), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_fc1_fwd_mfma, dim3(G * mtiles * FC1_KS), dim3(WG),
                      0, s, a);
@@ -1792,8 +1952,12 @@ torch::Tensor cnn_eval(
   auto s = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(cnn_eval_conv1, dim3((int)n_slots), dim3(WG), 0, s,
                      a);
-  hipLaunchKernelGGL(cnn_eval_conv2_mfma, dim3((int)(n_slots * 9)),
-                     dim3(WG), 0, s, a);
+  if (fwd_v2())
+    hipLaunchKernelGGL(cnn_eval_conv2_mfma2, dim3((int)(n_slots * 5)),
+                       dim3(WG), 0, s, a);
+  else
+    hipLaunchKernelGGL(cnn_eval_conv2_mfma, dim3((int)(n_slots * 9)),
+                       dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_eval_pool, dim3((int)n_slots), dim3(WG), 0, s, a);
   hipLaunchKernelGGL(cnn_eval_fc1_mfma,
                      dim3((int)blk_row.size(0) * EVAL_FC1_KS),
