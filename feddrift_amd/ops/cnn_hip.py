@@ -99,13 +99,24 @@ class CnnHipEngine:
             "wtf": f(G * 9 * 2048),
             "wtd": f(G * 9 * 2048),
             "z1part": f(G * 8 * B * NH),
-            "w2ms": min(32, max(1, 512 // max(G, 1))),
-            "w2part": f(G * min(32, max(1, 512 // max(G, 1))) * 9 * 2048),
+            "w2ms": self._w2ms(G),
+            "w2part": f(G * self._w2ms(G) * 9 * 2048),
             "b2part": f(G * B * 64),
             "grad": f(G, self.P),
         }
         self._ws = ws
         return ws
+
+    @staticmethod
+    def _w2ms(G: int) -> int:
+        """conv2-wgrad m-split count: G*w2ms blocks at 2 blocks/CU need
+        >= 512 resident slots plus queued tail to fill 256 CUs (env
+        FEDDRIFT_W2MS overrides for A/B runs)."""
+        import os
+        ov = os.environ.get("FEDDRIFT_W2MS")
+        if ov:
+            return max(1, int(ov))
+        return min(64, max(1, 1024 // max(G, 1)))
 
     def _chunk_pairs(self, B: int) -> int:
         per_pair = (2 * X1N + 3 * NF + Z2N + 2 * NH + 64) * B * 4 \
