@@ -303,3 +303,25 @@ def test_cnn_hip_engine_not_selected_on_cpu(tmp_path):
     job = FLJob(cfg, Communicator(device=__import__("torch").device("cpu")),
                 MetricLogger(enabled=False, to_file=False), dataset=ds)
     assert isinstance(job.mod_engine, VmapEngine), type(job.mod_engine)
+
+
+def test_cnn_hip_engine_tunables(monkeypatch):
+    """Host-side sizing contracts of the CNN kernel engine (pure
+    functions, no GPU): the wgrad m-split fills >=1024 blocks capped at
+    64 (FEDDRIFT_W2MS overrides), and pair chunking respects the
+    workspace budget with the 2048-pair kernel cap."""
+    from feddrift_amd.ops.cnn_hip import CnnHipEngine
+    w2ms = CnnHipEngine._w2ms
+    assert w2ms(1) == 64 and w2ms(13) == 64     # small fleets hit the cap
+    assert w2ms(100) == 10                       # 1024 // 100
+    assert w2ms(2048) == 1                       # scale: no extra split
+    monkeypatch.setenv("FEDDRIFT_W2MS", "32")
+    assert w2ms(13) == 32                        # A/B override
+    monkeypatch.delenv("FEDDRIFT_W2MS")
+    eng = CnnHipEngine.__new__(CnnHipEngine)     # no GPU init
+    eng.WS_BUDGET = CnnHipEngine.WS_BUDGET
+    eng.P = 1_199_882
+    assert eng._chunk_pairs(1) >= eng._chunk_pairs(100)  # bigger B, fewer
+    assert 1 <= eng._chunk_pairs(500) <= 2048
+    eng.WS_BUDGET = 1
+    assert eng._chunk_pairs(100) == 1            # floor at one pair
