@@ -20,7 +20,8 @@ from feddrift_amd.ops.module_vmap import VmapEngine
 
 DEV = torch.device("cuda:0")
 N_CONFIGS = int(sys.argv[1]) if len(sys.argv) > 1 else 16
-rng = np.random.default_rng(20260914)
+SEED = int(sys.argv[2]) if len(sys.argv) > 2 else 20260914
+rng = np.random.default_rng(SEED)
 t0 = time.time()
 
 for trial in range(N_CONFIGS):
